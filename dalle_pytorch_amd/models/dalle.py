@@ -1,0 +1,352 @@
+"""DALLE: text+image autoregressive transformer.
+
+Parity target: reference dalle_pytorch.py:352-671. Checkpoint-compatible key
+schema: ``text_emb``, ``image_emb``, ``to_logits.{0,1}``,
+``transformer.*``; the frozen VAE is a submodule so ``vae.*`` keys appear in
+the state dict, exactly as in the reference.
+
+Behavior kept bit-for-bit:
+* per-position unique padding ids substituted where text==0 (:595-596),
+* <bos> prepended and the final position dropped at train time (:600,629-631),
+* logits masked so text positions predict text vocab and image positions
+  image vocab, with finite ``-finfo.max`` fill (:450-455,648-652),
+* loss = (CE_text + w * CE_img) / (w + 1), default w=7 (:667-670),
+* cached decode passes only the newest token through the stack (:637-639),
+* classifier-free guidance via a second null-cond forward (:564-574).
+"""
+
+import torch
+import torch.nn.functional as F
+from torch import nn
+
+from dalle_pytorch_amd.models.dvae import DiscreteVAE
+from dalle_pytorch_amd.models.vae_adapters import OpenAIDiscreteVAE, VQGanVAE
+from dalle_pytorch_amd.models.transformer import Transformer, DivideMax
+from dalle_pytorch_amd.models.positional import AxialPositionalEmbedding
+
+
+class _Zero:
+    """Stand-in for a positional embedding when rotary is active."""
+
+    def __call__(self, *args, **kwargs):
+        return 0
+
+
+def _safe_log(t, eps=1e-20):
+    return torch.log(t.clamp(min=eps))
+
+
+def gumbel_sample(t, temperature=1., dim=-1):
+    noise = torch.zeros_like(t).uniform_(0, 1)
+    gumbel = -_safe_log(-_safe_log(noise))
+    return (t / temperature + gumbel).argmax(dim=dim)
+
+
+def top_k(logits, thres=0.5):
+    num_logits = logits.shape[-1]
+    k = max(int((1 - thres) * num_logits), 1)
+    val, ind = torch.topk(logits, k)
+    out = torch.full_like(logits, float('-inf'))
+    out.scatter_(1, ind, val)
+    return out
+
+
+class SharedEmbedding(nn.Embedding):
+    """Embedding view onto a slice of the output head's weight (weight tying;
+    reference dalle_pytorch.py:71-83)."""
+
+    def __init__(self, linear, start_index, end_index, **kwargs):
+        super().__init__(end_index - start_index, linear.weight.shape[1], **kwargs)
+        del self.weight
+        self.linear = linear
+        self.start_index = start_index
+        self.end_index = end_index
+
+    def forward(self, input):
+        return F.embedding(
+            input, self.linear.weight[self.start_index:self.end_index],
+            self.padding_idx, self.max_norm, self.norm_type,
+            self.scale_grad_by_freq, self.sparse)
+
+
+class DALLE(nn.Module):
+    def __init__(
+        self,
+        *,
+        dim,
+        vae,
+        num_text_tokens=10000,
+        text_seq_len=256,
+        depth,
+        heads=8,
+        dim_head=64,
+        reversible=False,
+        attn_dropout=0.,
+        ff_dropout=0,
+        sparse_attn=False,
+        attn_types=None,
+        loss_img_weight=7,
+        stable=False,
+        sandwich_norm=False,
+        shift_tokens=True,
+        rotary_emb=True,
+        shared_attn_ids=None,
+        shared_ff_ids=None,
+        share_input_output_emb=False,
+        optimize_for_inference=False,
+    ):
+        super().__init__()
+        assert isinstance(vae, (DiscreteVAE, OpenAIDiscreteVAE, VQGanVAE)), \
+            'vae must be an instance of DiscreteVAE'
+
+        image_size = vae.image_size
+        num_image_tokens = vae.num_tokens
+        image_fmap_size = vae.image_size // (2 ** vae.num_layers)
+        image_seq_len = image_fmap_size ** 2
+
+        # one unique padding token per text position
+        num_text_tokens = num_text_tokens + text_seq_len
+
+        self.text_pos_emb = nn.Embedding(text_seq_len + 1, dim) if not rotary_emb else _Zero()
+        self.image_pos_emb = AxialPositionalEmbedding(
+            dim, axial_shape=(image_fmap_size, image_fmap_size)) if not rotary_emb else _Zero()
+
+        self.num_text_tokens = num_text_tokens
+        self.num_image_tokens = num_image_tokens
+        self.text_seq_len = text_seq_len
+        self.image_seq_len = image_seq_len
+
+        seq_len = text_seq_len + image_seq_len
+        total_tokens = num_text_tokens + num_image_tokens
+        self.total_tokens = total_tokens
+        self.total_seq_len = seq_len
+
+        self.vae = vae
+        for p in self.vae.parameters():
+            p.requires_grad = False
+
+        self.transformer = Transformer(
+            dim=dim,
+            causal=True,
+            seq_len=seq_len,
+            depth=depth,
+            heads=heads,
+            dim_head=dim_head,
+            reversible=reversible,
+            attn_dropout=attn_dropout,
+            ff_dropout=ff_dropout,
+            attn_types=attn_types,
+            image_fmap_size=image_fmap_size,
+            sparse_attn=sparse_attn,
+            stable=stable,
+            sandwich_norm=sandwich_norm,
+            shift_tokens=shift_tokens,
+            rotary_emb=rotary_emb,
+            shared_attn_ids=shared_attn_ids,
+            shared_ff_ids=shared_ff_ids,
+            optimize_for_inference=optimize_for_inference,
+        )
+
+        self.stable = stable
+        if stable:
+            self.norm_by_max = DivideMax(dim=-1)
+
+        self.to_logits = nn.Sequential(
+            nn.LayerNorm(dim),
+            nn.Linear(dim, self.total_tokens),
+        )
+
+        if share_input_output_emb:
+            self.text_emb = SharedEmbedding(self.to_logits[1], 0, num_text_tokens)
+            self.image_emb = SharedEmbedding(self.to_logits[1], num_text_tokens, total_tokens)
+        else:
+            self.text_emb = nn.Embedding(num_text_tokens, dim)
+            self.image_emb = nn.Embedding(num_image_tokens, dim)
+
+        seq_range = torch.arange(seq_len).reshape(1, -1, 1)
+        logits_range = torch.arange(total_tokens).reshape(1, 1, -1)
+        logits_mask = (
+            ((seq_range >= text_seq_len) & (logits_range < num_text_tokens)) |
+            ((seq_range < text_seq_len) & (logits_range >= num_text_tokens))
+        )
+        self.register_buffer('logits_mask', logits_mask, persistent=False)
+        self.loss_img_weight = loss_img_weight
+
+    # ------------------------------------------------------------------ gen
+
+    @torch.no_grad()
+    def generate_texts(self, tokenizer, text=None, *, filter_thres=0.5, temperature=1.):
+        was_training = self.training
+        self.eval()
+        device = next(self.parameters()).device
+        if text is None or text == '':
+            text_tokens = torch.tensor([[0]], device=device)
+        else:
+            text_tokens = torch.tensor(
+                tokenizer.tokenizer.encode(text), device=device).unsqueeze(0)
+
+        for _ in range(text_tokens.shape[1], self.text_seq_len):
+            tokens = self.text_emb(text_tokens)
+            tokens = tokens + self.text_pos_emb(
+                torch.arange(text_tokens.shape[1], device=device))
+
+            seq_len = tokens.shape[1]
+            out = self.transformer(tokens)
+            if self.stable:
+                out = self.norm_by_max(out)
+            logits = self.to_logits(out)
+
+            logits_mask = self.logits_mask[:, :seq_len]
+            logits = logits.masked_fill(logits_mask, -torch.finfo(logits.dtype).max)
+            logits = logits[:, -1, :]
+
+            filtered = top_k(logits, thres=filter_thres)
+            sample = gumbel_sample(filtered, temperature=temperature, dim=-1)
+            text_tokens = torch.cat((text_tokens, sample[:, None]), dim=-1)
+
+        self.train(was_training)
+        pad_tokens = set(range(self.num_text_tokens - self.text_seq_len,
+                               self.num_text_tokens))
+        texts = [tokenizer.tokenizer.decode(t, pad_tokens=pad_tokens)
+                 for t in text_tokens]
+        return text_tokens, texts
+
+    @torch.no_grad()
+    def generate_images(
+        self,
+        text,
+        *,
+        clip=None,
+        filter_thres=0.5,
+        temperature=1.,
+        img=None,
+        num_init_img_tokens=None,
+        cond_scale=1.,
+        use_cache=False,
+    ):
+        was_training = self.training
+        self.eval()
+        vae = self.vae
+        text_seq_len, image_seq_len = self.text_seq_len, self.image_seq_len
+        num_text_tokens = self.num_text_tokens
+        total_len = text_seq_len + image_seq_len
+
+        text = text[:, :text_seq_len]
+        out = text
+
+        if img is not None:
+            image_size = vae.image_size
+            assert img.shape[1:] == (3, image_size, image_size), \
+                f'input image must have the correct image size {image_size}'
+            indices = vae.get_codebook_indices(img)
+            num_img_tokens = num_init_img_tokens if num_init_img_tokens is not None \
+                else int(0.4375 * image_seq_len)
+            assert num_img_tokens < image_seq_len, \
+                'priming token count must be < image sequence length'
+            out = torch.cat((out, indices[:, :num_img_tokens]), dim=-1)
+
+        cache = {} if use_cache else None
+        for cur_len in range(out.shape[1], total_len):
+            is_image = cur_len >= text_seq_len
+            text_part, image_part = out[:, :text_seq_len], out[:, text_seq_len:]
+            logits = self.forward_with_cond_scale(
+                text_part, image_part, cond_scale=cond_scale, cache=cache)
+            logits = logits[:, -1, :]
+            filtered = top_k(logits, thres=filter_thres)
+            sample = gumbel_sample(filtered, temperature=temperature, dim=-1)
+            sample -= num_text_tokens if is_image else 0
+            out = torch.cat((out, sample[:, None]), dim=-1)
+
+        text_seq = out[:, :text_seq_len]
+        img_seq = out[:, -image_seq_len:]
+        images = vae.decode(img_seq)
+        self.train(was_training)
+
+        if clip is not None:
+            scores = clip(text_seq, images, return_loss=False)
+            return images, scores
+        return images
+
+    def forward_with_cond_scale(self, *args, cond_scale=1, cache=None, **kwargs):
+        if cond_scale == 1:
+            return self(*args, cache=cache, **kwargs)
+        prev_cache = cache.copy() if cache is not None else None
+        logits = self(*args, cache=cache, **kwargs)
+        null_logits = self(*args, null_cond_prob=1., cache=prev_cache, **kwargs)
+        return null_logits + (logits - null_logits) * cond_scale
+
+    # ---------------------------------------------------------------- train
+
+    def forward(self, text, image=None, return_loss=False, null_cond_prob=0., cache=None):
+        assert text.shape[-1] == self.text_seq_len, \
+            f'text length {text.shape[-1]} != text_seq_len {self.text_seq_len}'
+        batch, device = text.shape[0], text.device
+        total_seq_len = self.total_seq_len
+
+        if null_cond_prob > 0:
+            null_mask = torch.rand(batch, device=device) < null_cond_prob
+            text = text * (~null_mask).unsqueeze(1)
+
+        # unique padding ids per position
+        text_range = torch.arange(self.text_seq_len, device=device) \
+            + (self.num_text_tokens - self.text_seq_len)
+        text = torch.where(text == 0, text_range, text)
+
+        text = F.pad(text, (1, 0), value=0)  # <bos>
+        tokens = self.text_emb(text)
+        tokens = tokens + self.text_pos_emb(torch.arange(text.shape[1], device=device))
+
+        seq_len = tokens.shape[1]
+
+        if image is not None and image.nelement() > 0:
+            if image.dim() == 4:
+                image_size = self.vae.image_size
+                channels = self.vae.channels
+                assert tuple(image.shape[1:]) == (channels, image_size, image_size), \
+                    f'invalid image of dimensions {tuple(image.shape)}'
+                image = self.vae.get_codebook_indices(image)
+            image_len = image.shape[1]
+            image_emb = self.image_emb(image)
+            image_emb = image_emb + self.image_pos_emb(image_emb)
+            tokens = torch.cat((tokens, image_emb), dim=1)
+            seq_len += image_len
+
+        # drop the final position at train time: nothing follows it
+        if tokens.shape[1] > total_seq_len:
+            seq_len -= 1
+            tokens = tokens[:, :-1]
+
+        if self.stable:
+            alpha = 0.1
+            tokens = tokens * alpha + tokens.detach() * (1 - alpha)
+
+        if cache is not None and cache.get('offset'):
+            tokens = tokens[:, -1:]
+        out = self.transformer(tokens, cache=cache)
+
+        if self.stable:
+            out = self.norm_by_max(out)
+
+        logits = self.to_logits(out)
+
+        logits_mask = self.logits_mask[:, :seq_len]
+        if cache is not None and cache.get('offset'):
+            logits_mask = logits_mask[:, -1:]
+        logits = logits.masked_fill(logits_mask, -torch.finfo(logits.dtype).max)
+
+        if cache is not None:
+            cache['offset'] = cache.get('offset', 0) + logits.shape[1]
+
+        if not return_loss:
+            return logits
+
+        assert image is not None, 'when training, image must be supplied'
+        offsetted_image = image + self.num_text_tokens
+        labels = torch.cat((text[:, 1:], offsetted_image), dim=1)
+
+        logits = logits.transpose(1, 2)
+        loss_text = F.cross_entropy(
+            logits[:, :, :self.text_seq_len], labels[:, :self.text_seq_len])
+        loss_img = F.cross_entropy(
+            logits[:, :, self.text_seq_len:], labels[:, self.text_seq_len:])
+        return (loss_text + self.loss_img_weight * loss_img) / (self.loss_img_weight + 1)
