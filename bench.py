@@ -1,0 +1,186 @@
+#!/usr/bin/env python3
+"""Benchmark: flagship DALL-E training step on 1..8 MI355X GPUs.
+
+Metric (BASELINE.json): img-tokens/sec training DALL-E dim=1024 / depth=12 /
+heads=16, seq 1280 (256 text + 1024 image tokens), bf16, synthetic data,
+random-init weights. ``value`` is the WHOLE-JOB aggregate across all ranks.
+
+Run directly (single GPU) or under torchrun for N>1:
+    python bench.py --gpus 1 --steps 10 --warmup 3
+    python -m torch.distributed.run --nnodes=1 --nproc-per-node 8 \
+        --master-addr 127.0.0.1 bench.py --gpus 8 --steps 10 --warmup 3
+
+Each timed step is a full training iteration: frozen-VAE image encode,
+transformer forward, loss, backward, bucketed RCCL gradient all-reduce,
+grad clip, Adam step — nothing cached, nothing skipped.
+"""
+
+import argparse
+import json
+import os
+import sys
+import time
+
+import torch
+
+CONFIGS = {
+    # BASELINE.json configs B-D (training). Default: the headline C shape.
+    'b': dict(dim=512, depth=12, heads=8, attn_types=('full',),
+              reversible=False, batch_size=8, vae='dvae'),
+    'c': dict(dim=1024, depth=12, heads=16, attn_types=('axial_row', 'axial_col'),
+              reversible=True, batch_size=8, vae='dvae'),
+    'c_full': dict(dim=1024, depth=12, heads=16, attn_types=('full',),
+                   reversible=True, batch_size=8, vae='dvae'),
+    'd': dict(dim=1024, depth=64, heads=16, attn_types=('axial_row', 'axial_col'),
+              reversible=True, batch_size=4, vae='vqgan16k'),
+}
+
+
+def parse_args():
+    p = argparse.ArgumentParser()
+    p.add_argument('--gpus', type=int, default=1)
+    p.add_argument('--steps', type=int, default=10)
+    p.add_argument('--warmup', type=int, default=3)
+    p.add_argument('--config', type=str, default='c', choices=sorted(CONFIGS))
+    p.add_argument('--batch_size', type=int, default=None, help='per-GPU batch')
+    p.add_argument('--mode', type=str, default='train', choices=['train', 'generate'])
+    p.add_argument('--gen_batch', type=int, default=64)
+    p.add_argument('--eager', action='store_true',
+                   help='force eager ops on GPU (baseline comparison)')
+    return p.parse_args()
+
+
+def maybe_self_launch(args):
+    """Convenience: `python bench.py --gpus N` outside torchrun re-execs
+    itself under torch.distributed.run."""
+    if args.gpus > 1 and 'RANK' not in os.environ:
+        import subprocess
+        cmd = [sys.executable, '-m', 'torch.distributed.run', '--nnodes=1',
+               f'--nproc-per-node={args.gpus}', '--master-addr', '127.0.0.1',
+               '--master-port', '29517', sys.argv[0], *sys.argv[1:]]
+        raise SystemExit(subprocess.call(cmd))
+
+
+def build_model(cfg, device):
+    from dalle_pytorch_amd import DALLE, DiscreteVAE, VQGanVAE
+    torch.manual_seed(1234)
+    if cfg['vae'] == 'dvae':
+        vae = DiscreteVAE(image_size=256, num_layers=3, num_tokens=8192,
+                          codebook_dim=512, hidden_dim=64)
+    else:
+        vae = VQGanVAE(num_tokens=16384)  # f=16 -> 256 image tokens... see note
+    dalle = DALLE(
+        dim=cfg['dim'], vae=vae, num_text_tokens=10000, text_seq_len=256,
+        depth=cfg['depth'], heads=cfg['heads'], dim_head=64,
+        attn_types=cfg['attn_types'], reversible=cfg['reversible'],
+        shift_tokens=True, rotary_emb=True)
+    return dalle.to(device)
+
+
+def main():
+    args = parse_args()
+    maybe_self_launch(args)
+    if args.eager:
+        os.environ['DALLE_AMD_ALLOW_EAGER'] = '1'
+        import dalle_pytorch_amd.ops.dispatch as _d
+        _d._HIP, _d._TRIED = None, True
+
+    from dalle_pytorch_amd.parallel import init_distributed, barrier
+    from dalle_pytorch_amd.parallel import DataParallelEngine
+
+    rank, world, local_rank = init_distributed()
+    use_cuda = torch.cuda.is_available()
+    device = torch.device(f'cuda:{local_rank}') if use_cuda else torch.device('cpu')
+
+    cfg = dict(CONFIGS[args.config])
+    if args.batch_size:
+        cfg['batch_size'] = args.batch_size
+    bsz = cfg['batch_size']
+
+    dalle = build_model(cfg, device)
+    image_seq_len = dalle.image_seq_len
+    seq_len = dalle.total_seq_len
+
+    engine = DataParallelEngine(dalle)
+    opt = torch.optim.Adam((p for p in dalle.parameters() if p.requires_grad), lr=3e-4)
+
+    # synthetic data pool: distinct per rank & step, generated once on device
+    torch.manual_seed(4321 + rank)
+    n_pool = 4
+    pool = [(torch.randint(1, 10000, (bsz, 256), device=device),
+             torch.rand(bsz, 3, 256, 256, device=device)) for _ in range(n_pool)]
+
+    autocast = torch.autocast(device_type='cuda', dtype=torch.bfloat16,
+                              enabled=use_cuda)
+
+    def train_step(i):
+        text, images = pool[i % n_pool]
+        with autocast:
+            loss = dalle(text, images, return_loss=True)
+        loss.backward()
+        engine.finish_gradient_sync()
+        torch.nn.utils.clip_grad_norm_(dalle.parameters(), 0.5)
+        opt.step()
+        engine.zero_grad()
+        return loss
+
+    def gen_step(_):
+        text = pool[0][0][:1].repeat(args.gen_batch, 1)
+        with autocast:
+            dalle.generate_images(text, use_cache=True, filter_thres=0.9)
+
+    step = train_step if args.mode == 'train' else gen_step
+    if args.mode == 'generate':
+        dalle.eval()
+
+    for i in range(args.warmup):
+        step(i)
+
+    barrier()
+    if use_cuda:
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for i in range(args.steps):
+        step(args.warmup + i)
+    if use_cuda:
+        torch.cuda.synchronize()
+    barrier()
+    elapsed = time.perf_counter() - t0
+
+    # MAX over ranks
+    t = torch.tensor([elapsed], device=device if use_cuda else 'cpu')
+    if world > 1:
+        import torch.distributed as dist
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+    elapsed = float(t.item())
+
+    if rank == 0:
+        ms_per_step = elapsed / args.steps * 1000
+        if args.mode == 'train':
+            global_batch = bsz * world
+            value = global_batch * image_seq_len * args.steps / elapsed
+            metric, unit = 'img-tokens/sec', 'img_tokens_per_sec'
+        else:
+            value = args.gen_batch * world * args.steps / elapsed
+            metric, unit = 'gen imgs/sec', 'images_per_sec'
+            global_batch = args.gen_batch * world
+        print(json.dumps({
+            'metric': metric, 'value': round(value, 2), 'unit': unit,
+            'n_gpus': world, 'steps': args.steps, 'warmup': args.warmup,
+            'ms_per_step': round(ms_per_step, 2),
+            'higher_is_better': True, 'scaling': 'weak',
+            'vs_baseline': None, 'dtype': 'bf16' if use_cuda else 'fp32',
+            'data': 'synthetic',
+            'config': {
+                'model': f"dalle-dim{cfg['dim']}-depth{cfg['depth']}-heads{cfg['heads']}",
+                'attn_types': list(cfg['attn_types']),
+                'reversible': cfg['reversible'],
+                'global_batch': global_batch, 'seq_len': seq_len,
+                'image_seq_len': image_seq_len,
+                'parallelism': f'dp{world}', 'mode': args.mode,
+                'eager': bool(args.eager)},
+        }))
+
+
+if __name__ == '__main__':
+    main()

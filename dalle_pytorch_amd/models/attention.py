@@ -310,17 +310,19 @@ class SparseAttention(Attention):
         pad = (self.block_size - rem) % self.block_size
         if mask is None:
             mask = torch.ones(b, n, dtype=torch.bool, device=x.device)
-        if pad:
-            x = F.pad(x, (0, 0, 0, pad))
-            mask = F.pad(mask, (0, pad), value=False)
 
         h = self.heads
         qkv = self.to_qkv(x).chunk(3, dim=-1)
         q, k, v = (_split_heads(t, h) for t in qkv)
         if rotary_pos_emb is not None:
             q, k, v = apply_rotary_to_qkv(rotary_pos_emb, (q, k, v))
+        if pad:
+            # rotary applied pre-pad (the table covers seq_len+1 positions);
+            # padded keys are masked out below, padded queries sliced at return
+            q, k, v = (F.pad(t, (0, 0, 0, pad)) for t in (q, k, v))
+            mask = F.pad(mask, (0, pad), value=False)
 
-        m = x.shape[1]
+        m = n + pad
         static = self.block_mask[:m, :m].to(x.device)
         out = attention_core(q, k, v, self.scale, causal=self.causal,
                              key_mask=mask, static_mask=static)

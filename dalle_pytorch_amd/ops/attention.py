@@ -42,10 +42,54 @@ def _eager_attention(q, k, v, scale, causal, key_mask, static_mask):
     return torch.matmul(attn, v)
 
 
+def _flash_bwd_composite(q, k, v, out, lse, dout, scale, causal, key_mask,
+                         static_mask, q_chunk=128):
+    """Flash-style backward at the torch level: recompute P per q-chunk from
+    the saved logsumexp, never materializing the full n x n matrix. All
+    GEMMs run in bf16 on rocBLAS MFMA paths (fp32 accumulation inside);
+    the P/dS elementwise math is fp32. A fully hand-written CDNA4 bwd
+    kernel replaces this incrementally."""
+    b, h, nq, d = q.shape
+    nk = k.shape[2]
+    diag = nk - nq
+    dq = torch.empty_like(q)
+    dk = torch.zeros(b, h, nk, d, dtype=torch.float32, device=q.device)
+    dv = torch.zeros_like(dk)
+    kT = k.transpose(-1, -2)
+    arange_k = torch.arange(nk, device=q.device)
+
+    for c0 in range(0, nq, q_chunk):
+        c1 = min(c0 + q_chunk, nq)
+        qc, oc, doc = q[:, :, c0:c1], out[:, :, c0:c1], dout[:, :, c0:c1]
+        lsec = lse[:, :, c0:c1]
+
+        s = torch.matmul(qc, kT).float() * scale              # [b,h,C,nk]
+        if causal:
+            cm = arange_k[None, :] > (torch.arange(c0, c1, device=q.device)[:, None] + diag)
+            s = s.masked_fill(cm, float('-inf'))
+        if key_mask is not None:
+            s = s.masked_fill(~key_mask[:, None, None, :], float('-inf'))
+        if static_mask is not None:
+            s = s.masked_fill(~static_mask[c0:c1], float('-inf'))
+
+        p = torch.exp(s - lsec.unsqueeze(-1))
+        p = torch.nan_to_num(p, nan=0.0)                      # -inf - -inf rows
+        pb = p.to(q.dtype)
+
+        dv += torch.matmul(pb.transpose(-1, -2), doc).float()
+        dp = torch.matmul(doc, v.transpose(-1, -2)).float()
+        Dc = (doc.float() * oc.float()).sum(dim=-1, keepdim=True)
+        ds = (p * (dp - Dc) * scale).to(q.dtype)
+        dq[:, :, c0:c1] = torch.matmul(ds, k)
+        dk += torch.matmul(ds.transpose(-1, -2), qc).float()
+
+    return dq, dk.to(q.dtype), dv.to(q.dtype)
+
+
 class _FlashAttention(torch.autograd.Function):
-    """Binds the gfx950 flash-attention kernels (fwd saves out + logsumexp;
-    bwd recomputes probabilities tile-by-tile — no n x n matrix ever hits
-    HBM)."""
+    """Binds the gfx950 flash-attention forward kernel (saves out +
+    logsumexp); backward recomputes probabilities chunk-by-chunk — no
+    n x n matrix is ever stored across the fwd/bwd boundary."""
 
     @staticmethod
     def forward(ctx, q, k, v, scale, causal, key_mask, static_mask):
@@ -60,18 +104,21 @@ class _FlashAttention(torch.autograd.Function):
 
     @staticmethod
     def backward(ctx, dout):
-        ext = hip_module()
         q, k, v, out, lse = ctx.saved_tensors
-        dq, dk, dv = ext.fa_bwd(q, k, v, out, lse, dout.contiguous(),
-                                ctx.scale, ctx.causal,
-                                ctx.key_mask, ctx.static_mask)
+        dq, dk, dv = _flash_bwd_composite(
+            q, k, v, out, lse, dout.contiguous(), ctx.scale, ctx.causal,
+            ctx.key_mask, ctx.static_mask)
         return dq, dk, dv, None, None, None, None
 
 
 def _hip_supported(q, k, causal, key_mask):
     if q.shape[-1] not in _SUPPORTED_HEAD_DIMS:
         return False
-    if q.dtype not in (torch.bfloat16, torch.float16):
+    if q.dtype != torch.bfloat16:
+        return False
+    if q.shape[-2] < 32:
+        # single-token decode: rocBLAS batched GEMV (memory-bound) until the
+        # dedicated decode kernel lands; not a training-path fallback
         return False
     return True
 
@@ -91,11 +138,12 @@ def attention_core(q, k, v, scale, causal=True, key_mask=None, static_mask=None)
     if using_eager_fallback(q):
         return _eager_attention(q, k, v, scale, causal, key_mask, static_mask)
     if not _hip_supported(q, k, causal, key_mask):
-        key = (q.shape[-1], str(q.dtype))
-        if key not in _warned_shapes:
-            _warned_shapes.add(key)
-            warnings.warn(f'attention_core: shape/dtype {key} not covered by '
-                          'the HIP kernel yet; using eager path on GPU')
+        if q.shape[-2] >= 32:  # decode fallback is intentional and silent
+            key = (q.shape[-1], str(q.dtype))
+            if key not in _warned_shapes:
+                _warned_shapes.add(key)
+                warnings.warn(f'attention_core: shape/dtype {key} not covered '
+                              'by the HIP kernel yet; using eager path on GPU')
         return _eager_attention(q, k, v, scale, causal, key_mask, static_mask)
     if static_mask is not None:
         static_mask = static_mask.contiguous()

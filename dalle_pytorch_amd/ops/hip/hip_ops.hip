@@ -1,0 +1,407 @@
+// dalle_pytorch_amd gfx950 (CDNA4) kernel library.
+//
+// First-party HIP kernels for the hot ops of the DALL-E stack
+// (SURVEY.md §2.5 K2-K7): flash-style causal attention forward with online
+// softmax (MFMA bf16 16x16x32, LDS-tiled K/V with transposed-V store and
+// padded rows for bank-conflict-free ds_read_b128), and fused GEGLU.
+// Written for wave64 / 8-XCD MI355X per the CDNA4 HIP guide —
+// NOT a port of any CUDA kernel.
+//
+// bf16 values are carried as raw `short` bit patterns end to end; float
+// math goes through explicit bit casts (bf2f/f2bf) so no accidental
+// numeric conversion happens on the storage path.
+//
+// Numerics contract (matches the eager oracle in ops/attention.py):
+//   S = scale * (Q K^T); masked entries -> -inf; P = softmax(S) with online
+//   max subtraction (identical to the reference's stable_softmax,
+//   attention.py:27-30); O = P V; lse = rowmax + log(rowsum).
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+#include <hip/hip_runtime.h>
+
+#define DEVFN __device__ __forceinline__
+
+using bf16x8 = __attribute__((ext_vector_type(8))) short;   // MFMA A/B frag
+using f32x4 = __attribute__((ext_vector_type(4))) float;    // MFMA C/D frag
+using int4v = __attribute__((ext_vector_type(4))) int;      // 16B copies
+
+constexpr float NEG_INF = -INFINITY;
+
+DEVFN float bf2f(short u) {
+  union { unsigned int i; float f; } c;
+  c.i = (unsigned int)(unsigned short)u << 16;
+  return c.f;
+}
+DEVFN short f2bf(float f) {
+  union { unsigned int i; float f; } c;
+  c.f = f;
+  unsigned int r = c.i + 0x7fff + ((c.i >> 16) & 1);  // round-nearest-even
+  return (short)(r >> 16);
+}
+
+// ---------------------------------------------------------------------------
+// Flash attention forward, head_dim = 64.
+//
+// Block: 256 threads = 4 waves; each block owns QBLK=64 query rows (16 per
+// wave); K/V streamed in KBLK=32 tiles through LDS.
+//
+// "Swapped" QK^T: S^T = mfma(K_tile, Q_tile), so the MFMA C layout
+// (col = lane&15, row = (lane>>4)*4 + reg) puts one query row per lane —
+// softmax reduces 8 in-lane values + a 4-lane shfl_xor, no serial-lane
+// section (guide §B attn / common-mistake 6). P round-trips through a
+// padded LDS tile to become the PV A-fragment; V is stored transposed at
+// stage time so the PV B-frag is one contiguous ds_read_b128.
+// ---------------------------------------------------------------------------
+
+constexpr int FA_D = 64;        // head dim
+constexpr int FA_QBLK = 64;     // q rows per block
+constexpr int FA_KBLK = 32;     // keys per LDS tile
+constexpr int FA_WAVES = 4;
+
+constexpr int KPAD = 72;        // K tile row stride (8-elem pad -> 2-way banks)
+constexpr int VPAD = 40;        // V^T tile row stride
+constexpr int PPAD = 40;        // P tile row stride
+
+DEVFN bf16x8 frag_from_lds(const short* base) {
+  return *reinterpret_cast<const bf16x8*>(base);  // ds_read_b128
+}
+
+__global__ __launch_bounds__(256, 2)
+void fa_fwd_d64_kernel(
+    const short* __restrict__ q,    // [bh, nq, 64] bf16 bits
+    const short* __restrict__ k,    // [bh, nk, 64]
+    const short* __restrict__ v,    // [bh, nk, 64]
+    short* __restrict__ out,        // [bh, nq, 64]
+    float* __restrict__ lse,        // [bh, nq]
+    const bool* __restrict__ key_mask,    // [b, nk] or null
+    const bool* __restrict__ static_mask, // [nq, nk] or null
+    int b, int h, int nq, int nk,
+    float scale, int causal) {
+
+  __shared__ short Kt[FA_KBLK][KPAD];
+  __shared__ short Vt[FA_D][VPAD];
+  __shared__ short Pl[FA_WAVES][16][PPAD];
+
+  const int bh = blockIdx.y;
+  const int batch = bh / h;
+  const int q0 = blockIdx.x * FA_QBLK;
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6;
+  const int lane = tid & 63;
+  const int lq = lane & 15;       // this lane's query row within the wave tile
+  const int grp = lane >> 4;      // lane group 0..3
+
+  const int qrow = q0 + wave * 16 + lq;   // this lane's global query row
+  const int diag = nk - nq;               // causal offset (ref triu(j-i+1))
+
+  const short* qp = q + (long)bh * nq * FA_D;
+  const short* kp = k + (long)bh * nk * FA_D;
+  const short* vp = v + (long)bh * nk * FA_D;
+
+  // Q fragments (B-operand of the swapped QK^T):
+  // lane holds Q[lq][8*grp + e + 32*c], c = 0,1
+  bf16x8 qfrag[2];
+  {
+    const bool qok = qrow < nq;
+    #pragma unroll
+    for (int c = 0; c < 2; ++c) {
+      qfrag[c] = qok
+          ? *reinterpret_cast<const bf16x8*>(qp + (long)qrow * FA_D + 8 * grp + 32 * c)
+          : bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
+    }
+  }
+
+  float m_run = NEG_INF;   // per-q-row online-softmax state (q = lq)
+  float l_run = 0.f;
+  // PV accumulators: acc[nt] covers d cols [16nt, 16nt+16); C layout rows
+  // are q = grp*4 + r of the wave tile
+  f32x4 acc[4] = {f32x4{0, 0, 0, 0}, f32x4{0, 0, 0, 0},
+                  f32x4{0, 0, 0, 0}, f32x4{0, 0, 0, 0}};
+
+  int ntiles = (nk + FA_KBLK - 1) / FA_KBLK;
+  if (causal) {
+    const int lim = min(nk - 1, q0 + FA_QBLK - 1 + diag);
+    ntiles = lim < 0 ? 0 : (lim / FA_KBLK + 1);
+  }
+
+  for (int kt = 0; kt < ntiles; ++kt) {
+    const int kbase = kt * FA_KBLK;
+
+    // ---- stage K (row-major, padded) and V (transposed) tiles
+    {
+      const int row = tid >> 3;       // 0..31
+      const int c8 = (tid & 7) * 8;   // 0,8,...,56
+      const int kg = kbase + row;
+      int4v kv{0, 0, 0, 0}, vv{0, 0, 0, 0};
+      if (kg < nk) {
+        kv = *reinterpret_cast<const int4v*>(kp + (long)kg * FA_D + c8);
+        vv = *reinterpret_cast<const int4v*>(vp + (long)kg * FA_D + c8);
+      }
+      __syncthreads();   // prior tile fully consumed before overwrite
+      *reinterpret_cast<int4v*>(&Kt[row][c8]) = kv;
+      const short* vs = reinterpret_cast<const short*>(&vv);
+      #pragma unroll
+      for (int e = 0; e < 8; ++e) Vt[c8 + e][row] = vs[e];
+    }
+    __syncthreads();
+
+    // ---- swapped QK^T: two 16-key subtiles, contraction K=64 in 2 steps
+    float s8[8];
+    #pragma unroll
+    for (int mt = 0; mt < 2; ++mt) {
+      f32x4 st{0, 0, 0, 0};
+      #pragma unroll
+      for (int c = 0; c < 2; ++c) {
+        bf16x8 af = frag_from_lds(&Kt[mt * 16 + lq][8 * grp + 32 * c]);
+        st = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, qfrag[c], st, 0, 0, 0);
+      }
+      #pragma unroll
+      for (int r = 0; r < 4; ++r) s8[mt * 4 + r] = st[r];
+    }
+    // After the swap: C col = lane&15 = q row; C row = grp*4 + r = key.
+    // s8[i] = S[qrow][kbase + (i>>2)*16 + grp*4 + (i&3)].
+
+    // ---- scale + masks
+    #pragma unroll
+    for (int i = 0; i < 8; ++i) {
+      const int kg = kbase + (i >> 2) * 16 + grp * 4 + (i & 3);
+      bool ok = (kg < nk) & (qrow < nq);
+      if (causal) ok &= kg <= qrow + diag;
+      if (key_mask != nullptr && ok) ok &= key_mask[(long)batch * nk + kg];
+      if (static_mask != nullptr && ok) ok &= static_mask[(long)qrow * nk + kg];
+      s8[i] = ok ? s8[i] * scale : NEG_INF;
+    }
+
+    // ---- online softmax update (lane-local + 4-lane reduce per q row)
+    float mt_part = NEG_INF;
+    #pragma unroll
+    for (int i = 0; i < 8; ++i) mt_part = fmaxf(mt_part, s8[i]);
+    mt_part = fmaxf(mt_part, __shfl_xor(mt_part, 16));
+    mt_part = fmaxf(mt_part, __shfl_xor(mt_part, 32));
+
+    const float m_new = fmaxf(m_run, mt_part);
+    float lsum = 0.f;
+    float p8[8];
+    #pragma unroll
+    for (int i = 0; i < 8; ++i) {
+      p8[i] = (s8[i] == NEG_INF) ? 0.f : __expf(s8[i] - m_new);
+      lsum += p8[i];
+    }
+    lsum += __shfl_xor(lsum, 16);
+    lsum += __shfl_xor(lsum, 32);
+
+    const float alpha = (m_run == NEG_INF) ? 0.f : __expf(m_run - m_new);
+    l_run = l_run * alpha + lsum;
+    if (m_new != NEG_INF) m_run = m_new;
+
+    // ---- P -> LDS (bf16) to reshape into the PV A-fragment
+    #pragma unroll
+    for (int i = 0; i < 8; ++i) {
+      const int kk = (i >> 2) * 16 + grp * 4 + (i & 3);
+      Pl[wave][lq][kk] = f2bf(p8[i]);
+    }
+
+    // ---- rescale accumulators; acc rows are q = grp*4 + r, alphas live on
+    // lanes whose (lane&15) equals that q row
+    #pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const float a_r = __shfl(alpha, grp * 4 + r);
+      #pragma unroll
+      for (int nt = 0; nt < 4; ++nt) acc[nt][r] *= a_r;
+    }
+
+    // ---- PV (compiler inserts the lgkmcnt wait for the same-wave Pl trip)
+    bf16x8 pf = frag_from_lds(&Pl[wave][lq][8 * grp]);
+    #pragma unroll
+    for (int nt = 0; nt < 4; ++nt) {
+      bf16x8 vf = frag_from_lds(&Vt[lq + 16 * nt][8 * grp]);
+      acc[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pf, vf, acc[nt], 0, 0, 0);
+    }
+  }
+
+  // ---- epilogue: divide by l per q row, store out + lse
+  #pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const int qr = q0 + wave * 16 + grp * 4 + r;
+    const float l_r = __shfl(l_run, grp * 4 + r);
+    const float inv = l_r > 0.f ? 1.f / l_r : 0.f;
+    if (qr < nq) {
+      #pragma unroll
+      for (int nt = 0; nt < 4; ++nt) {
+        out[((long)bh * nq + qr) * FA_D + 16 * nt + lq] = f2bf(acc[nt][r] * inv);
+      }
+    }
+  }
+  if (grp == 0 && qrow < nq) {
+    lse[(long)bh * nq + qrow] = (l_run > 0.f) ? m_run + __logf(l_run) : NEG_INF;
+  }
+}
+
+// ---------------------------------------------------------------------------
+// GEGLU forward/backward: x [N, 2H] -> out [N, H] = a * gelu(g)
+// (a = x[:, :H], g = x[:, H:]); exact erf gelu (transformer.py:106-109).
+// Memory-bound; 8-element strides for vectorizable bf16 access (guide G13).
+// ---------------------------------------------------------------------------
+
+DEVFN float gelu_f(float x) { return 0.5f * x * (1.f + erff(x * 0.70710678f)); }
+DEVFN float gelu_grad_f(float x) {
+  const float cdf = 0.5f * (1.f + erff(x * 0.70710678f));
+  const float pdf = 0.3989422804f * __expf(-0.5f * x * x);
+  return cdf + x * pdf;
+}
+
+template <typename T>
+__global__ void geglu_fwd_kernel(const T* __restrict__ x, T* __restrict__ out,
+                                 long rows, int H) {
+  const long row = blockIdx.y;
+  const int i0 = (blockIdx.x * blockDim.x + threadIdx.x) * 8;
+  if (row >= rows || i0 >= H) return;
+  const T* a = x + row * (2L * H) + i0;
+  const T* g = a + H;
+  T* o = out + row * (long)H + i0;
+  #pragma unroll
+  for (int e = 0; e < 8; ++e)
+    if (i0 + e < H) o[e] = T(float(a[e]) * gelu_f(float(g[e])));
+}
+
+template <typename T>
+__global__ void geglu_bwd_kernel(const T* __restrict__ x,
+                                 const T* __restrict__ dout,
+                                 T* __restrict__ dx, long rows, int H) {
+  const long row = blockIdx.y;
+  const int i0 = (blockIdx.x * blockDim.x + threadIdx.x) * 8;
+  if (row >= rows || i0 >= H) return;
+  const T* a = x + row * (2L * H) + i0;
+  const T* g = a + H;
+  const T* dO = dout + row * (long)H + i0;
+  T* da = dx + row * (2L * H) + i0;
+  T* dg = da + H;
+  #pragma unroll
+  for (int e = 0; e < 8; ++e) {
+    if (i0 + e < H) {
+      const float av = float(a[e]), gv = float(g[e]), dv = float(dO[e]);
+      da[e] = T(dv * gelu_f(gv));
+      dg[e] = T(dv * av * gelu_grad_f(gv));
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// MFMA layout probe (test support): one 16x16x32 bf16 MFMA with the exact
+// fragment mappings the attention kernel assumes. The GPU test compares
+// C against torch.matmul on asymmetric inputs (guide G9) so a wrong operand
+// layout is caught in isolation.
+// ---------------------------------------------------------------------------
+
+__global__ void mfma_probe_kernel(const short* __restrict__ A,   // [16,32]
+                                  const short* __restrict__ B,   // [32,16]
+                                  float* __restrict__ C) {       // [16,16]
+  const int lane = threadIdx.x & 63;
+  const int lq = lane & 15, grp = lane >> 4;
+  bf16x8 af = *reinterpret_cast<const bf16x8*>(A + lq * 32 + 8 * grp);
+  bf16x8 bf;
+  #pragma unroll
+  for (int e = 0; e < 8; ++e) bf[e] = B[(8 * grp + e) * 16 + lq];
+  f32x4 c{0, 0, 0, 0};
+  c = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, bf, c, 0, 0, 0);
+  #pragma unroll
+  for (int r = 0; r < 4; ++r) C[(grp * 4 + r) * 16 + lq] = c[r];
+}
+
+// ===========================================================================
+// Bindings
+// ===========================================================================
+
+#define CHK(x) TORCH_CHECK(x, #x)
+
+static hipStream_t cur_stream() {
+  return at::hip::getCurrentHIPStream().stream();
+}
+
+std::vector<torch::Tensor> fa_fwd(torch::Tensor q, torch::Tensor k, torch::Tensor v,
+                                  double scale, bool causal,
+                                  std::optional<torch::Tensor> key_mask,
+                                  std::optional<torch::Tensor> static_mask) {
+  CHK(q.is_cuda() && k.is_cuda() && v.is_cuda());
+  CHK(q.dtype() == torch::kBFloat16);
+  CHK(q.size(-1) == FA_D);
+  CHK(q.is_contiguous() && k.is_contiguous() && v.is_contiguous());
+  const int b = q.size(0), h = q.size(1), nq = q.size(2), nk = k.size(2);
+
+  auto out = torch::empty_like(q);
+  auto lse = torch::empty({b, h, nq}, q.options().dtype(torch::kFloat32));
+
+  const bool* km = nullptr;
+  const bool* sm = nullptr;
+  if (key_mask.has_value()) {
+    CHK(key_mask->dtype() == torch::kBool && key_mask->is_contiguous());
+    CHK(key_mask->size(0) == b && key_mask->size(1) == nk);
+    km = key_mask->data_ptr<bool>();
+  }
+  if (static_mask.has_value()) {
+    CHK(static_mask->dtype() == torch::kBool && static_mask->is_contiguous());
+    CHK(static_mask->size(0) == nq && static_mask->size(1) == nk);
+    sm = static_mask->data_ptr<bool>();
+  }
+
+  dim3 grid((nq + FA_QBLK - 1) / FA_QBLK, b * h);
+  hipLaunchKernelGGL(fa_fwd_d64_kernel, grid, dim3(256), 0, cur_stream(),
+                     reinterpret_cast<const short*>(q.data_ptr()),
+                     reinterpret_cast<const short*>(k.data_ptr()),
+                     reinterpret_cast<const short*>(v.data_ptr()),
+                     reinterpret_cast<short*>(out.data_ptr()),
+                     lse.data_ptr<float>(), km, sm,
+                     b, h, nq, nk, (float)scale, causal ? 1 : 0);
+  return {out, lse};
+}
+
+torch::Tensor geglu_fwd(torch::Tensor x) {
+  CHK(x.is_cuda() && x.is_contiguous());
+  const int H = (int)x.size(-1) / 2;
+  const long rows = x.numel() / (2L * H);
+  auto sizes = x.sizes().vec();
+  sizes.back() = H;
+  auto out = torch::empty(sizes, x.options());
+  dim3 grid((H + 256 * 8 - 1) / (256 * 8), rows);
+  AT_DISPATCH_FLOATING_TYPES_AND2(at::ScalarType::BFloat16, at::ScalarType::Half,
+                                  x.scalar_type(), "geglu_fwd", [&] {
+    hipLaunchKernelGGL(geglu_fwd_kernel<scalar_t>, grid, dim3(256), 0, cur_stream(),
+                       x.data_ptr<scalar_t>(), out.data_ptr<scalar_t>(), rows, H);
+  });
+  return out;
+}
+
+torch::Tensor geglu_bwd(torch::Tensor x, torch::Tensor dout) {
+  CHK(x.is_cuda() && x.is_contiguous() && dout.is_contiguous());
+  const int H = (int)x.size(-1) / 2;
+  const long rows = x.numel() / (2L * H);
+  auto dx = torch::empty_like(x);
+  dim3 grid((H + 256 * 8 - 1) / (256 * 8), rows);
+  AT_DISPATCH_FLOATING_TYPES_AND2(at::ScalarType::BFloat16, at::ScalarType::Half,
+                                  x.scalar_type(), "geglu_bwd", [&] {
+    hipLaunchKernelGGL(geglu_bwd_kernel<scalar_t>, grid, dim3(256), 0, cur_stream(),
+                       x.data_ptr<scalar_t>(), dout.data_ptr<scalar_t>(),
+                       dx.data_ptr<scalar_t>(), rows, H);
+  });
+  return dx;
+}
+
+torch::Tensor mfma_probe(torch::Tensor A, torch::Tensor B) {
+  CHK(A.is_cuda() && A.dtype() == torch::kBFloat16);
+  auto Ac = A.contiguous();
+  auto Bc = B.contiguous();
+  auto C = torch::zeros({16, 16}, A.options().dtype(torch::kFloat32));
+  hipLaunchKernelGGL(mfma_probe_kernel, dim3(1), dim3(64), 0, cur_stream(),
+                     reinterpret_cast<const short*>(Ac.data_ptr()),
+                     reinterpret_cast<const short*>(Bc.data_ptr()),
+                     C.data_ptr<float>());
+  return C;
+}
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("fa_fwd", &fa_fwd, "flash attention forward (gfx950, d=64)");
+  m.def("geglu_fwd", &geglu_fwd, "fused GEGLU forward");
+  m.def("geglu_bwd", &geglu_bwd, "fused GEGLU backward");
+  m.def("mfma_probe", &mfma_probe, "MFMA 16x16x32 bf16 layout probe");
+}

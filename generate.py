@@ -1,0 +1,102 @@
+#!/usr/bin/env python3
+"""Generate images from a trained DALL-E checkpoint (reference generate.py
+parity): load checkpoint -> rebuild VAE+DALLE -> autoregressive sampling
+(cached decode by default on GPU) -> save PNGs; optional text completion
+first (--gentxt) and CLIP re-ranking.
+"""
+
+import argparse
+from pathlib import Path
+
+import torch
+
+from dalle_pytorch_amd.utils.checkpoint import (
+    load_dalle_checkpoint, build_dalle_from_checkpoint)
+from dalle_pytorch_amd.utils import tokenizer as tokenizer_mod
+
+
+def parse_args(argv=None):
+    p = argparse.ArgumentParser(description='DALL-E image generation (MI355X-native)')
+    p.add_argument('--dalle_path', type=str, required=True)
+    p.add_argument('--vqgan_model_path', type=str, default=None)
+    p.add_argument('--vqgan_config_path', type=str, default=None)
+    p.add_argument('--text', type=str, required=True,
+                   help='prompt(s); separate multiple with |')
+    p.add_argument('--num_images', type=int, default=128)
+    p.add_argument('--batch_size', type=int, default=4)
+    p.add_argument('--top_k', type=float, default=0.9)
+    p.add_argument('--temperature', type=float, default=1.)
+    p.add_argument('--outputs_dir', type=str, default='./outputs')
+    p.add_argument('--bpe_path', type=str, default=None)
+    p.add_argument('--hug', action='store_true')
+    p.add_argument('--chinese', action='store_true')
+    p.add_argument('--gentxt', action='store_true',
+                   help='complete the prompt with the model before generating')
+    p.add_argument('--no_cache', action='store_true')
+    p.add_argument('--cond_scale', type=float, default=1.0)
+    return p.parse_args(argv)
+
+
+def get_tokenizer(args):
+    if args.chinese:
+        return tokenizer_mod.ChineseTokenizer()
+    if args.hug:
+        return tokenizer_mod.HugTokenizer(args.bpe_path)
+    if args.bpe_path is not None:
+        suffix = Path(args.bpe_path).suffix
+        if suffix == '.json':
+            return tokenizer_mod.HugTokenizer(args.bpe_path)
+        if suffix == '.model':
+            return tokenizer_mod.YttmTokenizer(args.bpe_path)
+        return tokenizer_mod.SimpleTokenizer(args.bpe_path)
+    return tokenizer_mod.tokenizer
+
+
+def main(argv=None):
+    args = parse_args(argv)
+    device = torch.device('cuda:0') if torch.cuda.is_available() else torch.device('cpu')
+    tok = get_tokenizer(args)
+
+    ckpt = load_dalle_checkpoint(args.dalle_path)
+    vae = None
+    if ckpt.get('vae_class_name') == 'VQGanVAE' and args.vqgan_model_path:
+        from dalle_pytorch_amd import VQGanVAE
+        vae = VQGanVAE(args.vqgan_model_path, args.vqgan_config_path)
+    dalle, vae = build_dalle_from_checkpoint(ckpt, vae=vae)
+    dalle = dalle.to(device).eval()
+
+    texts = args.text.split('|')
+    out_root = Path(args.outputs_dir)
+
+    for raw_text in texts:
+        if args.gentxt:
+            text_tokens, gen_texts = dalle.generate_texts(tok, text=raw_text)
+            raw_text = gen_texts[0]
+            print(f'completed text: {raw_text}')
+            text_tokens = text_tokens.repeat(args.num_images, 1)
+        else:
+            text_tokens = tok.tokenize([raw_text], dalle.text_seq_len,
+                                       truncate_text=True).to(device)
+            text_tokens = text_tokens.repeat(args.num_images, 1)
+
+        images = []
+        for i in range(0, text_tokens.shape[0], args.batch_size):
+            chunk = text_tokens[i:i + args.batch_size]
+            images.append(dalle.generate_images(
+                chunk, filter_thres=args.top_k, temperature=args.temperature,
+                use_cache=not args.no_cache, cond_scale=args.cond_scale))
+        images = torch.cat(images, dim=0)
+
+        subdir = out_root / raw_text.replace(' ', '_')[:100]
+        subdir.mkdir(parents=True, exist_ok=True)
+        try:
+            from torchvision.utils import save_image
+            for j, img in enumerate(images):
+                save_image(img, subdir / f'{j}.png', normalize=False)
+        except ImportError:
+            torch.save(images, subdir / 'images.pt')
+        print(f'created {images.shape[0]} images at "{subdir}"')
+
+
+if __name__ == '__main__':
+    main()

@@ -1,0 +1,176 @@
+"""Attention variants vs independent oracles (reference semantics from
+SURVEY.md §2.5 K2-K4/K9-K11)."""
+
+import math
+
+import pytest
+import torch
+
+from dalle_pytorch_amd.models.attention import (
+    Attention, SparseAxialCausalAttention, SparseConvCausalAttention,
+    SparseAttention)
+from dalle_pytorch_amd.models.positional import (
+    build_dalle_rotary_table, apply_rotary, rotary_freqs, rotary_angles)
+from dalle_pytorch_amd.ops.attention import attention_core
+
+torch.manual_seed(0)
+
+
+def dense_oracle(q, k, v, scale, causal=True, key_mask=None, static_mask=None):
+    dots = (q * scale) @ k.transpose(-1, -2)
+    big = -torch.finfo(dots.dtype).max
+    if key_mask is not None:
+        dots = dots.masked_fill(~key_mask[:, None, None, :], big)
+    if causal:
+        i, j = dots.shape[-2:]
+        dots = dots.masked_fill(torch.ones(i, j, dtype=torch.bool).triu_(j - i + 1), big)
+    if static_mask is not None:
+        dots = dots.masked_fill(~static_mask, big)
+    return dots.softmax(-1) @ v
+
+
+def test_attention_core_matches_oracle():
+    q, k, v = torch.randn(3, 2, 4, 10, 16), torch.randn(2, 4, 10, 16), None
+    q = torch.randn(2, 4, 10, 16)
+    v = torch.randn(2, 4, 10, 16)
+    out = attention_core(q, k, v, 0.25, causal=True)
+    assert torch.allclose(out, dense_oracle(q, k, v, 0.25), atol=1e-6)
+
+
+def test_attention_core_masks():
+    q = torch.randn(2, 2, 8, 16)
+    k = torch.randn(2, 2, 8, 16)
+    v = torch.randn(2, 2, 8, 16)
+    km = torch.rand(2, 8) > 0.3
+    km[:, 0] = True
+    sm = torch.rand(8, 8) > 0.3
+    sm.fill_diagonal_(True)
+    out = attention_core(q, k, v, 0.25, causal=True, key_mask=km, static_mask=sm)
+    ref = dense_oracle(q, k, v, 0.25, causal=True, key_mask=km, static_mask=sm)
+    assert torch.allclose(out, ref, atol=1e-6)
+
+
+def test_cached_attention_matches_full():
+    attn = Attention(dim=32, seq_len=12, heads=2, dim_head=16).eval()
+    x = torch.randn(1, 8, 32)
+    full = attn(x)
+    cache = {}
+    outs = []
+    for i in range(8):
+        cache['offset'] = i
+        outs.append(attn(x[:, i:i + 1], cache=cache, cache_key='a'))
+    step = torch.cat(outs, dim=1)
+    assert torch.allclose(full, step, atol=1e-5)
+
+
+def test_rotary_table_shape_and_values():
+    # dim_head=64: 60 rotated channels
+    t = build_dalle_rotary_table(64, 257, 32)
+    assert t.shape == (1, 257 + 1024, 60)
+    # text 1-D branch at position p rotates with lang freqs
+    lang = rotary_freqs(64 // 3, 'lang')
+    ang5 = rotary_angles(torch.tensor([5.0]), lang)[0]
+    assert torch.allclose(t[0, 5, :20], ang5)
+    # image rows all share the pinned 8192 position in the 1-D branch
+    ang_far = rotary_angles(torch.tensor([8192.0]), lang)[0]
+    assert torch.allclose(t[0, 300, :20], ang_far)
+
+
+def test_apply_rotary_rotates_pairs():
+    ang = torch.tensor([[math.pi / 2, math.pi / 2]])  # one pair, 90 degrees
+    x = torch.tensor([[1.0, 2.0, 7.0]])
+    out = apply_rotary(ang, x)
+    # (x0,x1) -> (x0 c - x1 s, x1 c + x0 s) = (-2, 1); x2 passes through
+    assert torch.allclose(out, torch.tensor([[-2.0, 1.0, 7.0]]), atol=1e-6)
+
+
+def test_axial_attention_matches_static_mask_dense():
+    """SparseAxialCausalAttention == dense attention under the axial static
+    mask (the reference's optimize_for_inference equivalence,
+    transformer.py:252-260)."""
+    torch.manual_seed(1)
+    S, text_len = 4, 3
+    seq_len = text_len + S * S - 1   # 18
+    for axis in (0, 1):
+        sparse = SparseAxialCausalAttention(
+            dim=32, seq_len=seq_len, image_size=S, axis=axis, heads=2,
+            dim_head=16)
+        # dense mask replicating transformer._get_attention_mask
+        m = torch.zeros(seq_len, seq_len, dtype=torch.bool)
+        m[:, :text_len] = True
+        if axis == 0:
+            for row in range(S):
+                lo = text_len + row * S
+                m[lo:lo + S, lo:lo + S] = True
+        else:
+            for col in range(S):
+                lo = text_len + col
+                m[lo::S, lo::S] = True
+        dense = Attention(dim=32, seq_len=seq_len, heads=2, dim_head=16,
+                          causal=True, static_mask=m)
+        dense.to_qkv.weight.data = sparse.to_qkv.weight.data.clone()
+        dense.to_out[0].weight.data = sparse.to_out[0].weight.data.clone()
+        dense.to_out[0].bias.data = sparse.to_out[0].bias.data.clone()
+
+        x = torch.randn(2, seq_len, 32)
+        assert torch.allclose(sparse(x), dense(x), atol=1e-5), f'axis={axis}'
+
+
+def test_conv_attention_matches_dense_mask():
+    """conv_like == dense attention under the unfolded-neighborhood mask."""
+    torch.manual_seed(2)
+    S, text_len, ks = 4, 3, 3
+    seq_len = text_len + S * S - 1
+    sparse = SparseConvCausalAttention(dim=32, seq_len=seq_len, image_size=S,
+                                       kernel_size=ks, heads=2, dim_head=16)
+    # build the dense mask: text causal; image attends text fully + its
+    # causally-padded ks x ks neighborhood
+    n = seq_len + 1
+    m = torch.zeros(n, n, dtype=torch.bool)
+    tri = torch.ones(text_len, text_len, dtype=torch.bool).tril_()
+    m[:text_len, :text_len] = tri
+    m[text_len:, :text_len] = True
+    half = ks // 2
+    for qi in range(S * S):
+        qr, qc = divmod(qi, S)
+        for di in range(ks):
+            for dj in range(ks):
+                kr, kc = qr - 2 * half + di, qc - 2 * half + dj
+                if 0 <= kr < S and 0 <= kc < S:
+                    m[text_len + qi, text_len + kr * S + kc] = True
+
+    x = torch.randn(2, seq_len, 32)
+    h, dh, scale = 2, 16, 16 ** -0.5
+    xp = torch.nn.functional.pad(x, (0, 0, 0, 1))
+    qkv = sparse.to_qkv(xp).chunk(3, -1)
+    q, k, v = (t.reshape(2, n, h, dh).permute(0, 2, 1, 3) for t in qkv)
+    ref = dense_oracle(q, k, v, scale, causal=False, static_mask=m)
+    ref = ref.permute(0, 2, 1, 3).reshape(2, n, h * dh)
+    ref = sparse.to_out(ref)[:, :seq_len]
+    assert torch.allclose(sparse(x), ref, atol=1e-5)
+
+
+def test_block_sparse_layout_properties():
+    attn = SparseAttention(32, 64, heads=2, dim_head=16, block_size=8,
+                           text_seq_len=16, num_random_blocks=1)
+    x = torch.randn(1, 64, 32)
+    out = attn(x)
+    assert out.shape == (1, 64, 32)
+    bm = attn.block_mask
+    nb = 64 // 8
+    blocks = bm[:64, :64].reshape(nb, 8, nb, 8).any(dim=(1, 3))
+    assert blocks.diagonal().all()                 # diagonal present
+    assert not blocks.triu(1).any()                # strictly causal
+    for r in range(nb):                            # global text blocks (causal-clipped)
+        assert blocks[r, :min(r + 1, 2)].all()
+
+
+def test_rotary_applied_to_v_quirk():
+    """The reference rotates v as well as q/k (attention.py:35,67); outputs
+    must differ from a no-rotary run even when q/k rotation cancels."""
+    attn = Attention(dim=32, seq_len=8, heads=2, dim_head=16).eval()
+    x = torch.randn(1, 8, 32)
+    table = build_dalle_rotary_table(16, 4, 2)[..., :8, :]
+    out_rot = attn(x, rotary_pos_emb=table)
+    out_plain = attn(x)
+    assert not torch.allclose(out_rot, out_plain, atol=1e-4)

@@ -1,0 +1,165 @@
+"""DALLE model: forward/loss semantics, generation parity, key schema."""
+
+import pytest
+import torch
+
+from dalle_pytorch_amd import DALLE, DiscreteVAE
+
+torch.manual_seed(0)
+
+
+def tiny_vae():
+    return DiscreteVAE(image_size=64, num_layers=3, num_tokens=64,
+                       codebook_dim=32, hidden_dim=8)
+
+
+def tiny_dalle(**kw):
+    args = dict(dim=32, num_text_tokens=50, text_seq_len=8, depth=2, heads=2,
+                dim_head=16, attn_types=('full',), shift_tokens=True)
+    args.update(kw)
+    return DALLE(vae=tiny_vae(), **args)
+
+
+@pytest.mark.parametrize('attn_types,reversible', [
+    (('full',), False),
+    (('axial_row', 'axial_col'), False),
+    (('conv_like',), False),
+    (('sparse',), False),
+    (('full', 'axial_row'), True),
+])
+def test_forward_backward_all_attention_types(attn_types, reversible):
+    d = tiny_dalle(attn_types=attn_types, reversible=reversible, stable=True)
+    text = torch.randint(1, 50, (2, 8))
+    imgs = torch.rand(2, 3, 64, 64)
+    loss = d(text, imgs, return_loss=True)
+    loss.backward()
+    assert torch.isfinite(loss)
+    grads = [p.grad for p in d.parameters() if p.requires_grad and p.grad is not None]
+    assert len(grads) > 0 and all(torch.isfinite(g).all() for g in grads)
+
+
+def test_vae_frozen():
+    d = tiny_dalle()
+    assert all(not p.requires_grad for p in d.vae.parameters())
+
+
+def test_loss_weighting_formula():
+    """loss = (CE_text + w * CE_img) / (w + 1) (reference :667-670)."""
+    d = tiny_dalle().eval()
+    text = torch.randint(1, 50, (2, 8))
+    imgs = torch.rand(2, 3, 64, 64)
+    torch.manual_seed(3)
+    image_codes = d.vae.get_codebook_indices(imgs)
+
+    loss = d(text, image_codes, return_loss=True)
+
+    logits = d(text, image_codes)
+    text_in = torch.where(text == 0,
+                          torch.arange(8) + (d.num_text_tokens - 8), text)
+    labels = torch.cat((text_in, image_codes + d.num_text_tokens), 1)
+    lg = logits.transpose(1, 2)
+    lt = torch.nn.functional.cross_entropy(lg[:, :, :8], labels[:, :8])
+    li = torch.nn.functional.cross_entropy(lg[:, :, 8:], labels[:, 8:])
+    expect = (lt + 7 * li) / 8
+    assert torch.allclose(loss, expect, atol=1e-5)
+
+
+def test_logits_mask_positions():
+    d = tiny_dalle().eval()
+    text = torch.randint(1, 50, (1, 8))
+    logits = d(text, None)
+    # text positions predict text vocab only, big-neg elsewhere
+    ntt = d.num_text_tokens
+    assert (logits[0, :8, ntt:] < -1e30).all()
+    # last position (first image position) predicts image vocab only
+    assert (logits[0, 8, :ntt] < -1e30).all()
+    assert (logits[0, 8, ntt:] > -1e30).all()
+
+
+def test_unique_padding_tokens():
+    """text==0 replaced by per-position padding ids (reference :595-596)."""
+    d = tiny_dalle().eval()
+    t1 = torch.zeros(1, 8, dtype=torch.long)
+    t2 = torch.zeros(1, 8, dtype=torch.long)
+    t2[0, 3] = 7
+    l1, l2 = d(t1, None), d(t2, None)
+    assert not torch.allclose(l1, l2)
+
+
+@pytest.mark.parametrize('attn_types', [('full',), ('axial_row', 'axial_col')])
+def test_generate_cached_equals_uncached(attn_types):
+    """SURVEY.md behavioral bar (c): cached generation allclose to uncached
+    at near-zero temperature."""
+    torch.manual_seed(5)
+    d = tiny_dalle(attn_types=attn_types, depth=2, stable=True).eval()
+    text = torch.randint(1, 50, (1, 8))
+    torch.manual_seed(7)
+    a = d.generate_images(text, use_cache=True, temperature=1e-8, filter_thres=0.99)
+    torch.manual_seed(7)
+    b = d.generate_images(text, use_cache=False, temperature=1e-8, filter_thres=0.99)
+    assert torch.allclose(a, b, atol=1e-5)
+
+
+def test_generate_with_cond_scale_and_priming():
+    d = tiny_dalle().eval()
+    text = torch.randint(1, 50, (1, 8))
+    img = torch.rand(1, 3, 64, 64)
+    out = d.generate_images(text, img=img, cond_scale=2.0, use_cache=True)
+    assert out.shape == (1, 3, 64, 64)
+
+
+def test_generate_texts():
+    from dalle_pytorch_amd.utils.tokenizer import SimpleTokenizer
+
+    class _Wrap:
+        def __init__(self):
+            self.tokenizer = SimpleTokenizer()
+
+    d = tiny_dalle(num_text_tokens=SimpleTokenizer().vocab_size).eval()
+    tokens, texts = d.generate_texts(_Wrap(), text='a red')
+    assert tokens.shape[1] == d.text_seq_len
+    assert isinstance(texts[0], str)
+
+
+def test_share_input_output_emb():
+    d = tiny_dalle(share_input_output_emb=True)
+    sd = d.state_dict()
+    assert 'text_emb.weight' not in sd and 'image_emb.weight' not in sd
+    text = torch.randint(1, 50, (1, 8))
+    imgs = torch.rand(1, 3, 64, 64)
+    d(text, imgs, return_loss=True).backward()
+
+
+def test_state_dict_schema():
+    """SURVEY.md §2.6 verified key layout."""
+    d = tiny_dalle()
+    sd = d.state_dict()
+    for key in [
+        'text_emb.weight', 'image_emb.weight', 'to_logits.0.weight',
+        'to_logits.1.weight', 'transformer.pos_emb',
+        'transformer.layers.layers.0.0.scale',
+        'transformer.layers.layers.0.0.fn.norm.weight',
+        'transformer.layers.layers.0.0.fn.fn.fn.fn.fn.to_qkv.weight',
+        'transformer.layers.layers.0.1.fn.fn.fn.fn.net.0.weight',
+        'transformer.layers.layers.0.1.fn.fn.fn.fn.net.3.bias',
+        'vae.codebook.weight',
+    ]:
+        assert key in sd, key
+    assert 'logits_mask' not in sd
+
+    dr = tiny_dalle(reversible=True)
+    sdr = dr.state_dict()
+    assert 'transformer.layers.blocks.0.f.net.scale' in sdr
+    assert 'transformer.layers.blocks.0.g.net.fn.fn.fn.fn.net.0.weight' in sdr
+
+    dn = tiny_dalle(rotary_emb=False)
+    sdn = dn.state_dict()
+    assert 'text_pos_emb.weight' in sdn
+    assert 'image_pos_emb.weights.0' in sdn
+    assert 'transformer.pos_emb' not in sdn
+
+
+def test_text_length_assertion():
+    d = tiny_dalle()
+    with pytest.raises(AssertionError):
+        d(torch.randint(1, 50, (1, 5)), None)

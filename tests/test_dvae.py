@@ -1,0 +1,99 @@
+"""DiscreteVAE: shapes, loss quirks, straight-through/reinmax."""
+
+import pytest
+import torch
+import torch.nn.functional as F
+
+from dalle_pytorch_amd import DiscreteVAE
+
+torch.manual_seed(0)
+
+
+def test_shapes_roundtrip():
+    vae = DiscreteVAE(image_size=64, num_layers=3, num_tokens=128,
+                      codebook_dim=32, hidden_dim=8)
+    img = torch.rand(2, 3, 64, 64)
+    logits = vae(img, return_logits=True)
+    assert logits.shape == (2, 128, 8, 8)
+    codes = vae.get_codebook_indices(img)
+    assert codes.shape == (2, 64)
+    recon = vae.decode(codes)
+    assert recon.shape == (2, 3, 64, 64)
+    out = vae(img)
+    assert out.shape == img.shape
+
+
+def test_resblocks_and_1x1_decoder_head():
+    vae = DiscreteVAE(image_size=32, num_layers=2, num_tokens=32,
+                      codebook_dim=16, hidden_dim=8, num_resnet_blocks=2)
+    img = torch.rand(1, 3, 32, 32)
+    assert vae(img).shape == img.shape
+    # resblock path inserts the codebook->hidden 1x1 conv at decoder[0]
+    assert isinstance(vae.decoder[0], torch.nn.Conv2d)
+    assert vae.decoder[0].kernel_size == (1, 1)
+
+
+def test_kl_term_reversed_args_quirk():
+    """Reference computes F.kl_div(log_uniform, log_qy, log_target=True)
+    with uniform as *input* (dalle_pytorch.py:256-261). Verify our loss
+    reproduces that exact expression."""
+    vae = DiscreteVAE(image_size=32, num_layers=2, num_tokens=16,
+                      codebook_dim=8, hidden_dim=4, kl_div_loss_weight=1.0)
+    img = torch.rand(2, 3, 32, 32)
+    torch.manual_seed(11)
+    loss = vae(img, return_loss=True)
+
+    torch.manual_seed(11)
+    normed = vae.norm(img)
+    logits = vae.encoder(normed)
+    one_hot = F.gumbel_softmax(logits, tau=vae.temperature, dim=1, hard=False)
+    sampled = torch.einsum('bnhw,nd->bdhw', one_hot, vae.codebook.weight)
+    out = vae.decoder(sampled)
+    recon = F.mse_loss(normed, out)
+    lg = logits.permute(0, 2, 3, 1).reshape(2, -1, 16)
+    log_qy = F.log_softmax(lg, dim=-1)
+    log_u = torch.log(torch.tensor([1. / 16]))
+    kl = F.kl_div(log_u, log_qy, None, None, 'batchmean', log_target=True)
+    assert torch.allclose(loss, recon + kl, atol=1e-5)
+
+
+def test_normalization_applied():
+    vae = DiscreteVAE(image_size=32, num_layers=2, num_tokens=16,
+                      codebook_dim=8, hidden_dim=4)
+    img = torch.rand(1, 3, 32, 32)
+    n = vae.norm(img)
+    assert torch.allclose(n, (img - 0.5) / 0.5)
+
+
+@pytest.mark.parametrize('st,rm', [(True, False), (True, True)])
+def test_straight_through_and_reinmax_backward(st, rm):
+    vae = DiscreteVAE(image_size=32, num_layers=2, num_tokens=16,
+                      codebook_dim=8, hidden_dim=4, straight_through=st,
+                      reinmax=rm)
+    img = torch.rand(2, 3, 32, 32)
+    loss = vae(img, return_loss=True)
+    loss.backward()
+    assert torch.isfinite(loss)
+    assert vae.codebook.weight.grad is not None
+
+
+def test_smooth_l1_option():
+    vae = DiscreteVAE(image_size=32, num_layers=1, num_tokens=8,
+                      codebook_dim=4, hidden_dim=4, smooth_l1_loss=True)
+    assert vae.loss_fn is F.smooth_l1_loss
+
+
+def test_temperature_argument():
+    vae = DiscreteVAE(image_size=32, num_layers=1, num_tokens=8,
+                      codebook_dim=4, hidden_dim=4)
+    img = torch.rand(1, 3, 32, 32)
+    torch.manual_seed(0)
+    a = vae(img, temp=0.1)
+    torch.manual_seed(0)
+    b = vae(img, temp=5.0)
+    assert not torch.allclose(a, b)
+
+
+def test_image_size_power_of_two_assert():
+    with pytest.raises(AssertionError):
+        DiscreteVAE(image_size=48)
