@@ -46,11 +46,15 @@ def apply_rotary(angles: torch.Tensor, t: torch.Tensor) -> torch.Tensor:
     """Rotate the first ``angles.shape[-1]`` channels of ``t``; pass the rest.
 
     ``angles`` broadcasts against ``t[..., :rot]``. Matches
-    rotary_embedding_torch.apply_rotary_emb with start_index=0.
+    rotary_embedding_torch.apply_rotary_emb with start_index=0. cos/sin are
+    cast to ``t``'s dtype so a bf16 autocast stream stays bf16 (keeping the
+    fused bf16 attention kernel on the hot path).
     """
     rot = angles.shape[-1]
     head, tail = t[..., :rot], t[..., rot:]
-    head = head * angles.cos() + _rotate_every_two(head) * angles.sin()
+    cos = angles.cos().to(t.dtype)
+    sin = angles.sin().to(t.dtype)
+    head = head * cos + _rotate_every_two(head) * sin
     return torch.cat((head, tail), dim=-1) if tail.shape[-1] else head
 
 

@@ -2,6 +2,11 @@
 """Diagnostic: MFMA layout probe + flash-attention error report.
 Run on a GPU box; prints everything needed to fix operand layouts offline."""
 
+import sys
+from pathlib import Path
+
+sys.path.insert(0, str(Path(__file__).resolve().parents[1]))
+
 import torch
 
 import dalle_pytorch_amd._hip as ext

@@ -130,6 +130,31 @@ def test_attention_core_uses_hip_kernel():
     assert called.get('yes'), 'eager fallback ran on GPU'
 
 
+def test_autocast_rotary_stays_on_hip_path():
+    """Rotary application must not promote q/k/v to fp32 under autocast —
+    that silently pushes training attention onto the eager path."""
+    from dalle_pytorch_amd.models.attention import Attention
+    from dalle_pytorch_amd.models.positional import build_dalle_rotary_table
+    from dalle_pytorch_amd.ops import attention as attn_mod
+    attn = Attention(dim=128, seq_len=68, heads=2, dim_head=64).cuda()
+    table = build_dalle_rotary_table(64, 4, 8).cuda()
+    x = torch.randn(1, 68, 128, device='cuda')
+    called = {}
+    orig = attn_mod._FlashAttention.apply
+
+    def spy(*a):
+        called['yes'] = True
+        return orig(*a)
+
+    attn_mod._FlashAttention.apply = spy
+    try:
+        with torch.autocast(device_type='cuda', dtype=torch.bfloat16):
+            attn(x, rotary_pos_emb=table[..., :68, :])
+    finally:
+        attn_mod._FlashAttention.apply = orig
+    assert called.get('yes'), 'rotary path fell back to eager attention'
+
+
 def test_geglu_vs_oracle(ext):
     torch.manual_seed(4)
     x0 = torch.randn(4, 96, 512, device='cuda')
@@ -138,14 +163,16 @@ def test_geglu_vs_oracle(ext):
     out = geglu(x)
     a, g = x0.chunk(2, -1)
     ref = a * torch.nn.functional.gelu(g)
-    assert (out.float() - ref).abs().max().item() < 2e-2
+    err = (out.float() - ref).abs().max().item() / ref.abs().max().item()
+    assert err < 1e-2, f'fwd rel err {err}'   # bf16 storage rounding
 
     dout = torch.randn_like(out, dtype=torch.float32)
     out.backward(dout.bfloat16())
     xr = x0.clone().requires_grad_()
     ar, gr = xr.chunk(2, -1)
     (ar * torch.nn.functional.gelu(gr)).backward(dout)
-    assert (x.grad.float() - xr.grad).abs().max().item() < 2e-2
+    gerr = (x.grad.float() - xr.grad).abs().max().item() / xr.grad.abs().max().item()
+    assert gerr < 1e-2, f'bwd rel err {gerr}'
 
 
 def test_dalle_train_step_gpu():
