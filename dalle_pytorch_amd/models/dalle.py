@@ -344,9 +344,14 @@ class DALLE(nn.Module):
         offsetted_image = image + self.num_text_tokens
         labels = torch.cat((text[:, 1:], offsetted_image), dim=1)
 
-        logits = logits.transpose(1, 2)
+        # CE over the contiguous [rows, vocab] layout: same math as the
+        # reference's 'b n c -> b c n' form (dalle_pytorch.py:665-669) but it
+        # hits the fast row-softmax kernel instead of SpatialSoftMax over a
+        # strided dim (measured 26% of step time on MI355X)
+        C = logits.shape[-1]
+        tlen = self.text_seq_len
         loss_text = F.cross_entropy(
-            logits[:, :, :self.text_seq_len], labels[:, :self.text_seq_len])
+            logits[:, :tlen].reshape(-1, C), labels[:, :tlen].reshape(-1))
         loss_img = F.cross_entropy(
-            logits[:, :, self.text_seq_len:], labels[:, self.text_seq_len:])
+            logits[:, tlen:].reshape(-1, C), labels[:, tlen:].reshape(-1))
         return (loss_text + self.loss_img_weight * loss_img) / (self.loss_img_weight + 1)

@@ -35,8 +35,12 @@ def route_args(router, args, depth):
 
 
 class Deterministic(nn.Module):
-    """Wraps a module so a later replay reproduces its RNG draws exactly
-    (reference reversible.py:20-50)."""
+    """Wraps a module so a later replay reproduces its forward exactly
+    (reference reversible.py:20-50): RNG state for dropout patterns AND the
+    autocast state. The latter goes beyond the reference — its recompute ran
+    in whatever precision the backward thread had, which on a bf16-autocast
+    MI355X run would silently recompute in fp32 (no bf16 MFMA, ~8x slower
+    GEMMs, measured in rocprof as fp32 Tensile kernels)."""
 
     def __init__(self, net):
         super().__init__()
@@ -45,12 +49,15 @@ class Deterministic(nn.Module):
         self._gpu_devices = None
         self._gpu_states = None
         self._had_gpu = False
+        self._autocast = None   # (enabled, dtype) for 'cuda'
 
     def record_rng(self, *tensors):
         self._cpu_state = torch.get_rng_state()
         if torch.cuda._initialized:
             self._had_gpu = True
             self._gpu_devices, self._gpu_states = get_device_states(*tensors)
+        self._autocast = (torch.is_autocast_enabled('cuda'),
+                          torch.get_autocast_dtype('cuda'))
 
     def forward(self, *args, record_rng=False, set_rng=False, **kwargs):
         if record_rng:
@@ -62,6 +69,9 @@ class Deterministic(nn.Module):
             torch.set_rng_state(self._cpu_state)
             if self._had_gpu:
                 set_device_states(self._gpu_devices, self._gpu_states)
+            if self._autocast is not None and self._autocast[0] and torch.cuda.is_available():
+                with torch.autocast('cuda', dtype=self._autocast[1], enabled=True):
+                    return self.net(*args, **kwargs)
             return self.net(*args, **kwargs)
 
 
