@@ -41,7 +41,13 @@ class DivideMax(nn.Module):
 
 class NonCached(nn.Module):
     """Rebuilds the full sequence for layers without native cache support,
-    then returns only the suffix (reference transformer.py:38-58)."""
+    then returns only the suffix (reference transformer.py:38-58).
+
+    When no cache_key is supplied (the PreShiftToken wrapper does not thread
+    one through, reference transformer.py:153,200 — under which the
+    reference's sparse layers would all share ``cache[None]`` and corrupt
+    each other during cached decode) a per-instance key is used instead.
+    """
 
     def __init__(self, fn):
         super().__init__()
@@ -50,6 +56,8 @@ class NonCached(nn.Module):
     def forward(self, x, *, cache=None, cache_key=None, **kwargs):
         n = x.shape[-2]
         if cache is not None:
+            if cache_key is None:
+                cache_key = ('noncached', id(self))
             if cache_key in cache:
                 x = torch.cat([cache[cache_key], x], dim=-2)
             cache[cache_key] = x

@@ -89,26 +89,33 @@ def _flash_bwd_composite(q, k, v, out, lse, dout, scale, causal, key_mask,
 class _FlashAttention(torch.autograd.Function):
     """Binds the gfx950 flash-attention forward kernel (saves out +
     logsumexp); backward recomputes probabilities chunk-by-chunk — no
-    n x n matrix is ever stored across the fwd/bwd boundary."""
+    n x n matrix is ever stored across the fwd/bwd boundary. With
+    ``fold_heads`` the kernel epilogue writes [b, n, h, d] directly (no
+    head-merge permute before the output projection)."""
 
     @staticmethod
-    def forward(ctx, q, k, v, scale, causal, key_mask, static_mask):
+    def forward(ctx, q, k, v, scale, causal, key_mask, static_mask,
+                tile_map, fold_heads):
         ext = hip_module()
         q, k, v = (t.contiguous() for t in (q, k, v))
         out, lse = ext.fa_fwd(q, k, v, scale, causal,
-                              key_mask, static_mask)
+                              key_mask, static_mask, tile_map, fold_heads)
         ctx.save_for_backward(q, k, v, out, lse)
         ctx.scale, ctx.causal = scale, causal
         ctx.key_mask, ctx.static_mask = key_mask, static_mask
+        ctx.fold_heads = fold_heads
         return out
 
     @staticmethod
     def backward(ctx, dout):
         q, k, v, out, lse = ctx.saved_tensors
+        if ctx.fold_heads:   # [b, n, h, d] -> [b, h, n, d]
+            out = out.permute(0, 2, 1, 3)
+            dout = dout.permute(0, 2, 1, 3)
         dq, dk, dv = _flash_bwd_composite(
             q, k, v, out, lse, dout.contiguous(), ctx.scale, ctx.causal,
             ctx.key_mask, ctx.static_mask)
-        return dq, dk, dv, None, None, None, None
+        return dq, dk, dv, None, None, None, None, None, None
 
 
 def _hip_supported(q, k, causal, key_mask):
@@ -126,17 +133,28 @@ def _hip_supported(q, k, causal, key_mask):
 _warned_shapes = set()
 
 
-def attention_core(q, k, v, scale, causal=True, key_mask=None, static_mask=None):
+def _fold(out):
+    b, h, n, d = out.shape
+    return out.permute(0, 2, 1, 3).reshape(b, n, h * d)
+
+
+def attention_core(q, k, v, scale, causal=True, key_mask=None, static_mask=None,
+                   static_tiles=None, fold_heads=False):
     """Scaled-dot-product attention with the reference's masking semantics.
 
     q: [b, h, nq, d] (unscaled), k/v: [b, h, nk, d],
     key_mask: optional bool [b, nk] (True = attend),
     static_mask: optional bool [nq, nk] (True = attend), already sliced for
-    any cache offset (reference attention.py:91-92).
-    Returns [b, h, nq, d].
+    any cache offset (reference attention.py:91-92),
+    static_tiles: optional uint8 [ceil(nq/64), ceil(nk/32)] block map of
+    static_mask — fully-zero tiles are skipped by the kernel (this is how
+    axial/conv/block-sparse patterns become truly sparse),
+    fold_heads: return [b, nq, h*d] (GPU kernel writes it directly).
+    Returns [b, h, nq, d] or [b, nq, h*d].
     """
     if using_eager_fallback(q):
-        return _eager_attention(q, k, v, scale, causal, key_mask, static_mask)
+        out = _eager_attention(q, k, v, scale, causal, key_mask, static_mask)
+        return _fold(out) if fold_heads else out
     if not _hip_supported(q, k, causal, key_mask):
         if q.shape[-2] >= 32:  # decode fallback is intentional and silent
             key = (q.shape[-1], str(q.dtype))
@@ -144,9 +162,28 @@ def attention_core(q, k, v, scale, causal=True, key_mask=None, static_mask=None)
                 _warned_shapes.add(key)
                 warnings.warn(f'attention_core: shape/dtype {key} not covered '
                               'by the HIP kernel yet; using eager path on GPU')
-        return _eager_attention(q, k, v, scale, causal, key_mask, static_mask)
+        out = _eager_attention(q, k, v, scale, causal, key_mask, static_mask)
+        return _fold(out) if fold_heads else out
     if static_mask is not None:
         static_mask = static_mask.contiguous()
     if key_mask is not None:
         key_mask = key_mask.contiguous()
-    return _FlashAttention.apply(q, k, v, scale, causal, key_mask, static_mask)
+    out = _FlashAttention.apply(q, k, v, scale, causal, key_mask, static_mask,
+                                static_tiles, fold_heads)
+    if fold_heads:
+        b, n, h, d = out.shape
+        return out.view(b, n, h * d)
+    return out
+
+
+def build_tile_map(static_mask, causal=False):
+    """uint8 [ceil(nq/64), ceil(nk/32)] block map: 1 where any mask entry in
+    the (64, 32) tile is True (conservative — kernel still applies the exact
+    element mask)."""
+    nq, nk = static_mask.shape
+    tq, tk = (nq + 63) // 64, (nk + 31) // 32
+    padded = torch.zeros(tq * 64, tk * 32, dtype=torch.bool,
+                         device=static_mask.device)
+    padded[:nq, :nk] = static_mask
+    tiles = padded.reshape(tq, 64, tk, 32).any(dim=3).any(dim=1)
+    return tiles.to(torch.uint8).contiguous()

@@ -72,12 +72,13 @@ void fa_fwd_d64_kernel(
     const short* __restrict__ q,    // [bh, nq, 64] bf16 bits
     const short* __restrict__ k,    // [bh, nk, 64]
     const short* __restrict__ v,    // [bh, nk, 64]
-    short* __restrict__ out,        // [bh, nq, 64]
+    short* __restrict__ out,        // [bh, nq, 64], or [b, nq, h, 64] if out_bnhd
     float* __restrict__ lse,        // [bh, nq]
     const bool* __restrict__ key_mask,    // [b, nk] or null
     const bool* __restrict__ static_mask, // [nq, nk] or null
+    const unsigned char* __restrict__ tile_map,  // [ceil(nq/64), ceil(nk/32)] or null
     int b, int h, int nq, int nk,
-    float scale, int causal) {
+    float scale, int causal, int out_bnhd) {
 
   __shared__ short Kt[FA_KBLK][KPAD];
   __shared__ short Vt[FA_D][VPAD];
@@ -120,12 +121,18 @@ void fa_fwd_d64_kernel(
                   f32x4{0, 0, 0, 0}, f32x4{0, 0, 0, 0}};
 
   int ntiles = (nk + FA_KBLK - 1) / FA_KBLK;
+  const int ntk = ntiles;
   if (causal) {
     const int lim = min(nk - 1, q0 + FA_QBLK - 1 + diag);
     ntiles = lim < 0 ? 0 : (lim / FA_KBLK + 1);
   }
+  const unsigned char* tmap_row =
+      tile_map ? tile_map + (long)blockIdx.x * ntk : nullptr;
 
   for (int kt = 0; kt < ntiles; ++kt) {
+    // block-sparse skip: the host precomputes, per (64-q, 32-k) tile, whether
+    // any static-mask entry is set (serves axial/conv/block-sparse patterns)
+    if (tmap_row && !tmap_row[kt]) continue;
     const int kbase = kt * FA_KBLK;
 
     // ---- stage K (row-major, padded) and V (transposed) tiles
@@ -220,22 +227,124 @@ void fa_fwd_d64_kernel(
     }
   }
 
-  // ---- epilogue: divide by l per q row, store out + lse
+  // ---- epilogue: divide by l per q row, store out + lse. out_bnhd writes
+  // the [b, n, h*d] layout directly so no head-merge permute kernel is
+  // needed before the output projection GEMM.
+  const int batch_i = bh / h, head_i = bh - batch_i * h;
   #pragma unroll
   for (int r = 0; r < 4; ++r) {
     const int qr = q0 + wave * 16 + grp * 4 + r;
     const float l_r = __shfl(l_run, grp * 4 + r);
     const float inv = l_r > 0.f ? 1.f / l_r : 0.f;
     if (qr < nq) {
+      const long base = out_bnhd
+          ? (((long)batch_i * nq + qr) * h + head_i) * FA_D
+          : ((long)bh * nq + qr) * FA_D;
       #pragma unroll
       for (int nt = 0; nt < 4; ++nt) {
-        out[((long)bh * nq + qr) * FA_D + 16 * nt + lq] = f2bf(acc[nt][r] * inv);
+        out[base + 16 * nt + lq] = f2bf(acc[nt][r] * inv);
       }
     }
   }
   if (grp == 0 && qrow < nq) {
     lse[(long)bh * nq + qrow] = (l_run > 0.f) ? m_run + __logf(l_run) : NEG_INF;
   }
+}
+
+// ---------------------------------------------------------------------------
+// Fused QKV split + rotary embedding (kernels K1-K2 glue).
+//
+// One pass turns the to_qkv GEMM output [b, n, 3*h*d] into contiguous
+// q/k/v [b, h, n, d] with the interleaved-pair rotary rotation applied to
+// the first `rot` channels of ALL THREE tensors (the reference's
+// rotary-on-v quirk, attention.py:35,67). Replaces ~10 eager kernels per
+// tensor (cos/sin/cast/mul/add/cat/permute/contiguous) with one
+// memory-bound sweep. d = 64 fixed; rot may be 0 (pure split).
+// ---------------------------------------------------------------------------
+
+__global__ void rope_split_fwd_kernel(
+    const short* __restrict__ qkv,   // [b, n, 3*h*64]
+    const float* __restrict__ cosv,  // [n, rot] (interleaved-duplicated)
+    const float* __restrict__ sinv,  // [n, rot]
+    short* __restrict__ qo,          // [b, h, n, 64]
+    short* __restrict__ ko,
+    short* __restrict__ vo,
+    int b, int h, int n, int rot) {
+  const int chunks_per_row = 3 * h * 8;
+  const int c = blockIdx.x * blockDim.x + threadIdx.x;
+  const int bi = blockIdx.y;
+  if (c >= (long)n * chunks_per_row) return;
+  const int ni = c / chunks_per_row;
+  const int rc = c - ni * chunks_per_row;   // chunk within the row
+  const int which = rc / (h * 8);           // 0=q 1=k 2=v
+  const int head = (rc / 8) % h;
+  const int d0 = (rc & 7) * 8;
+
+  int4v x16 = *reinterpret_cast<const int4v*>(
+      qkv + ((long)bi * n + ni) * (3L * h * 64) + rc * 8);
+  const short* xs = reinterpret_cast<const short*>(&x16);
+  short y[8];
+  #pragma unroll
+  for (int e = 0; e < 8; e += 2) {
+    const int d = d0 + e;
+    if (d < rot) {
+      const float cs = cosv[(long)ni * rot + d];
+      const float sn = sinv[(long)ni * rot + d];
+      const float a = bf2f(xs[e]), bb = bf2f(xs[e + 1]);
+      y[e] = f2bf(a * cs - bb * sn);
+      y[e + 1] = f2bf(bb * cs + a * sn);
+    } else {
+      y[e] = xs[e];
+      y[e + 1] = xs[e + 1];
+    }
+  }
+  short* dst = (which == 0 ? qo : which == 1 ? ko : vo);
+  *reinterpret_cast<int4v*>(
+      dst + (((long)bi * h + head) * n + ni) * 64 + d0) =
+      *reinterpret_cast<const int4v*>(y);
+}
+
+__global__ void rope_split_bwd_kernel(
+    const short* __restrict__ dq,    // [b, h, n, 64]
+    const short* __restrict__ dk,
+    const short* __restrict__ dv,
+    const float* __restrict__ cosv,
+    const float* __restrict__ sinv,
+    short* __restrict__ dqkv,        // [b, n, 3*h*64]
+    int b, int h, int n, int rot) {
+  const int chunks_per_row = 3 * h * 8;
+  const int c = blockIdx.x * blockDim.x + threadIdx.x;
+  const int bi = blockIdx.y;
+  if (c >= (long)n * chunks_per_row) return;
+  const int ni = c / chunks_per_row;
+  const int rc = c - ni * chunks_per_row;
+  const int which = rc / (h * 8);
+  const int head = (rc / 8) % h;
+  const int d0 = (rc & 7) * 8;
+
+  const short* src = (which == 0 ? dq : which == 1 ? dk : dv);
+  int4v x16 = *reinterpret_cast<const int4v*>(
+      src + (((long)bi * h + head) * n + ni) * 64 + d0);
+  const short* xs = reinterpret_cast<const short*>(&x16);
+  short y[8];
+  #pragma unroll
+  for (int e = 0; e < 8; e += 2) {
+    const int d = d0 + e;
+    if (d < rot) {
+      // transpose rotation: dx = dy*cos - rotate_half(dy)*sin
+      const float cs = cosv[(long)ni * rot + d];
+      const float sn = sinv[(long)ni * rot + d];
+      const float g1 = bf2f(xs[e]), g2 = bf2f(xs[e + 1]);
+      y[e] = f2bf(g1 * cs + g2 * sn);
+      y[e + 1] = f2bf(g2 * cs - g1 * sn);
+    } else {
+      y[e] = xs[e];
+      y[e + 1] = xs[e + 1];
+    }
+  }
+  *reinterpret_cast<int4v*>(
+      dqkv + ((long)bi * n + ni) * (3L * h * 64) + rc * 8) =
+      *reinterpret_cast<const int4v*>(y);
 }
 
 // ---------------------------------------------------------------------------
@@ -322,18 +431,22 @@ static hipStream_t cur_stream() {
 std::vector<torch::Tensor> fa_fwd(torch::Tensor q, torch::Tensor k, torch::Tensor v,
                                   double scale, bool causal,
                                   std::optional<torch::Tensor> key_mask,
-                                  std::optional<torch::Tensor> static_mask) {
+                                  std::optional<torch::Tensor> static_mask,
+                                  std::optional<torch::Tensor> tile_map,
+                                  bool out_bnhd) {
   CHK(q.is_cuda() && k.is_cuda() && v.is_cuda());
   CHK(q.dtype() == torch::kBFloat16);
   CHK(q.size(-1) == FA_D);
   CHK(q.is_contiguous() && k.is_contiguous() && v.is_contiguous());
   const int b = q.size(0), h = q.size(1), nq = q.size(2), nk = k.size(2);
 
-  auto out = torch::empty_like(q);
+  auto out = out_bnhd ? torch::empty({b, nq, h, FA_D}, q.options())
+                      : torch::empty_like(q);
   auto lse = torch::empty({b, h, nq}, q.options().dtype(torch::kFloat32));
 
   const bool* km = nullptr;
   const bool* sm = nullptr;
+  const unsigned char* tm = nullptr;
   if (key_mask.has_value()) {
     CHK(key_mask->dtype() == torch::kBool && key_mask->is_contiguous());
     CHK(key_mask->size(0) == b && key_mask->size(1) == nk);
@@ -344,6 +457,12 @@ std::vector<torch::Tensor> fa_fwd(torch::Tensor q, torch::Tensor k, torch::Tenso
     CHK(static_mask->size(0) == nq && static_mask->size(1) == nk);
     sm = static_mask->data_ptr<bool>();
   }
+  if (tile_map.has_value()) {
+    CHK(tile_map->dtype() == torch::kUInt8 && tile_map->is_contiguous());
+    CHK(tile_map->size(0) == (nq + FA_QBLK - 1) / FA_QBLK);
+    CHK(tile_map->size(1) == (nk + FA_KBLK - 1) / FA_KBLK);
+    tm = tile_map->data_ptr<uint8_t>();
+  }
 
   dim3 grid((nq + FA_QBLK - 1) / FA_QBLK, b * h);
   hipLaunchKernelGGL(fa_fwd_d64_kernel, grid, dim3(256), 0, cur_stream(),
@@ -351,9 +470,68 @@ std::vector<torch::Tensor> fa_fwd(torch::Tensor q, torch::Tensor k, torch::Tenso
                      reinterpret_cast<const short*>(k.data_ptr()),
                      reinterpret_cast<const short*>(v.data_ptr()),
                      reinterpret_cast<short*>(out.data_ptr()),
-                     lse.data_ptr<float>(), km, sm,
-                     b, h, nq, nk, (float)scale, causal ? 1 : 0);
+                     lse.data_ptr<float>(), km, sm, tm,
+                     b, h, nq, nk, (float)scale, causal ? 1 : 0,
+                     out_bnhd ? 1 : 0);
   return {out, lse};
+}
+
+std::vector<torch::Tensor> rope_split_fwd(torch::Tensor qkv, int64_t heads,
+                                          std::optional<torch::Tensor> cosv,
+                                          std::optional<torch::Tensor> sinv) {
+  CHK(qkv.is_cuda() && qkv.dtype() == torch::kBFloat16 && qkv.is_contiguous());
+  const int b = qkv.size(0), n = qkv.size(1);
+  const int h = (int)heads;
+  CHK(qkv.size(2) == 3L * h * FA_D);
+  int rot = 0;
+  const float* cp = nullptr;
+  const float* sp = nullptr;
+  if (cosv.has_value()) {
+    CHK(cosv->dtype() == torch::kFloat32 && cosv->is_contiguous());
+    CHK(sinv->dtype() == torch::kFloat32 && sinv->is_contiguous());
+    CHK(cosv->size(0) == n && sinv->size(0) == n);
+    rot = cosv->size(1);
+    CHK(rot % 2 == 0 && rot <= FA_D);
+    cp = cosv->data_ptr<float>();
+    sp = sinv->data_ptr<float>();
+  }
+  auto opts = qkv.options();
+  auto q = torch::empty({b, h, n, FA_D}, opts);
+  auto k = torch::empty({b, h, n, FA_D}, opts);
+  auto v = torch::empty({b, h, n, FA_D}, opts);
+  const long chunks = (long)n * 3 * h * 8;
+  dim3 grid((chunks + 255) / 256, b);
+  hipLaunchKernelGGL(rope_split_fwd_kernel, grid, dim3(256), 0, cur_stream(),
+                     reinterpret_cast<const short*>(qkv.data_ptr()), cp, sp,
+                     reinterpret_cast<short*>(q.data_ptr()),
+                     reinterpret_cast<short*>(k.data_ptr()),
+                     reinterpret_cast<short*>(v.data_ptr()), b, h, n, rot);
+  return {q, k, v};
+}
+
+torch::Tensor rope_split_bwd(torch::Tensor dq, torch::Tensor dk, torch::Tensor dv,
+                             std::optional<torch::Tensor> cosv,
+                             std::optional<torch::Tensor> sinv) {
+  CHK(dq.is_cuda() && dq.dtype() == torch::kBFloat16);
+  auto dqc = dq.contiguous(), dkc = dk.contiguous(), dvc = dv.contiguous();
+  const int b = dqc.size(0), h = dqc.size(1), n = dqc.size(2);
+  int rot = 0;
+  const float* cp = nullptr;
+  const float* sp = nullptr;
+  if (cosv.has_value()) {
+    rot = cosv->size(1);
+    cp = cosv->data_ptr<float>();
+    sp = sinv->data_ptr<float>();
+  }
+  auto dqkv = torch::empty({b, n, 3L * h * FA_D}, dqc.options());
+  const long chunks = (long)n * 3 * h * 8;
+  dim3 grid((chunks + 255) / 256, b);
+  hipLaunchKernelGGL(rope_split_bwd_kernel, grid, dim3(256), 0, cur_stream(),
+                     reinterpret_cast<const short*>(dqc.data_ptr()),
+                     reinterpret_cast<const short*>(dkc.data_ptr()),
+                     reinterpret_cast<const short*>(dvc.data_ptr()), cp, sp,
+                     reinterpret_cast<short*>(dqkv.data_ptr()), b, h, n, rot);
+  return dqkv;
 }
 
 torch::Tensor geglu_fwd(torch::Tensor x) {
@@ -401,6 +579,9 @@ torch::Tensor mfma_probe(torch::Tensor A, torch::Tensor B) {
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("fa_fwd", &fa_fwd, "flash attention forward (gfx950, d=64)");
+  m.def("rope_split_fwd", &rope_split_fwd,
+        "fused qkv split + rotary (q,k,v all rotated)");
+  m.def("rope_split_bwd", &rope_split_bwd, "rope_split backward");
   m.def("geglu_fwd", &geglu_fwd, "fused GEGLU forward");
   m.def("geglu_bwd", &geglu_bwd, "fused GEGLU backward");
   m.def("mfma_probe", &mfma_probe, "MFMA 16x16x32 bf16 layout probe");
