@@ -66,12 +66,14 @@ _MASK_CACHE = {}
 
 
 def _cached_mask(key, device, build):
-    """(mask[n,n] bool, tiles uint8) on device, built once per pattern."""
+    """(mask bool, tiles uint8 [nq/64,nk/32], tiles_t uint8 [nk/64,nq/32])
+    on device, built once per pattern. tiles serves the forward + dQ kernels;
+    tiles_t the dK/dV kernel (keys-major blocking)."""
     full_key = (*key, str(device))
     hit = _MASK_CACHE.get(full_key)
     if hit is None:
         mask = build().to(device)
-        hit = (mask, build_tile_map(mask))
+        hit = (mask, build_tile_map(mask), build_tile_map(mask.t()))
         _MASK_CACHE[full_key] = hit
     return hit
 
@@ -168,13 +170,14 @@ class Attention(nn.Module):
 
     def _masks(self, offset, n_q, n_k, device):
         if self.static_mask is None:
-            return None, None
+            return None, None, None
         sm = self.static_mask[offset:offset + n_q, :n_k]
-        tiles = None
+        tiles = tiles_t = None
         if sm.is_cuda and offset == 0 and n_k == self.static_mask.shape[1]:
-            _, tiles = _cached_mask(('attn-static', id(self.static_mask), n_q, n_k),
-                                    device, lambda: sm)
-        return sm, tiles
+            _, tiles, tiles_t = _cached_mask(
+                ('attn-static', id(self.static_mask), n_q, n_k),
+                device, lambda: sm)
+        return sm, tiles, tiles_t
 
     def forward(self, x, mask=None, rotary_pos_emb=None, cache=None, cache_key=None):
         h = self.heads
@@ -190,12 +193,12 @@ class Attention(nn.Module):
             cache[cache_key] = (k, v)
 
         n_q, n_k = q.shape[-2], k.shape[-2]
-        static, tiles = self._masks(offset, n_q, n_k, x.device)
+        static, tiles, tiles_t = self._masks(offset, n_q, n_k, x.device)
         out = attention_core(
             q, k, v, self.scale,
             causal=self.causal and offset == 0,
             key_mask=mask, static_mask=static, static_tiles=tiles,
-            fold_heads=True)
+            static_tiles_t=tiles_t, fold_heads=True)
         return self.to_out(out)
 
 
@@ -219,12 +222,13 @@ class _StaticMaskSparseAttention(nn.Module):
 
     # subclasses define: _mask_key(), _build_mask() -> [seq_len+1?, ...] bool
     def _pattern(self, n, device):
-        mask, tiles = _cached_mask(self._mask_key(), device, self._build_mask)
+        mask, tiles, tiles_t = _cached_mask(self._mask_key(), device,
+                                            self._build_mask)
         if n == mask.shape[0]:
-            return mask, tiles
+            return mask, tiles, tiles_t
         sm = mask[:n, :n]
-        _, t = _cached_mask((*self._mask_key(), 'n', n), device, lambda: sm)
-        return sm, t
+        _, t, tt = _cached_mask((*self._mask_key(), 'n', n), device, lambda: sm)
+        return sm, t, tt
 
     def _key_mask(self, mask, b, n, text_len, device):
         if mask is None:
@@ -238,11 +242,12 @@ class _StaticMaskSparseAttention(nn.Module):
         b, n, _ = x.shape
         q, k, v = _qkv_heads(x, self.to_qkv, self.heads, self.dim_head,
                              rotary_pos_emb, 0)
-        static, tiles = self._pattern(n, x.device)
+        static, tiles, tiles_t = self._pattern(n, x.device)
         km = self._key_mask(mask, b, n, getattr(self, 'text_len', n), x.device)
         out = attention_core(q, k, v, self.scale, causal=self.causal,
                              key_mask=km, static_mask=static,
-                             static_tiles=tiles, fold_heads=True)
+                             static_tiles=tiles, static_tiles_t=tiles_t,
+                             fold_heads=True)
         return self.to_out(out)
 
 

@@ -95,7 +95,7 @@ class _FlashAttention(torch.autograd.Function):
 
     @staticmethod
     def forward(ctx, q, k, v, scale, causal, key_mask, static_mask,
-                tile_map, fold_heads):
+                tile_map, tile_map_t, fold_heads):
         ext = hip_module()
         q, k, v = (t.contiguous() for t in (q, k, v))
         out, lse = ext.fa_fwd(q, k, v, scale, causal,
@@ -103,6 +103,7 @@ class _FlashAttention(torch.autograd.Function):
         ctx.save_for_backward(q, k, v, out, lse)
         ctx.scale, ctx.causal = scale, causal
         ctx.key_mask, ctx.static_mask = key_mask, static_mask
+        ctx.tile_map, ctx.tile_map_t = tile_map, tile_map_t
         ctx.fold_heads = fold_heads
         return out
 
@@ -112,10 +113,17 @@ class _FlashAttention(torch.autograd.Function):
         if ctx.fold_heads:   # [b, n, h, d] -> [b, h, n, d]
             out = out.permute(0, 2, 1, 3)
             dout = dout.permute(0, 2, 1, 3)
-        dq, dk, dv = _flash_bwd_composite(
-            q, k, v, out, lse, dout.contiguous(), ctx.scale, ctx.causal,
-            ctx.key_mask, ctx.static_mask)
-        return dq, dk, dv, None, None, None, None, None, None
+        ext = hip_module()
+        if hasattr(ext, 'fa_bwd'):
+            dq, dk, dv = ext.fa_bwd(
+                q, k, v, out.contiguous(), lse, dout.contiguous(),
+                ctx.scale, ctx.causal, ctx.key_mask, ctx.static_mask,
+                ctx.tile_map, ctx.tile_map_t)
+        else:
+            dq, dk, dv = _flash_bwd_composite(
+                q, k, v, out, lse, dout.contiguous(), ctx.scale, ctx.causal,
+                ctx.key_mask, ctx.static_mask)
+        return dq, dk, dv, None, None, None, None, None, None, None
 
 
 def _hip_supported(q, k, causal, key_mask):
@@ -139,7 +147,7 @@ def _fold(out):
 
 
 def attention_core(q, k, v, scale, causal=True, key_mask=None, static_mask=None,
-                   static_tiles=None, fold_heads=False):
+                   static_tiles=None, static_tiles_t=None, fold_heads=False):
     """Scaled-dot-product attention with the reference's masking semantics.
 
     q: [b, h, nq, d] (unscaled), k/v: [b, h, nk, d],
@@ -169,7 +177,7 @@ def attention_core(q, k, v, scale, causal=True, key_mask=None, static_mask=None,
     if key_mask is not None:
         key_mask = key_mask.contiguous()
     out = _FlashAttention.apply(q, k, v, scale, causal, key_mask, static_mask,
-                                static_tiles, fold_heads)
+                                static_tiles, static_tiles_t, fold_heads)
     if fold_heads:
         b, n, h, d = out.shape
         return out.view(b, n, h * d)

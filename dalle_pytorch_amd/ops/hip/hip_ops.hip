@@ -252,6 +252,299 @@ void fa_fwd_d64_kernel(
 }
 
 // ---------------------------------------------------------------------------
+// Flash attention backward, head_dim = 64 (two passes, standard flash
+// decomposition — no n x n matrix in HBM):
+//   D[q]   = rowsum(dO * O)                       (host-side fused reduce)
+//   P      = exp(scale*QK^T - lse)  (masked -> 0) (recomputed per tile)
+//   dV     = P^T dO;   dP = dO V^T
+//   dS     = P * (dP - D) * scale
+//   dQ     = dS K;     dK = dS^T Q
+// Pass 1 (dq): blocks own 64 q rows, stream K/V tiles — same swapped-MFMA
+// structure as the forward. Pass 2 (dkv): blocks own 64 keys, stream Q/dO
+// tiles with the roles mirrored. Both honor the causal bound and the
+// (64,32) tile maps, so sparse patterns stay sparse in backward too.
+// ---------------------------------------------------------------------------
+
+__global__ __launch_bounds__(256, 2)
+void fa_bwd_dq_kernel(
+    const short* __restrict__ q,     // [bh, nq, 64]
+    const short* __restrict__ k,     // [bh, nk, 64]
+    const short* __restrict__ v,     // [bh, nk, 64]
+    const short* __restrict__ dout,  // [bh, nq, 64]
+    const float* __restrict__ lse,   // [bh, nq]
+    const float* __restrict__ Dv,    // [bh, nq]
+    short* __restrict__ dq,          // [bh, nq, 64]
+    const bool* __restrict__ key_mask,
+    const bool* __restrict__ static_mask,
+    const unsigned char* __restrict__ tile_map,   // [nq/64, nk/32]
+    int b, int h, int nq, int nk, float scale, int causal) {
+
+  __shared__ short Kt[FA_KBLK][KPAD];    // K row-major
+  __shared__ short Vr[FA_KBLK][KPAD];    // V row-major (A-operand of dP^T)
+  __shared__ short Ktr[FA_D][VPAD];      // K transposed (B-operand of dS*K)
+  __shared__ short DSl[FA_WAVES][16][PPAD];
+
+  const int bh = blockIdx.y;
+  const int batch = bh / h;
+  const int q0 = blockIdx.x * FA_QBLK;
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6, lane = tid & 63;
+  const int lq = lane & 15, grp = lane >> 4;
+  const int qrow = q0 + wave * 16 + lq;
+  const int diag = nk - nq;
+
+  const short* qp = q + (long)bh * nq * FA_D;
+  const short* kp = k + (long)bh * nk * FA_D;
+  const short* vp = v + (long)bh * nk * FA_D;
+  const short* dop = dout + (long)bh * nq * FA_D;
+
+  bf16x8 qfrag[2], dofrag[2];
+  float lse_q = 0.f, D_q = 0.f;
+  {
+    const bool qok = qrow < nq;
+    #pragma unroll
+    for (int c = 0; c < 2; ++c) {
+      qfrag[c] = qok ? *reinterpret_cast<const bf16x8*>(qp + (long)qrow * FA_D + 8 * grp + 32 * c)
+                     : bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
+      dofrag[c] = qok ? *reinterpret_cast<const bf16x8*>(dop + (long)qrow * FA_D + 8 * grp + 32 * c)
+                      : bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
+    }
+    if (qok) {
+      lse_q = lse[(long)bh * nq + qrow];
+      D_q = Dv[(long)bh * nq + qrow];
+    }
+  }
+
+  f32x4 acc[4] = {f32x4{0, 0, 0, 0}, f32x4{0, 0, 0, 0},
+                  f32x4{0, 0, 0, 0}, f32x4{0, 0, 0, 0}};
+
+  int ntiles = (nk + FA_KBLK - 1) / FA_KBLK;
+  const int ntk = ntiles;
+  if (causal) {
+    const int lim = min(nk - 1, q0 + FA_QBLK - 1 + diag);
+    ntiles = lim < 0 ? 0 : (lim / FA_KBLK + 1);
+  }
+  const unsigned char* tmap_row =
+      tile_map ? tile_map + (long)blockIdx.x * ntk : nullptr;
+
+  for (int kt = 0; kt < ntiles; ++kt) {
+    if (tmap_row && !tmap_row[kt]) continue;
+    const int kbase = kt * FA_KBLK;
+    {
+      const int row = tid >> 3;
+      const int c8 = (tid & 7) * 8;
+      const int kg = kbase + row;
+      int4v kv{0, 0, 0, 0}, vv{0, 0, 0, 0};
+      if (kg < nk) {
+        kv = *reinterpret_cast<const int4v*>(kp + (long)kg * FA_D + c8);
+        vv = *reinterpret_cast<const int4v*>(vp + (long)kg * FA_D + c8);
+      }
+      __syncthreads();
+      *reinterpret_cast<int4v*>(&Kt[row][c8]) = kv;
+      *reinterpret_cast<int4v*>(&Vr[row][c8]) = vv;
+      const short* ks = reinterpret_cast<const short*>(&kv);
+      #pragma unroll
+      for (int e = 0; e < 8; ++e) Ktr[c8 + e][row] = ks[e];
+    }
+    __syncthreads();
+
+    // s^T and dp^T, swapped layout: lane -> q = lq, keys = grp*4+r (+16mt)
+    float s8[8], dp8[8];
+    #pragma unroll
+    for (int mt = 0; mt < 2; ++mt) {
+      f32x4 st{0, 0, 0, 0}, dpt{0, 0, 0, 0};
+      #pragma unroll
+      for (int c = 0; c < 2; ++c) {
+        bf16x8 kf = frag_from_lds(&Kt[mt * 16 + lq][8 * grp + 32 * c]);
+        bf16x8 vf = frag_from_lds(&Vr[mt * 16 + lq][8 * grp + 32 * c]);
+        st = __builtin_amdgcn_mfma_f32_16x16x32_bf16(kf, qfrag[c], st, 0, 0, 0);
+        dpt = __builtin_amdgcn_mfma_f32_16x16x32_bf16(vf, dofrag[c], dpt, 0, 0, 0);
+      }
+      #pragma unroll
+      for (int r = 0; r < 4; ++r) { s8[mt * 4 + r] = st[r]; dp8[mt * 4 + r] = dpt[r]; }
+    }
+
+    #pragma unroll
+    for (int i = 0; i < 8; ++i) {
+      const int kg = kbase + (i >> 2) * 16 + grp * 4 + (i & 3);
+      bool ok = (kg < nk) & (qrow < nq);
+      if (causal) ok &= kg <= qrow + diag;
+      if (key_mask != nullptr && ok) ok &= key_mask[(long)batch * nk + kg];
+      if (static_mask != nullptr && ok) ok &= static_mask[(long)qrow * nk + kg];
+      const float p = ok ? __expf(s8[i] * scale - lse_q) : 0.f;
+      const float ds = p * (dp8[i] - D_q) * scale;
+      const int kk = (i >> 2) * 16 + grp * 4 + (i & 3);
+      DSl[wave][lq][kk] = f2bf(ds);
+    }
+
+    bf16x8 dsf = frag_from_lds(&DSl[wave][lq][8 * grp]);
+    #pragma unroll
+    for (int nt = 0; nt < 4; ++nt) {
+      bf16x8 kf = frag_from_lds(&Ktr[lq + 16 * nt][8 * grp]);
+      acc[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(dsf, kf, acc[nt], 0, 0, 0);
+    }
+  }
+
+  #pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const int qr = q0 + wave * 16 + grp * 4 + r;
+    if (qr < nq) {
+      #pragma unroll
+      for (int nt = 0; nt < 4; ++nt) {
+        dq[((long)bh * nq + qr) * FA_D + 16 * nt + lq] = f2bf(acc[nt][r]);
+      }
+    }
+  }
+}
+
+__global__ __launch_bounds__(256, 2)
+void fa_bwd_dkv_kernel(
+    const short* __restrict__ q,
+    const short* __restrict__ k,
+    const short* __restrict__ v,
+    const short* __restrict__ dout,
+    const float* __restrict__ lse,
+    const float* __restrict__ Dv,
+    short* __restrict__ dk,          // [bh, nk, 64]
+    short* __restrict__ dv,          // [bh, nk, 64]
+    const bool* __restrict__ key_mask,
+    const bool* __restrict__ static_mask,
+    const unsigned char* __restrict__ tile_map_t,  // [nk/64, nq/32]
+    int b, int h, int nq, int nk, float scale, int causal) {
+
+  __shared__ short Qr[FA_KBLK][KPAD];     // Q rows (B-operand of s^T)
+  __shared__ short dOr[FA_KBLK][KPAD];    // dO rows (B-operand of dp^T)
+  __shared__ short Qtr[FA_D][VPAD];       // Q transposed (dK = dS^T Q)
+  __shared__ short dOtr[FA_D][VPAD];      // dO transposed (dV = P^T dO)
+  __shared__ short Pt[FA_WAVES][16][PPAD];
+  __shared__ short DSt[FA_WAVES][16][PPAD];
+
+  const int bh = blockIdx.y;
+  const int batch = bh / h;
+  const int k0 = blockIdx.x * FA_QBLK;    // 64 keys per block
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6, lane = tid & 63;
+  const int lq = lane & 15, grp = lane >> 4;
+  const int krow = k0 + wave * 16 + lq;   // this lane's key (A-frag row)
+  const int diag = nk - nq;
+
+  const short* qp = q + (long)bh * nq * FA_D;
+  const short* kp = k + (long)bh * nk * FA_D;
+  const short* vp = v + (long)bh * nk * FA_D;
+  const short* dop = dout + (long)bh * nq * FA_D;
+
+  bf16x8 kfrag[2], vfrag[2];
+  {
+    const bool kok = krow < nk;
+    #pragma unroll
+    for (int c = 0; c < 2; ++c) {
+      kfrag[c] = kok ? *reinterpret_cast<const bf16x8*>(kp + (long)krow * FA_D + 8 * grp + 32 * c)
+                     : bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
+      vfrag[c] = kok ? *reinterpret_cast<const bf16x8*>(vp + (long)krow * FA_D + 8 * grp + 32 * c)
+                     : bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
+    }
+  }
+
+  f32x4 acc_dk[4] = {f32x4{0, 0, 0, 0}, f32x4{0, 0, 0, 0},
+                     f32x4{0, 0, 0, 0}, f32x4{0, 0, 0, 0}};
+  f32x4 acc_dv[4] = {f32x4{0, 0, 0, 0}, f32x4{0, 0, 0, 0},
+                     f32x4{0, 0, 0, 0}, f32x4{0, 0, 0, 0}};
+
+  const int nqt = (nq + FA_KBLK - 1) / FA_KBLK;   // q tiles of 32
+  int qt_start = 0;
+  if (causal) {
+    // smallest q that can see any key in this block: q >= k - diag
+    const int qmin = k0 - diag;
+    qt_start = qmin <= 0 ? 0 : qmin / FA_KBLK;
+  }
+  const unsigned char* tmap_row =
+      tile_map_t ? tile_map_t + (long)blockIdx.x * nqt : nullptr;
+
+  for (int qt = qt_start; qt < nqt; ++qt) {
+    if (tmap_row && !tmap_row[qt]) continue;
+    const int qbase = qt * FA_KBLK;
+    {
+      const int row = tid >> 3;
+      const int c8 = (tid & 7) * 8;
+      const int qg = qbase + row;
+      int4v qv{0, 0, 0, 0}, dv16{0, 0, 0, 0};
+      if (qg < nq) {
+        qv = *reinterpret_cast<const int4v*>(qp + (long)qg * FA_D + c8);
+        dv16 = *reinterpret_cast<const int4v*>(dop + (long)qg * FA_D + c8);
+      }
+      __syncthreads();
+      *reinterpret_cast<int4v*>(&Qr[row][c8]) = qv;
+      *reinterpret_cast<int4v*>(&dOr[row][c8]) = dv16;
+      const short* qs = reinterpret_cast<const short*>(&qv);
+      const short* ds_ = reinterpret_cast<const short*>(&dv16);
+      #pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        Qtr[c8 + e][row] = qs[e];
+        dOtr[c8 + e][row] = ds_[e];
+      }
+    }
+    __syncthreads();
+
+    // s^T[key, q], dp^T[key, q]: lane -> key rows grp*4+r (of the wave's
+    // 16), q col = lq (+16mt). Note roles vs the dq kernel are mirrored:
+    // here the A-operand is the wave's OWN key fragment and q comes from LDS.
+    #pragma unroll
+    for (int mt = 0; mt < 2; ++mt) {
+      f32x4 st{0, 0, 0, 0}, dpt{0, 0, 0, 0};
+      #pragma unroll
+      for (int c = 0; c < 2; ++c) {
+        bf16x8 qf = frag_from_lds(&Qr[mt * 16 + lq][8 * grp + 32 * c]);
+        bf16x8 dof = frag_from_lds(&dOr[mt * 16 + lq][8 * grp + 32 * c]);
+        st = __builtin_amdgcn_mfma_f32_16x16x32_bf16(kfrag[c], qf, st, 0, 0, 0);
+        dpt = __builtin_amdgcn_mfma_f32_16x16x32_bf16(vfrag[c], dof, dpt, 0, 0, 0);
+      }
+      // output tile is [M=16 keys, N=16 q]: C row grp*4+r = key within the
+      // wave's 16, C col = lane&15 = q within the subtile
+      #pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int key = k0 + wave * 16 + grp * 4 + r;
+        const int qg = qbase + mt * 16 + lq;
+        bool ok = (key < nk) & (qg < nq);
+        if (causal) ok &= key <= qg + diag;
+        if (key_mask != nullptr && ok) ok &= key_mask[(long)batch * nk + key];
+        if (static_mask != nullptr && ok) ok &= static_mask[(long)qg * nk + key];
+        float p = 0.f, ds = 0.f;
+        if (ok) {
+          const float l = lse[(long)bh * nq + qg];
+          const float Dq = Dv[(long)bh * nq + qg];
+          p = __expf(st[r] * scale - l);
+          ds = p * (dpt[r] - Dq) * scale;
+        }
+        Pt[wave][grp * 4 + r][mt * 16 + lq] = f2bf(p);
+        DSt[wave][grp * 4 + r][mt * 16 + lq] = f2bf(ds);
+      }
+    }
+
+    bf16x8 pf = frag_from_lds(&Pt[wave][lq][8 * grp]);
+    bf16x8 dsf = frag_from_lds(&DSt[wave][lq][8 * grp]);
+    #pragma unroll
+    for (int nt = 0; nt < 4; ++nt) {
+      bf16x8 dof = frag_from_lds(&dOtr[lq + 16 * nt][8 * grp]);
+      bf16x8 qf = frag_from_lds(&Qtr[lq + 16 * nt][8 * grp]);
+      acc_dv[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pf, dof, acc_dv[nt], 0, 0, 0);
+      acc_dk[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(dsf, qf, acc_dk[nt], 0, 0, 0);
+    }
+  }
+
+  #pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const int kr = k0 + wave * 16 + grp * 4 + r;
+    if (kr < nk) {
+      #pragma unroll
+      for (int nt = 0; nt < 4; ++nt) {
+        dk[((long)bh * nk + kr) * FA_D + 16 * nt + lq] = f2bf(acc_dk[nt][r]);
+        dv[((long)bh * nk + kr) * FA_D + 16 * nt + lq] = f2bf(acc_dv[nt][r]);
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
 // Fused QKV split + rotary embedding (kernels K1-K2 glue).
 //
 // One pass turns the to_qkv GEMM output [b, n, 3*h*d] into contiguous
@@ -476,6 +769,58 @@ std::vector<torch::Tensor> fa_fwd(torch::Tensor q, torch::Tensor k, torch::Tenso
   return {out, lse};
 }
 
+std::vector<torch::Tensor> fa_bwd(torch::Tensor q, torch::Tensor k, torch::Tensor v,
+                                  torch::Tensor out, torch::Tensor lse,
+                                  torch::Tensor dout,
+                                  double scale, bool causal,
+                                  std::optional<torch::Tensor> key_mask,
+                                  std::optional<torch::Tensor> static_mask,
+                                  std::optional<torch::Tensor> tile_map,
+                                  std::optional<torch::Tensor> tile_map_t) {
+  CHK(q.is_cuda() && q.dtype() == torch::kBFloat16);
+  CHK(q.is_contiguous() && k.is_contiguous() && v.is_contiguous());
+  CHK(out.is_contiguous() && dout.is_contiguous());
+  const int b = q.size(0), h = q.size(1), nq = q.size(2), nk = k.size(2);
+
+  // D = rowsum(dO * O), fp32 (single fused pass through ATen)
+  auto Dv = (dout.to(torch::kFloat32) * out.to(torch::kFloat32)).sum(-1);
+  Dv = Dv.contiguous();
+
+  auto dq = torch::empty_like(q);
+  auto dk = torch::empty_like(k);
+  auto dv = torch::empty_like(v);
+
+  const bool* km = nullptr;
+  const bool* sm = nullptr;
+  const unsigned char* tm = nullptr;
+  const unsigned char* tmt = nullptr;
+  if (key_mask.has_value()) km = key_mask->data_ptr<bool>();
+  if (static_mask.has_value()) sm = static_mask->data_ptr<bool>();
+  if (tile_map.has_value()) tm = tile_map->data_ptr<uint8_t>();
+  if (tile_map_t.has_value()) tmt = tile_map_t->data_ptr<uint8_t>();
+
+  dim3 grid_q((nq + FA_QBLK - 1) / FA_QBLK, b * h);
+  hipLaunchKernelGGL(fa_bwd_dq_kernel, grid_q, dim3(256), 0, cur_stream(),
+                     reinterpret_cast<const short*>(q.data_ptr()),
+                     reinterpret_cast<const short*>(k.data_ptr()),
+                     reinterpret_cast<const short*>(v.data_ptr()),
+                     reinterpret_cast<const short*>(dout.data_ptr()),
+                     lse.data_ptr<float>(), Dv.data_ptr<float>(),
+                     reinterpret_cast<short*>(dq.data_ptr()),
+                     km, sm, tm, b, h, nq, nk, (float)scale, causal ? 1 : 0);
+  dim3 grid_k((nk + FA_QBLK - 1) / FA_QBLK, b * h);
+  hipLaunchKernelGGL(fa_bwd_dkv_kernel, grid_k, dim3(256), 0, cur_stream(),
+                     reinterpret_cast<const short*>(q.data_ptr()),
+                     reinterpret_cast<const short*>(k.data_ptr()),
+                     reinterpret_cast<const short*>(v.data_ptr()),
+                     reinterpret_cast<const short*>(dout.data_ptr()),
+                     lse.data_ptr<float>(), Dv.data_ptr<float>(),
+                     reinterpret_cast<short*>(dk.data_ptr()),
+                     reinterpret_cast<short*>(dv.data_ptr()),
+                     km, sm, tmt, b, h, nq, nk, (float)scale, causal ? 1 : 0);
+  return {dq, dk, dv};
+}
+
 std::vector<torch::Tensor> rope_split_fwd(torch::Tensor qkv, int64_t heads,
                                           std::optional<torch::Tensor> cosv,
                                           std::optional<torch::Tensor> sinv) {
@@ -579,6 +924,7 @@ torch::Tensor mfma_probe(torch::Tensor A, torch::Tensor B) {
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("fa_fwd", &fa_fwd, "flash attention forward (gfx950, d=64)");
+  m.def("fa_bwd", &fa_bwd, "flash attention backward (gfx950, d=64)");
   m.def("rope_split_fwd", &rope_split_fwd,
         "fused qkv split + rotary (q,k,v all rotated)");
   m.def("rope_split_bwd", &rope_split_bwd, "rope_split backward");

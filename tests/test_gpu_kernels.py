@@ -51,7 +51,7 @@ def test_fa_fwd_vs_oracle(ext, b, h, nq, nk, causal):
     k = torch.randn(b, h, nk, 64, device='cuda').bfloat16()
     v = torch.randn(b, h, nk, 64, device='cuda').bfloat16()
     scale = 64 ** -0.5
-    out, lse = ext.fa_fwd(q, k, v, scale, causal, None, None)
+    out, lse = ext.fa_fwd(q, k, v, scale, causal, None, None, None, False)
     ref = fp32_oracle(q, k, v, scale, causal)
     err = (out.float() - ref).abs().max().item()
     assert err < 2e-2, f'max err {err}'
@@ -74,7 +74,7 @@ def test_fa_fwd_masks(ext):
     km[:, 0] = True
     sm = (torch.rand(n, n, device='cuda') > 0.2)
     sm.fill_diagonal_(True)
-    out, _ = ext.fa_fwd(q, k, v, 0.125, True, km, sm)
+    out, _ = ext.fa_fwd(q, k, v, 0.125, True, km, sm, None, False)
     ref = fp32_oracle(q, k, v, 0.125, True, km, sm)
     assert (out.float() - ref).abs().max().item() < 2e-2
 
@@ -106,6 +106,45 @@ def test_fa_backward_vs_oracle():
                             (v.grad, vr.grad, 'dv')):
         err = (got.float() - want).abs().max().item()
         rel = err / want.abs().max().item()
+        assert rel < 5e-2, f'{name} rel err {rel}'
+
+
+def test_fa_backward_masked_vs_oracle():
+    """Backward kernels with static mask + tile maps vs fp32 autograd."""
+    from dalle_pytorch_amd.ops import attention_core
+    from dalle_pytorch_amd.ops.attention import build_tile_map
+    torch.manual_seed(8)
+    b, h, n = 2, 2, 192
+    sm = torch.zeros(n, n, dtype=torch.bool, device='cuda')
+    sm[:, :33] = True
+    for r in range(0, n, 32):
+        sm[r:r + 32, r:r + 32] = True
+    tiles = build_tile_map(sm)
+    tiles_t = build_tile_map(sm.t())
+    km = torch.ones(b, n, dtype=torch.bool, device='cuda')
+    km[:, 10:20] = False
+
+    q0 = torch.randn(b, h, n, 64, device='cuda')
+    k0 = torch.randn(b, h, n, 64, device='cuda')
+    v0 = torch.randn(b, h, n, 64, device='cuda')
+    dout = torch.randn(b, h, n, 64, device='cuda')
+
+    q = q0.bfloat16().requires_grad_()
+    k = k0.bfloat16().requires_grad_()
+    v = v0.bfloat16().requires_grad_()
+    out = attention_core(q, k, v, 0.125, causal=True, key_mask=km,
+                         static_mask=sm, static_tiles=tiles,
+                         static_tiles_t=tiles_t)
+    out.backward(dout.bfloat16())
+
+    qr, kr, vr = (t.clone().requires_grad_() for t in (q0, k0, v0))
+    ref = fp32_oracle(qr, kr, vr, 0.125, True, km, sm)
+    ref = torch.nan_to_num(ref)   # fully-masked rows
+    ref.backward(torch.nan_to_num(dout))
+    for got, want, name in ((q.grad, qr.grad, 'dq'), (k.grad, kr.grad, 'dk'),
+                            (v.grad, vr.grad, 'dv')):
+        want = torch.nan_to_num(want)
+        rel = (got.float() - want).abs().max().item() / max(want.abs().max().item(), 1e-6)
         assert rel < 5e-2, f'{name} rel err {rel}'
 
 
@@ -149,7 +188,9 @@ def test_rope_split_vs_eager(ext):
     qe, ke, ve = (t.reshape(b, n, h, d).permute(0, 2, 1, 3) for t in parts)
     qe, ke, ve = apply_rotary_to_qkv(table, (qe, ke, ve))
     for got, want in ((q, qe), (k, ke), (v, ve)):
-        assert (got.float() - want.float()).abs().max().item() < 2e-2
+        err = (got.float() - want.float()).abs().max().item()
+        rel = err / want.float().abs().max().item()
+        assert rel < 2e-2, f'rope fwd rel err {rel}'
 
     # backward through the autograd wrapper
     from dalle_pytorch_amd.ops.rope import rope_split
