@@ -316,6 +316,11 @@ class DALLE(nn.Module):
             seq_len -= 1
             tokens = tokens[:, :-1]
 
+        # under bf16 autocast, carry the residual stream in bf16 end-to-end
+        # (embeddings emit fp32): halves every elementwise/norm/shift pass
+        if tokens.is_cuda and torch.is_autocast_enabled('cuda'):
+            tokens = tokens.to(torch.get_autocast_dtype('cuda'))
+
         if self.stable:
             alpha = 0.1
             tokens = tokens * alpha + tokens.detach() * (1 - alpha)

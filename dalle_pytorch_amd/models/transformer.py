@@ -21,6 +21,7 @@ from dalle_pytorch_amd.models.attention import (
     Attention, SparseAttention, SparseConvCausalAttention, SparseAxialCausalAttention)
 from dalle_pytorch_amd.models.positional import build_dalle_rotary_table
 from dalle_pytorch_amd.ops import geglu
+from dalle_pytorch_amd.ops.fused import token_shift, token_shift_supported
 
 
 def _as_tuple(val, depth=1):
@@ -92,7 +93,10 @@ class LayerScale(nn.Module):
         self.fn = fn
 
     def forward(self, x, **kwargs):
-        return self.fn(x, **kwargs) * self.scale
+        out = self.fn(x, **kwargs)
+        # multiply in the stream dtype: a fp32 scale would silently promote
+        # the bf16 residual stream (2x elementwise traffic downstream)
+        return out * self.scale.to(out.dtype)
 
 
 class PreNorm(nn.Module):
@@ -167,6 +171,11 @@ class PreShiftToken(nn.Module):
         n = x.shape[1]
         padding = seq_len - n + 1
         if n < text_len:
+            return self.fn(x, **kwargs)
+
+        if cache is None and token_shift_supported(x):
+            # fused gather kernel (no pad/cat/fill chain) — training hot path
+            x = token_shift(x, text_len, image_size)
             return self.fn(x, **kwargs)
 
         x_text, x_img = x[:, :text_len], x[:, text_len:]

@@ -35,3 +35,29 @@ def geglu(x: torch.Tensor) -> torch.Tensor:
         a, b = x.chunk(2, dim=-1)
         return a * F.gelu(b)
     return _GegluFn.apply(x)
+
+
+class _TokenShiftFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, text_len, image_size):
+        ext = hip_module()
+        ctx.text_len, ctx.image_size = text_len, image_size
+        return ext.token_shift(x.contiguous(), text_len, image_size, False)
+
+    @staticmethod
+    def backward(ctx, dout):
+        ext = hip_module()
+        dx = ext.token_shift(dout.contiguous(), ctx.text_len, ctx.image_size, True)
+        return dx, None, None
+
+
+def token_shift(x, text_len, image_size):
+    """Fused training-path token shift (reference transformer.py:165-186):
+    text halves shift one token; image quarters shift from the grid row
+    above / left neighbor. One gather kernel each way."""
+    return _TokenShiftFn.apply(x, text_len, image_size)
+
+
+def token_shift_supported(x):
+    return (x.is_cuda and hip_module() is not None
+            and x.shape[-1] * x.element_size() % 64 == 0)

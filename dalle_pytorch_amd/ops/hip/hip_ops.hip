@@ -653,6 +653,51 @@ DEVFN float gelu_grad_f(float x) {
   return cdf + x * pdf;
 }
 
+// bf16-specialized: short8 vector loads (guide G13 — hipcc does not
+// auto-vectorize scalar bf16 access; measured ~2x on memory-bound kernels)
+__global__ void geglu_fwd_bf16_kernel(const short* __restrict__ x,
+                                      short* __restrict__ out,
+                                      long rows, int H) {
+  const long row = blockIdx.y;
+  const int i0 = (blockIdx.x * blockDim.x + threadIdx.x) * 8;
+  if (row >= rows || i0 >= H) return;
+  int4v av = *reinterpret_cast<const int4v*>(x + row * (2L * H) + i0);
+  int4v gv = *reinterpret_cast<const int4v*>(x + row * (2L * H) + H + i0);
+  const short* a = reinterpret_cast<const short*>(&av);
+  const short* g = reinterpret_cast<const short*>(&gv);
+  short y[8];
+  #pragma unroll
+  for (int e = 0; e < 8; ++e) y[e] = f2bf(bf2f(a[e]) * gelu_f(bf2f(g[e])));
+  *reinterpret_cast<int4v*>(out + row * (long)H + i0) =
+      *reinterpret_cast<const int4v*>(y);
+}
+
+__global__ void geglu_bwd_bf16_kernel(const short* __restrict__ x,
+                                      const short* __restrict__ dout,
+                                      short* __restrict__ dx,
+                                      long rows, int H) {
+  const long row = blockIdx.y;
+  const int i0 = (blockIdx.x * blockDim.x + threadIdx.x) * 8;
+  if (row >= rows || i0 >= H) return;
+  int4v av = *reinterpret_cast<const int4v*>(x + row * (2L * H) + i0);
+  int4v gv = *reinterpret_cast<const int4v*>(x + row * (2L * H) + H + i0);
+  int4v dv = *reinterpret_cast<const int4v*>(dout + row * (long)H + i0);
+  const short* a = reinterpret_cast<const short*>(&av);
+  const short* g = reinterpret_cast<const short*>(&gv);
+  const short* d = reinterpret_cast<const short*>(&dv);
+  short da[8], dg[8];
+  #pragma unroll
+  for (int e = 0; e < 8; ++e) {
+    const float afv = bf2f(a[e]), gfv = bf2f(g[e]), dfv = bf2f(d[e]);
+    da[e] = f2bf(dfv * gelu_f(gfv));
+    dg[e] = f2bf(dfv * afv * gelu_grad_f(gfv));
+  }
+  *reinterpret_cast<int4v*>(dx + row * (2L * H) + i0) =
+      *reinterpret_cast<const int4v*>(da);
+  *reinterpret_cast<int4v*>(dx + row * (2L * H) + H + i0) =
+      *reinterpret_cast<const int4v*>(dg);
+}
+
 template <typename T>
 __global__ void geglu_fwd_kernel(const T* __restrict__ x, T* __restrict__ out,
                                  long rows, int H) {
@@ -687,6 +732,59 @@ __global__ void geglu_bwd_kernel(const T* __restrict__ x,
       dg[e] = T(dv * av * gelu_grad_f(gv));
     }
   }
+}
+
+// ---------------------------------------------------------------------------
+// Token shift (kernel K8, reference transformer.py:126-200 training path).
+//
+// Pure data movement, done in 16-byte chunks over the raw rows (dtype
+// agnostic): text positions take their first half-channels from the
+// previous token; image grid positions take quarter 0 from the row above
+// and quarter 1 from the left neighbor; everything else passes through.
+// backward=1 computes the transpose (each channel group has exactly one
+// destination, so the adjoint is another gather). Replaces ~8 eager
+// pad/cat/fill kernels per call.
+// ---------------------------------------------------------------------------
+
+__global__ void token_shift_kernel(
+    const char* __restrict__ src, char* __restrict__ dst,
+    int n, int row_bytes, int text_len, int S, int backward) {
+  const int cpr = row_bytes / 16;
+  const long c = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= (long)n * cpr) return;
+  const int pos = c / cpr;
+  const int off = (int)(c - (long)pos * cpr) * 16;
+  const long row0 = (long)blockIdx.y * n * row_bytes;
+
+  const int half = row_bytes / 2;
+  const int quarter = row_bytes / 4;
+
+  int srcpos = pos;
+  bool zero = false;
+  if (!backward) {
+    if (pos < text_len) {
+      if (off < half) { srcpos = pos - 1; zero = pos == 0; }
+    } else {
+      const int g = pos - text_len;
+      const int gr = g / S, gc = g - gr * S;
+      if (off < quarter) { srcpos = pos - S; zero = gr == 0; }
+      else if (off < half) { srcpos = pos - 1; zero = gc == 0; }
+    }
+  } else {
+    if (pos < text_len) {
+      if (off < half) { srcpos = pos + 1; zero = pos + 1 >= text_len; }
+    } else {
+      const int g = pos - text_len;
+      const int gr = g / S, gc = g - gr * S;
+      if (off < quarter) { srcpos = pos + S; zero = gr + 1 >= S || pos + S >= n; }
+      else if (off < half) { srcpos = pos + 1; zero = gc + 1 >= S || pos + 1 >= n; }
+    }
+  }
+
+  int4v val{0, 0, 0, 0};
+  if (!zero)
+    val = *reinterpret_cast<const int4v*>(src + row0 + (long)srcpos * row_bytes + off);
+  *reinterpret_cast<int4v*>(dst + row0 + (long)pos * row_bytes + off) = val;
 }
 
 // ---------------------------------------------------------------------------
@@ -887,6 +985,12 @@ torch::Tensor geglu_fwd(torch::Tensor x) {
   sizes.back() = H;
   auto out = torch::empty(sizes, x.options());
   dim3 grid((H + 256 * 8 - 1) / (256 * 8), rows);
+  if (x.dtype() == torch::kBFloat16 && H % 8 == 0) {
+    hipLaunchKernelGGL(geglu_fwd_bf16_kernel, grid, dim3(256), 0, cur_stream(),
+                       reinterpret_cast<const short*>(x.data_ptr()),
+                       reinterpret_cast<short*>(out.data_ptr()), rows, H);
+    return out;
+  }
   AT_DISPATCH_FLOATING_TYPES_AND2(at::ScalarType::BFloat16, at::ScalarType::Half,
                                   x.scalar_type(), "geglu_fwd", [&] {
     hipLaunchKernelGGL(geglu_fwd_kernel<scalar_t>, grid, dim3(256), 0, cur_stream(),
@@ -901,6 +1005,13 @@ torch::Tensor geglu_bwd(torch::Tensor x, torch::Tensor dout) {
   const long rows = x.numel() / (2L * H);
   auto dx = torch::empty_like(x);
   dim3 grid((H + 256 * 8 - 1) / (256 * 8), rows);
+  if (x.dtype() == torch::kBFloat16 && H % 8 == 0) {
+    hipLaunchKernelGGL(geglu_bwd_bf16_kernel, grid, dim3(256), 0, cur_stream(),
+                       reinterpret_cast<const short*>(x.data_ptr()),
+                       reinterpret_cast<const short*>(dout.data_ptr()),
+                       reinterpret_cast<short*>(dx.data_ptr()), rows, H);
+    return dx;
+  }
   AT_DISPATCH_FLOATING_TYPES_AND2(at::ScalarType::BFloat16, at::ScalarType::Half,
                                   x.scalar_type(), "geglu_bwd", [&] {
     hipLaunchKernelGGL(geglu_bwd_kernel<scalar_t>, grid, dim3(256), 0, cur_stream(),
@@ -908,6 +1019,23 @@ torch::Tensor geglu_bwd(torch::Tensor x, torch::Tensor dout) {
                        dx.data_ptr<scalar_t>(), rows, H);
   });
   return dx;
+}
+
+torch::Tensor token_shift(torch::Tensor x, int64_t text_len, int64_t image_size,
+                          bool backward) {
+  CHK(x.is_cuda() && x.is_contiguous() && x.dim() == 3);
+  const int b = x.size(0), n = x.size(1);
+  const long row_bytes = x.size(2) * x.element_size();
+  CHK(row_bytes % 64 == 0);   // 16B chunks must not cross quarter bounds
+  auto out = torch::empty_like(x);
+  const long chunks = (long)n * (row_bytes / 16);
+  dim3 grid((chunks + 255) / 256, b);
+  hipLaunchKernelGGL(token_shift_kernel, grid, dim3(256), 0, cur_stream(),
+                     reinterpret_cast<const char*>(x.data_ptr()),
+                     reinterpret_cast<char*>(out.data_ptr()),
+                     n, (int)row_bytes, (int)text_len, (int)image_size,
+                     backward ? 1 : 0);
+  return out;
 }
 
 torch::Tensor mfma_probe(torch::Tensor A, torch::Tensor B) {
@@ -928,6 +1056,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rope_split_fwd", &rope_split_fwd,
         "fused qkv split + rotary (q,k,v all rotated)");
   m.def("rope_split_bwd", &rope_split_bwd, "rope_split backward");
+  m.def("token_shift", &token_shift, "fused token shift (fwd/transpose)");
   m.def("geglu_fwd", &geglu_fwd, "fused GEGLU forward");
   m.def("geglu_bwd", &geglu_bwd, "fused GEGLU backward");
   m.def("mfma_probe", &mfma_probe, "MFMA 16x16x32 bf16 layout probe");

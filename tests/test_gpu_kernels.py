@@ -259,6 +259,33 @@ def test_autocast_rotary_stays_on_hip_path():
     assert called.get('yes'), 'rotary path fell back to eager attention'
 
 
+def test_token_shift_vs_eager(ext):
+    """Fused token shift == the eager pad/cat formulation (fwd + adjoint)."""
+    from dalle_pytorch_amd.models.transformer import PreShiftToken
+
+    class Capture(torch.nn.Module):
+        def forward(self, x, **kw):
+            return x
+
+    S, text_len = 8, 5
+    seq_len = text_len + S * S - 1
+    shift = PreShiftToken(Capture(), image_size=S, seq_len=seq_len)
+    torch.manual_seed(9)
+    for n in (seq_len, seq_len - 7):
+        x0 = torch.randn(2, n, 64, device='cuda')
+        # eager reference on CPU path semantics (run the module on fp32 CPU)
+        ref = shift(x0.cpu()).cuda()
+        got = ext.token_shift(x0.contiguous(), text_len, S, False)
+        assert torch.allclose(got, ref, atol=1e-6), n
+
+    # adjoint check: <shift(x), y> == <x, shift^T(y)>
+    x = torch.randn(2, seq_len, 64, device='cuda')
+    y = torch.randn(2, seq_len, 64, device='cuda')
+    lhs = (ext.token_shift(x, text_len, S, False) * y).sum()
+    rhs = (x * ext.token_shift(y, text_len, S, True)).sum()
+    assert torch.allclose(lhs, rhs, rtol=1e-4)
+
+
 def test_geglu_vs_oracle(ext):
     torch.manual_seed(4)
     x0 = torch.randn(4, 96, 512, device='cuda')
