@@ -80,9 +80,13 @@ void fa_fwd_d64_kernel(
     int b, int h, int nq, int nk,
     float scale, int causal, int out_bnhd) {
 
-  __shared__ short Kt[FA_KBLK][KPAD];
-  __shared__ short Vt[FA_D][VPAD];
-  __shared__ short Pl[FA_WAVES][16][PPAD];
+  // KV tiles are 64 keys (2 map granules); staging is software-pipelined:
+  // the next live tile's global loads are issued before this tile's MFMA
+  // work so HBM latency hides under the compute (guide G15 async-STAGE).
+  constexpr int KV = 2 * FA_KBLK;             // 64 keys per LDS tile
+  __shared__ short Kt[KV][KPAD];
+  __shared__ short Vt[FA_D][KV + 8];
+  __shared__ short Pl[FA_WAVES][16][KV + 8];
 
   const int bh = blockIdx.y;
   const int batch = bh / h;
@@ -115,48 +119,80 @@ void fa_fwd_d64_kernel(
 
   float m_run = NEG_INF;   // per-q-row online-softmax state (q = lq)
   float l_run = 0.f;
-  // PV accumulators: acc[nt] covers d cols [16nt, 16nt+16); C layout rows
-  // are q = grp*4 + r of the wave tile
   f32x4 acc[4] = {f32x4{0, 0, 0, 0}, f32x4{0, 0, 0, 0},
                   f32x4{0, 0, 0, 0}, f32x4{0, 0, 0, 0}};
 
-  int ntiles = (nk + FA_KBLK - 1) / FA_KBLK;
-  const int ntk = ntiles;
+  const int ntk = (nk + FA_KBLK - 1) / FA_KBLK;   // 32-key map granules
+  int ntiles = (nk + KV - 1) / KV;                // 64-key compute tiles
   if (causal) {
     const int lim = min(nk - 1, q0 + FA_QBLK - 1 + diag);
-    ntiles = lim < 0 ? 0 : (lim / FA_KBLK + 1);
+    ntiles = lim < 0 ? 0 : (lim / KV + 1);
   }
   const unsigned char* tmap_row =
       tile_map ? tile_map + (long)blockIdx.x * ntk : nullptr;
 
-  for (int kt = 0; kt < ntiles; ++kt) {
-    // block-sparse skip: the host precomputes, per (64-q, 32-k) tile, whether
-    // any static-mask entry is set (serves axial/conv/block-sparse patterns)
-    if (tmap_row && !tmap_row[kt]) continue;
-    const int kbase = kt * FA_KBLK;
+  // block-sparse skip: the host precomputes, per (64q, 32k) granule,
+  // whether any static-mask entry is set; a 64-key tile is live if either
+  // of its granules is (axial/conv/block-sparse patterns)
+  auto tile_live = [&](int t) -> bool {
+    if (!tmap_row) return true;
+    const int g0 = 2 * t;
+    bool live = tmap_row[g0] != 0;
+    if (g0 + 1 < ntk) live |= tmap_row[g0 + 1] != 0;
+    return live;
+  };
+  auto next_live = [&](int t) -> int {
+    while (t < ntiles && !tile_live(t)) ++t;
+    return t;
+  };
 
-    // ---- stage K (row-major, padded) and V (transposed) tiles
-    {
-      const int row = tid >> 3;       // 0..31
-      const int c8 = (tid & 7) * 8;   // 0,8,...,56
-      const int kg = kbase + row;
-      int4v kv{0, 0, 0, 0}, vv{0, 0, 0, 0};
+  // staging geometry: 64 rows x 64 cols = 512 16B chunks, 2 per thread
+  const int srow0 = tid >> 3;           // chunk-0 row (0..31)
+  const int sc8 = (tid & 7) * 8;        // chunk col
+  int4v kreg[2], vreg[2];
+
+  auto prefetch = [&](int t) {
+    const int kbase = t * KV;
+    #pragma unroll
+    for (int half = 0; half < 2; ++half) {
+      const int kg = kbase + srow0 + 32 * half;
       if (kg < nk) {
-        kv = *reinterpret_cast<const int4v*>(kp + (long)kg * FA_D + c8);
-        vv = *reinterpret_cast<const int4v*>(vp + (long)kg * FA_D + c8);
+        kreg[half] = *reinterpret_cast<const int4v*>(kp + (long)kg * FA_D + sc8);
+        vreg[half] = *reinterpret_cast<const int4v*>(vp + (long)kg * FA_D + sc8);
+      } else {
+        kreg[half] = int4v{0, 0, 0, 0};
+        vreg[half] = int4v{0, 0, 0, 0};
       }
-      __syncthreads();   // prior tile fully consumed before overwrite
-      *reinterpret_cast<int4v*>(&Kt[row][c8]) = kv;
-      const short* vs = reinterpret_cast<const short*>(&vv);
+    }
+  };
+
+  int kt = next_live(0);
+  if (kt < ntiles) prefetch(kt);
+
+  while (kt < ntiles) {
+    const int kbase = kt * KV;
+    const int kt_next = next_live(kt + 1);
+
+    __syncthreads();   // prior tile fully consumed before overwrite
+    #pragma unroll
+    for (int half = 0; half < 2; ++half) {
+      const int row = srow0 + 32 * half;
+      *reinterpret_cast<int4v*>(&Kt[row][sc8]) = kreg[half];
+      const short* vs = reinterpret_cast<const short*>(&vreg[half]);
       #pragma unroll
-      for (int e = 0; e < 8; ++e) Vt[c8 + e][row] = vs[e];
+      for (int e = 0; e < 8; ++e) Vt[sc8 + e][row] = vs[e];
     }
     __syncthreads();
 
-    // ---- swapped QK^T: two 16-key subtiles, contraction K=64 in 2 steps
-    float s8[8];
+    // issue the NEXT tile's loads now: HBM latency hides under the MFMA +
+    // softmax work below
+    if (kt_next < ntiles) prefetch(kt_next);
+
+    // ---- swapped QK^T: four 16-key subtiles, contraction K=64 in 2 steps
+    float s16[16];
+    __builtin_amdgcn_s_setprio(1);
     #pragma unroll
-    for (int mt = 0; mt < 2; ++mt) {
+    for (int mt = 0; mt < 4; ++mt) {
       f32x4 st{0, 0, 0, 0};
       #pragma unroll
       for (int c = 0; c < 2; ++c) {
@@ -164,67 +200,84 @@ void fa_fwd_d64_kernel(
         st = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, qfrag[c], st, 0, 0, 0);
       }
       #pragma unroll
-      for (int r = 0; r < 4; ++r) s8[mt * 4 + r] = st[r];
+      for (int r = 0; r < 4; ++r) s16[mt * 4 + r] = st[r];
     }
+    __builtin_amdgcn_s_setprio(0);
     // After the swap: C col = lane&15 = q row; C row = grp*4 + r = key.
-    // s8[i] = S[qrow][kbase + (i>>2)*16 + grp*4 + (i&3)].
+    // s16[i] = S[qrow][kbase + (i>>2)*16 + grp*4 + (i&3)].
 
     // ---- scale + masks
     #pragma unroll
-    for (int i = 0; i < 8; ++i) {
+    for (int i = 0; i < 16; ++i) {
       const int kg = kbase + (i >> 2) * 16 + grp * 4 + (i & 3);
       bool ok = (kg < nk) & (qrow < nq);
       if (causal) ok &= kg <= qrow + diag;
       if (key_mask != nullptr && ok) ok &= key_mask[(long)batch * nk + kg];
       if (static_mask != nullptr && ok) ok &= static_mask[(long)qrow * nk + kg];
-      s8[i] = ok ? s8[i] * scale : NEG_INF;
+      s16[i] = ok ? s16[i] * scale : NEG_INF;
     }
 
     // ---- online softmax update (lane-local + 4-lane reduce per q row)
     float mt_part = NEG_INF;
     #pragma unroll
-    for (int i = 0; i < 8; ++i) mt_part = fmaxf(mt_part, s8[i]);
+    for (int i = 0; i < 16; ++i) mt_part = fmaxf(mt_part, s16[i]);
     mt_part = fmaxf(mt_part, __shfl_xor(mt_part, 16));
     mt_part = fmaxf(mt_part, __shfl_xor(mt_part, 32));
 
-    const float m_new = fmaxf(m_run, mt_part);
+    // defer-max (guide T13): if no row's max grew by more than 8, keep the
+    // old running max — P stays bounded by e^8 and the O-rescale is skipped
+    const bool defer = __all(mt_part <= m_run + 8.f);
+    const float m_new = defer ? m_run : fmaxf(m_run, mt_part);
+
     float lsum = 0.f;
-    float p8[8];
+    float p16[16];
     #pragma unroll
-    for (int i = 0; i < 8; ++i) {
-      p8[i] = (s8[i] == NEG_INF) ? 0.f : __expf(s8[i] - m_new);
-      lsum += p8[i];
+    for (int i = 0; i < 16; ++i) {
+      p16[i] = (s16[i] == NEG_INF) ? 0.f : __expf(s16[i] - m_new);
+      lsum += p16[i];
     }
     lsum += __shfl_xor(lsum, 16);
     lsum += __shfl_xor(lsum, 32);
 
-    const float alpha = (m_run == NEG_INF) ? 0.f : __expf(m_run - m_new);
+    float alpha = 1.f;
+    if (!defer) {
+      alpha = (m_run == NEG_INF) ? 0.f : __expf(m_run - m_new);
+      if (m_new != NEG_INF) m_run = m_new;
+    }
     l_run = l_run * alpha + lsum;
-    if (m_new != NEG_INF) m_run = m_new;
 
     // ---- P -> LDS (bf16) to reshape into the PV A-fragment
     #pragma unroll
-    for (int i = 0; i < 8; ++i) {
+    for (int i = 0; i < 16; ++i) {
       const int kk = (i >> 2) * 16 + grp * 4 + (i & 3);
-      Pl[wave][lq][kk] = f2bf(p8[i]);
+      Pl[wave][lq][kk] = f2bf(p16[i]);
     }
 
-    // ---- rescale accumulators; acc rows are q = grp*4 + r, alphas live on
-    // lanes whose (lane&15) equals that q row
-    #pragma unroll
-    for (int r = 0; r < 4; ++r) {
-      const float a_r = __shfl(alpha, grp * 4 + r);
+    if (!defer) {
+      // rescale accumulators; acc rows are q = grp*4 + r, alphas live on
+      // lanes whose (lane&15) equals that q row
       #pragma unroll
-      for (int nt = 0; nt < 4; ++nt) acc[nt][r] *= a_r;
+      for (int r = 0; r < 4; ++r) {
+        const float a_r = __shfl(alpha, grp * 4 + r);
+        #pragma unroll
+        for (int nt = 0; nt < 4; ++nt) acc[nt][r] *= a_r;
+      }
     }
 
-    // ---- PV (compiler inserts the lgkmcnt wait for the same-wave Pl trip)
-    bf16x8 pf = frag_from_lds(&Pl[wave][lq][8 * grp]);
+    // ---- PV: contraction over this tile's 64 keys in two 32-key steps
+    bf16x8 pf0 = frag_from_lds(&Pl[wave][lq][8 * grp]);
+    bf16x8 pf1 = frag_from_lds(&Pl[wave][lq][32 + 8 * grp]);
+    __builtin_amdgcn_s_setprio(1);
     #pragma unroll
     for (int nt = 0; nt < 4; ++nt) {
-      bf16x8 vf = frag_from_lds(&Vt[lq + 16 * nt][8 * grp]);
-      acc[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pf, vf, acc[nt], 0, 0, 0);
+      bf16x8 vf0 = frag_from_lds(&Vt[lq + 16 * nt][8 * grp]);
+      bf16x8 vf1 = frag_from_lds(&Vt[lq + 16 * nt][32 + 8 * grp]);
+      acc[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pf0, vf0, acc[nt], 0, 0, 0);
+      acc[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pf1, vf1, acc[nt], 0, 0, 0);
     }
+    __builtin_amdgcn_s_setprio(0);
+
+    kt = kt_next;
   }
 
   // ---- epilogue: divide by l per q row, store out + lse. out_bnhd writes
@@ -279,10 +332,11 @@ void fa_bwd_dq_kernel(
     const unsigned char* __restrict__ tile_map,   // [nq/64, nk/32]
     int b, int h, int nq, int nk, float scale, int causal) {
 
-  __shared__ short Kt[FA_KBLK][KPAD];    // K row-major
-  __shared__ short Vr[FA_KBLK][KPAD];    // V row-major (A-operand of dP^T)
-  __shared__ short Ktr[FA_D][VPAD];      // K transposed (B-operand of dS*K)
-  __shared__ short DSl[FA_WAVES][16][PPAD];
+  constexpr int KV = 2 * FA_KBLK;          // 64 keys per LDS tile
+  __shared__ short Kt[KV][KPAD];           // K row-major
+  __shared__ short Vr[KV][KPAD];           // V row-major (A-operand of dP^T)
+  __shared__ short Ktr[FA_D][KV + 8];      // K transposed (B-operand of dS*K)
+  __shared__ short DSl[FA_WAVES][16][KV + 8];
 
   const int bh = blockIdx.y;
   const int batch = bh / h;
@@ -318,40 +372,69 @@ void fa_bwd_dq_kernel(
   f32x4 acc[4] = {f32x4{0, 0, 0, 0}, f32x4{0, 0, 0, 0},
                   f32x4{0, 0, 0, 0}, f32x4{0, 0, 0, 0}};
 
-  int ntiles = (nk + FA_KBLK - 1) / FA_KBLK;
-  const int ntk = ntiles;
+  const int ntk = (nk + FA_KBLK - 1) / FA_KBLK;
+  int ntiles = (nk + KV - 1) / KV;
   if (causal) {
     const int lim = min(nk - 1, q0 + FA_QBLK - 1 + diag);
-    ntiles = lim < 0 ? 0 : (lim / FA_KBLK + 1);
+    ntiles = lim < 0 ? 0 : (lim / KV + 1);
   }
   const unsigned char* tmap_row =
       tile_map ? tile_map + (long)blockIdx.x * ntk : nullptr;
 
-  for (int kt = 0; kt < ntiles; ++kt) {
-    if (tmap_row && !tmap_row[kt]) continue;
-    const int kbase = kt * FA_KBLK;
-    {
-      const int row = tid >> 3;
-      const int c8 = (tid & 7) * 8;
-      const int kg = kbase + row;
-      int4v kv{0, 0, 0, 0}, vv{0, 0, 0, 0};
+  auto tile_live = [&](int t) -> bool {
+    if (!tmap_row) return true;
+    const int g0 = 2 * t;
+    bool live = tmap_row[g0] != 0;
+    if (g0 + 1 < ntk) live |= tmap_row[g0 + 1] != 0;
+    return live;
+  };
+  auto next_live = [&](int t) -> int {
+    while (t < ntiles && !tile_live(t)) ++t;
+    return t;
+  };
+
+  const int srow0 = tid >> 3;
+  const int sc8 = (tid & 7) * 8;
+  int4v kreg[2], vreg[2];
+  auto prefetch = [&](int t) {
+    #pragma unroll
+    for (int half = 0; half < 2; ++half) {
+      const int kg = t * KV + srow0 + 32 * half;
       if (kg < nk) {
-        kv = *reinterpret_cast<const int4v*>(kp + (long)kg * FA_D + c8);
-        vv = *reinterpret_cast<const int4v*>(vp + (long)kg * FA_D + c8);
+        kreg[half] = *reinterpret_cast<const int4v*>(kp + (long)kg * FA_D + sc8);
+        vreg[half] = *reinterpret_cast<const int4v*>(vp + (long)kg * FA_D + sc8);
+      } else {
+        kreg[half] = int4v{0, 0, 0, 0};
+        vreg[half] = int4v{0, 0, 0, 0};
       }
-      __syncthreads();
-      *reinterpret_cast<int4v*>(&Kt[row][c8]) = kv;
-      *reinterpret_cast<int4v*>(&Vr[row][c8]) = vv;
-      const short* ks = reinterpret_cast<const short*>(&kv);
+    }
+  };
+
+  int kt = next_live(0);
+  if (kt < ntiles) prefetch(kt);
+
+  while (kt < ntiles) {
+    const int kbase = kt * KV;
+    const int kt_next = next_live(kt + 1);
+
+    __syncthreads();
+    #pragma unroll
+    for (int half = 0; half < 2; ++half) {
+      const int row = srow0 + 32 * half;
+      *reinterpret_cast<int4v*>(&Kt[row][sc8]) = kreg[half];
+      *reinterpret_cast<int4v*>(&Vr[row][sc8]) = vreg[half];
+      const short* ks = reinterpret_cast<const short*>(&kreg[half]);
       #pragma unroll
-      for (int e = 0; e < 8; ++e) Ktr[c8 + e][row] = ks[e];
+      for (int e = 0; e < 8; ++e) Ktr[sc8 + e][row] = ks[e];
     }
     __syncthreads();
+    if (kt_next < ntiles) prefetch(kt_next);
 
     // s^T and dp^T, swapped layout: lane -> q = lq, keys = grp*4+r (+16mt)
-    float s8[8], dp8[8];
+    float s16[16], dp16[16];
+    __builtin_amdgcn_s_setprio(1);
     #pragma unroll
-    for (int mt = 0; mt < 2; ++mt) {
+    for (int mt = 0; mt < 4; ++mt) {
       f32x4 st{0, 0, 0, 0}, dpt{0, 0, 0, 0};
       #pragma unroll
       for (int c = 0; c < 2; ++c) {
@@ -361,28 +444,36 @@ void fa_bwd_dq_kernel(
         dpt = __builtin_amdgcn_mfma_f32_16x16x32_bf16(vf, dofrag[c], dpt, 0, 0, 0);
       }
       #pragma unroll
-      for (int r = 0; r < 4; ++r) { s8[mt * 4 + r] = st[r]; dp8[mt * 4 + r] = dpt[r]; }
+      for (int r = 0; r < 4; ++r) { s16[mt * 4 + r] = st[r]; dp16[mt * 4 + r] = dpt[r]; }
     }
+    __builtin_amdgcn_s_setprio(0);
 
     #pragma unroll
-    for (int i = 0; i < 8; ++i) {
+    for (int i = 0; i < 16; ++i) {
       const int kg = kbase + (i >> 2) * 16 + grp * 4 + (i & 3);
       bool ok = (kg < nk) & (qrow < nq);
       if (causal) ok &= kg <= qrow + diag;
       if (key_mask != nullptr && ok) ok &= key_mask[(long)batch * nk + kg];
       if (static_mask != nullptr && ok) ok &= static_mask[(long)qrow * nk + kg];
-      const float p = ok ? __expf(s8[i] * scale - lse_q) : 0.f;
-      const float ds = p * (dp8[i] - D_q) * scale;
+      const float p = ok ? __expf(s16[i] * scale - lse_q) : 0.f;
+      const float ds = p * (dp16[i] - D_q) * scale;
       const int kk = (i >> 2) * 16 + grp * 4 + (i & 3);
       DSl[wave][lq][kk] = f2bf(ds);
     }
 
-    bf16x8 dsf = frag_from_lds(&DSl[wave][lq][8 * grp]);
+    bf16x8 dsf0 = frag_from_lds(&DSl[wave][lq][8 * grp]);
+    bf16x8 dsf1 = frag_from_lds(&DSl[wave][lq][32 + 8 * grp]);
+    __builtin_amdgcn_s_setprio(1);
     #pragma unroll
     for (int nt = 0; nt < 4; ++nt) {
-      bf16x8 kf = frag_from_lds(&Ktr[lq + 16 * nt][8 * grp]);
-      acc[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(dsf, kf, acc[nt], 0, 0, 0);
+      bf16x8 kf0 = frag_from_lds(&Ktr[lq + 16 * nt][8 * grp]);
+      bf16x8 kf1 = frag_from_lds(&Ktr[lq + 16 * nt][32 + 8 * grp]);
+      acc[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(dsf0, kf0, acc[nt], 0, 0, 0);
+      acc[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(dsf1, kf1, acc[nt], 0, 0, 0);
     }
+    __builtin_amdgcn_s_setprio(0);
+
+    kt = kt_next;
   }
 
   #pragma unroll
@@ -412,12 +503,13 @@ void fa_bwd_dkv_kernel(
     const unsigned char* __restrict__ tile_map_t,  // [nk/64, nq/32]
     int b, int h, int nq, int nk, float scale, int causal) {
 
-  __shared__ short Qr[FA_KBLK][KPAD];     // Q rows (B-operand of s^T)
-  __shared__ short dOr[FA_KBLK][KPAD];    // dO rows (B-operand of dp^T)
-  __shared__ short Qtr[FA_D][VPAD];       // Q transposed (dK = dS^T Q)
-  __shared__ short dOtr[FA_D][VPAD];      // dO transposed (dV = P^T dO)
-  __shared__ short Pt[FA_WAVES][16][PPAD];
-  __shared__ short DSt[FA_WAVES][16][PPAD];
+  constexpr int KV = 2 * FA_KBLK;          // 64 q rows per LDS tile
+  __shared__ short Qr[KV][KPAD];           // Q rows (B-operand of s^T)
+  __shared__ short dOr[KV][KPAD];          // dO rows (B-operand of dp^T)
+  __shared__ short Qtr[FA_D][KV + 8];      // Q transposed (dK = dS^T Q)
+  __shared__ short dOtr[FA_D][KV + 8];     // dO transposed (dV = P^T dO)
+  __shared__ short Pt[FA_WAVES][16][KV + 8];
+  __shared__ short DSt[FA_WAVES][16][KV + 8];
 
   const int bh = blockIdx.y;
   const int batch = bh / h;
@@ -450,46 +542,76 @@ void fa_bwd_dkv_kernel(
   f32x4 acc_dv[4] = {f32x4{0, 0, 0, 0}, f32x4{0, 0, 0, 0},
                      f32x4{0, 0, 0, 0}, f32x4{0, 0, 0, 0}};
 
-  const int nqt = (nq + FA_KBLK - 1) / FA_KBLK;   // q tiles of 32
+  const int nqg = (nq + FA_KBLK - 1) / FA_KBLK;   // 32-row map granules
+  const int nqt = (nq + KV - 1) / KV;             // 64-row compute tiles
   int qt_start = 0;
   if (causal) {
-    // smallest q that can see any key in this block: q >= k - diag
-    const int qmin = k0 - diag;
-    qt_start = qmin <= 0 ? 0 : qmin / FA_KBLK;
+    const int qmin = k0 - diag;   // smallest q that sees any key here
+    qt_start = qmin <= 0 ? 0 : qmin / KV;
   }
   const unsigned char* tmap_row =
-      tile_map_t ? tile_map_t + (long)blockIdx.x * nqt : nullptr;
+      tile_map_t ? tile_map_t + (long)blockIdx.x * nqg : nullptr;
 
-  for (int qt = qt_start; qt < nqt; ++qt) {
-    if (tmap_row && !tmap_row[qt]) continue;
-    const int qbase = qt * FA_KBLK;
-    {
-      const int row = tid >> 3;
-      const int c8 = (tid & 7) * 8;
-      const int qg = qbase + row;
-      int4v qv{0, 0, 0, 0}, dv16{0, 0, 0, 0};
+  auto tile_live = [&](int t) -> bool {
+    if (!tmap_row) return true;
+    const int g0 = 2 * t;
+    bool live = tmap_row[g0] != 0;
+    if (g0 + 1 < nqg) live |= tmap_row[g0 + 1] != 0;
+    return live;
+  };
+  auto next_live = [&](int t) -> int {
+    while (t < nqt && !tile_live(t)) ++t;
+    return t;
+  };
+
+  const int srow0 = tid >> 3;
+  const int sc8 = (tid & 7) * 8;
+  int4v qreg[2], doreg[2];
+  auto prefetch = [&](int t) {
+    #pragma unroll
+    for (int half = 0; half < 2; ++half) {
+      const int qg = t * KV + srow0 + 32 * half;
       if (qg < nq) {
-        qv = *reinterpret_cast<const int4v*>(qp + (long)qg * FA_D + c8);
-        dv16 = *reinterpret_cast<const int4v*>(dop + (long)qg * FA_D + c8);
+        qreg[half] = *reinterpret_cast<const int4v*>(qp + (long)qg * FA_D + sc8);
+        doreg[half] = *reinterpret_cast<const int4v*>(dop + (long)qg * FA_D + sc8);
+      } else {
+        qreg[half] = int4v{0, 0, 0, 0};
+        doreg[half] = int4v{0, 0, 0, 0};
       }
-      __syncthreads();
-      *reinterpret_cast<int4v*>(&Qr[row][c8]) = qv;
-      *reinterpret_cast<int4v*>(&dOr[row][c8]) = dv16;
-      const short* qs = reinterpret_cast<const short*>(&qv);
-      const short* ds_ = reinterpret_cast<const short*>(&dv16);
+    }
+  };
+
+  int qt = next_live(qt_start);
+  if (qt < nqt) prefetch(qt);
+
+  while (qt < nqt) {
+    const int qbase = qt * KV;
+    const int qt_next = next_live(qt + 1);
+
+    __syncthreads();
+    #pragma unroll
+    for (int half = 0; half < 2; ++half) {
+      const int row = srow0 + 32 * half;
+      *reinterpret_cast<int4v*>(&Qr[row][sc8]) = qreg[half];
+      *reinterpret_cast<int4v*>(&dOr[row][sc8]) = doreg[half];
+      const short* qs = reinterpret_cast<const short*>(&qreg[half]);
+      const short* ds_ = reinterpret_cast<const short*>(&doreg[half]);
       #pragma unroll
       for (int e = 0; e < 8; ++e) {
-        Qtr[c8 + e][row] = qs[e];
-        dOtr[c8 + e][row] = ds_[e];
+        Qtr[sc8 + e][row] = qs[e];
+        dOtr[sc8 + e][row] = ds_[e];
       }
     }
     __syncthreads();
+    if (qt_next < nqt) prefetch(qt_next);
 
-    // s^T[key, q], dp^T[key, q]: lane -> key rows grp*4+r (of the wave's
-    // 16), q col = lq (+16mt). Note roles vs the dq kernel are mirrored:
-    // here the A-operand is the wave's OWN key fragment and q comes from LDS.
+    // s^T[key, q], dp^T[key, q]: A = the wave's own key/value fragments,
+    // B = Q/dO columns from LDS. Output [M=16 keys, N=16 q]: C row
+    // grp*4+r = key within the wave's 16, C col = lane&15 = q in subtile.
+    __builtin_amdgcn_s_setprio(1);
+    f32x4 st4[4], dpt4[4];
     #pragma unroll
-    for (int mt = 0; mt < 2; ++mt) {
+    for (int mt = 0; mt < 4; ++mt) {
       f32x4 st{0, 0, 0, 0}, dpt{0, 0, 0, 0};
       #pragma unroll
       for (int c = 0; c < 2; ++c) {
@@ -498,8 +620,13 @@ void fa_bwd_dkv_kernel(
         st = __builtin_amdgcn_mfma_f32_16x16x32_bf16(kfrag[c], qf, st, 0, 0, 0);
         dpt = __builtin_amdgcn_mfma_f32_16x16x32_bf16(vfrag[c], dof, dpt, 0, 0, 0);
       }
-      // output tile is [M=16 keys, N=16 q]: C row grp*4+r = key within the
-      // wave's 16, C col = lane&15 = q within the subtile
+      st4[mt] = st;
+      dpt4[mt] = dpt;
+    }
+    __builtin_amdgcn_s_setprio(0);
+
+    #pragma unroll
+    for (int mt = 0; mt < 4; ++mt) {
       #pragma unroll
       for (int r = 0; r < 4; ++r) {
         const int key = k0 + wave * 16 + grp * 4 + r;
@@ -512,23 +639,33 @@ void fa_bwd_dkv_kernel(
         if (ok) {
           const float l = lse[(long)bh * nq + qg];
           const float Dq = Dv[(long)bh * nq + qg];
-          p = __expf(st[r] * scale - l);
-          ds = p * (dpt[r] - Dq) * scale;
+          p = __expf(st4[mt][r] * scale - l);
+          ds = p * (dpt4[mt][r] - Dq) * scale;
         }
         Pt[wave][grp * 4 + r][mt * 16 + lq] = f2bf(p);
         DSt[wave][grp * 4 + r][mt * 16 + lq] = f2bf(ds);
       }
     }
 
-    bf16x8 pf = frag_from_lds(&Pt[wave][lq][8 * grp]);
-    bf16x8 dsf = frag_from_lds(&DSt[wave][lq][8 * grp]);
+    bf16x8 pf0 = frag_from_lds(&Pt[wave][lq][8 * grp]);
+    bf16x8 pf1 = frag_from_lds(&Pt[wave][lq][32 + 8 * grp]);
+    bf16x8 dsf0 = frag_from_lds(&DSt[wave][lq][8 * grp]);
+    bf16x8 dsf1 = frag_from_lds(&DSt[wave][lq][32 + 8 * grp]);
+    __builtin_amdgcn_s_setprio(1);
     #pragma unroll
     for (int nt = 0; nt < 4; ++nt) {
-      bf16x8 dof = frag_from_lds(&dOtr[lq + 16 * nt][8 * grp]);
-      bf16x8 qf = frag_from_lds(&Qtr[lq + 16 * nt][8 * grp]);
-      acc_dv[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pf, dof, acc_dv[nt], 0, 0, 0);
-      acc_dk[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(dsf, qf, acc_dk[nt], 0, 0, 0);
+      bf16x8 dof0 = frag_from_lds(&dOtr[lq + 16 * nt][8 * grp]);
+      bf16x8 dof1 = frag_from_lds(&dOtr[lq + 16 * nt][32 + 8 * grp]);
+      bf16x8 qf0 = frag_from_lds(&Qtr[lq + 16 * nt][8 * grp]);
+      bf16x8 qf1 = frag_from_lds(&Qtr[lq + 16 * nt][32 + 8 * grp]);
+      acc_dv[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pf0, dof0, acc_dv[nt], 0, 0, 0);
+      acc_dv[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pf1, dof1, acc_dv[nt], 0, 0, 0);
+      acc_dk[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(dsf0, qf0, acc_dk[nt], 0, 0, 0);
+      acc_dk[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(dsf1, qf1, acc_dk[nt], 0, 0, 0);
     }
+    __builtin_amdgcn_s_setprio(0);
+
+    qt = qt_next;
   }
 
   #pragma unroll
