@@ -110,16 +110,18 @@ class _FlashAttention(torch.autograd.Function):
     @staticmethod
     def backward(ctx, dout):
         q, k, v, out, lse = ctx.saved_tensors
-        if ctx.fold_heads:   # [b, n, h, d] -> [b, h, n, d]
-            out = out.permute(0, 2, 1, 3)
-            dout = dout.permute(0, 2, 1, 3)
         ext = hip_module()
         if hasattr(ext, 'fa_bwd'):
+            # with fold_heads, out/dout stay in their contiguous [b,n,h,d]
+            # layout — the bwd kernels read them with bnhd strides directly
             dq, dk, dv = ext.fa_bwd(
-                q, k, v, out.contiguous(), lse, dout.contiguous(),
+                q, k, v, out, lse, dout.contiguous(),
                 ctx.scale, ctx.causal, ctx.key_mask, ctx.static_mask,
-                ctx.tile_map, ctx.tile_map_t)
+                ctx.tile_map, ctx.tile_map_t, ctx.fold_heads)
         else:
+            if ctx.fold_heads:
+                out = out.permute(0, 2, 1, 3)
+                dout = dout.permute(0, 2, 1, 3)
             dq, dk, dv = _flash_bwd_composite(
                 q, k, v, out, lse, dout.contiguous(), ctx.scale, ctx.causal,
                 ctx.key_mask, ctx.static_mask)

@@ -87,6 +87,7 @@ void fa_fwd_d64_kernel(
   __shared__ short Kt[KV][KPAD];
   __shared__ short Vt[FA_D][KV + 8];
   __shared__ short Pl[FA_WAVES][16][KV + 8];
+  __shared__ unsigned char Mtile[FA_QBLK][KV];   // static-mask tile
 
   const int bh = blockIdx.y;
   const int batch = bh / h;
@@ -149,7 +150,10 @@ void fa_fwd_d64_kernel(
   // staging geometry: 64 rows x 64 cols = 512 16B chunks, 2 per thread
   const int srow0 = tid >> 3;           // chunk-0 row (0..31)
   const int sc8 = (tid & 7) * 8;        // chunk col
-  int4v kreg[2], vreg[2];
+  int4v kreg[2], vreg[2], mreg;
+  // mask staging: 64x64 bytes = one 16B chunk per thread
+  const int mrow = tid >> 2;            // 0..63
+  const int mc16 = (tid & 3) * 16;
 
   auto prefetch = [&](int t) {
     const int kbase = t * KV;
@@ -162,6 +166,23 @@ void fa_fwd_d64_kernel(
       } else {
         kreg[half] = int4v{0, 0, 0, 0};
         vreg[half] = int4v{0, 0, 0, 0};
+      }
+    }
+    if (static_mask != nullptr) {
+      const int mq = q0 + mrow;
+      const long base = (long)mq * nk + kbase + mc16;
+      if (mq < nq && kbase + mc16 + 16 <= nk && (base & 15) == 0) {
+        mreg = *reinterpret_cast<const int4v*>(
+            reinterpret_cast<const char*>(static_mask) + base);
+      } else {
+        unsigned char mb[16];
+        #pragma unroll
+        for (int e = 0; e < 16; ++e) {
+          const int kg = kbase + mc16 + e;
+          mb[e] = (mq < nq && kg < nk)
+              ? (unsigned char)static_mask[(long)mq * nk + kg] : 0;
+        }
+        mreg = *reinterpret_cast<const int4v*>(mb);
       }
     }
   };
@@ -182,6 +203,8 @@ void fa_fwd_d64_kernel(
       #pragma unroll
       for (int e = 0; e < 8; ++e) Vt[sc8 + e][row] = vs[e];
     }
+    if (static_mask != nullptr)
+      *reinterpret_cast<int4v*>(&Mtile[mrow][mc16]) = mreg;
     __syncthreads();
 
     // issue the NEXT tile's loads now: HBM latency hides under the MFMA +
@@ -213,7 +236,8 @@ void fa_fwd_d64_kernel(
       bool ok = (kg < nk) & (qrow < nq);
       if (causal) ok &= kg <= qrow + diag;
       if (key_mask != nullptr && ok) ok &= key_mask[(long)batch * nk + kg];
-      if (static_mask != nullptr && ok) ok &= static_mask[(long)qrow * nk + kg];
+      if (static_mask != nullptr && ok)
+        ok &= Mtile[wave * 16 + lq][kg - kbase] != 0;
       s16[i] = ok ? s16[i] * scale : NEG_INF;
     }
 
@@ -330,13 +354,14 @@ void fa_bwd_dq_kernel(
     const bool* __restrict__ key_mask,
     const bool* __restrict__ static_mask,
     const unsigned char* __restrict__ tile_map,   // [nq/64, nk/32]
-    int b, int h, int nq, int nk, float scale, int causal) {
+    int b, int h, int nq, int nk, float scale, int causal, int do_bnhd) {
 
   constexpr int KV = 2 * FA_KBLK;          // 64 keys per LDS tile
   __shared__ short Kt[KV][KPAD];           // K row-major
   __shared__ short Vr[KV][KPAD];           // V row-major (A-operand of dP^T)
   __shared__ short Ktr[FA_D][KV + 8];      // K transposed (B-operand of dS*K)
   __shared__ short DSl[FA_WAVES][16][KV + 8];
+  __shared__ unsigned char Mtile[FA_QBLK][KV];
 
   const int bh = blockIdx.y;
   const int batch = bh / h;
@@ -350,7 +375,10 @@ void fa_bwd_dq_kernel(
   const short* qp = q + (long)bh * nq * FA_D;
   const short* kp = k + (long)bh * nk * FA_D;
   const short* vp = v + (long)bh * nk * FA_D;
-  const short* dop = dout + (long)bh * nq * FA_D;
+  const long do_stride = do_bnhd ? (long)h * FA_D : FA_D;
+  const short* dop = do_bnhd
+      ? dout + ((long)batch * nq * h + (bh - batch * h)) * FA_D
+      : dout + (long)bh * nq * FA_D;
 
   bf16x8 qfrag[2], dofrag[2];
   float lse_q = 0.f, D_q = 0.f;
@@ -360,7 +388,7 @@ void fa_bwd_dq_kernel(
     for (int c = 0; c < 2; ++c) {
       qfrag[c] = qok ? *reinterpret_cast<const bf16x8*>(qp + (long)qrow * FA_D + 8 * grp + 32 * c)
                      : bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
-      dofrag[c] = qok ? *reinterpret_cast<const bf16x8*>(dop + (long)qrow * FA_D + 8 * grp + 32 * c)
+      dofrag[c] = qok ? *reinterpret_cast<const bf16x8*>(dop + (long)qrow * do_stride + 8 * grp + 32 * c)
                       : bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
     }
     if (qok) {
@@ -395,7 +423,9 @@ void fa_bwd_dq_kernel(
 
   const int srow0 = tid >> 3;
   const int sc8 = (tid & 7) * 8;
-  int4v kreg[2], vreg[2];
+  const int mrow = tid >> 2;
+  const int mc16 = (tid & 3) * 16;
+  int4v kreg[2], vreg[2], mreg;
   auto prefetch = [&](int t) {
     #pragma unroll
     for (int half = 0; half < 2; ++half) {
@@ -406,6 +436,24 @@ void fa_bwd_dq_kernel(
       } else {
         kreg[half] = int4v{0, 0, 0, 0};
         vreg[half] = int4v{0, 0, 0, 0};
+      }
+    }
+    if (static_mask != nullptr) {
+      const int mq = q0 + mrow;
+      const int kb = t * KV;
+      const long base = (long)mq * nk + kb + mc16;
+      if (mq < nq && kb + mc16 + 16 <= nk && (base & 15) == 0) {
+        mreg = *reinterpret_cast<const int4v*>(
+            reinterpret_cast<const char*>(static_mask) + base);
+      } else {
+        unsigned char mb[16];
+        #pragma unroll
+        for (int e = 0; e < 16; ++e) {
+          const int kg = kb + mc16 + e;
+          mb[e] = (mq < nq && kg < nk)
+              ? (unsigned char)static_mask[(long)mq * nk + kg] : 0;
+        }
+        mreg = *reinterpret_cast<const int4v*>(mb);
       }
     }
   };
@@ -427,6 +475,8 @@ void fa_bwd_dq_kernel(
       #pragma unroll
       for (int e = 0; e < 8; ++e) Ktr[sc8 + e][row] = ks[e];
     }
+    if (static_mask != nullptr)
+      *reinterpret_cast<int4v*>(&Mtile[mrow][mc16]) = mreg;
     __syncthreads();
     if (kt_next < ntiles) prefetch(kt_next);
 
@@ -454,7 +504,8 @@ void fa_bwd_dq_kernel(
       bool ok = (kg < nk) & (qrow < nq);
       if (causal) ok &= kg <= qrow + diag;
       if (key_mask != nullptr && ok) ok &= key_mask[(long)batch * nk + kg];
-      if (static_mask != nullptr && ok) ok &= static_mask[(long)qrow * nk + kg];
+      if (static_mask != nullptr && ok)
+        ok &= Mtile[wave * 16 + lq][kg - kbase] != 0;
       const float p = ok ? __expf(s16[i] * scale - lse_q) : 0.f;
       const float ds = p * (dp16[i] - D_q) * scale;
       const int kk = (i >> 2) * 16 + grp * 4 + (i & 3);
@@ -501,7 +552,7 @@ void fa_bwd_dkv_kernel(
     const bool* __restrict__ key_mask,
     const bool* __restrict__ static_mask,
     const unsigned char* __restrict__ tile_map_t,  // [nk/64, nq/32]
-    int b, int h, int nq, int nk, float scale, int causal) {
+    int b, int h, int nq, int nk, float scale, int causal, int do_bnhd) {
 
   constexpr int KV = 2 * FA_KBLK;          // 64 q rows per LDS tile
   __shared__ short Qr[KV][KPAD];           // Q rows (B-operand of s^T)
@@ -510,6 +561,7 @@ void fa_bwd_dkv_kernel(
   __shared__ short dOtr[FA_D][KV + 8];     // dO transposed (dV = P^T dO)
   __shared__ short Pt[FA_WAVES][16][KV + 8];
   __shared__ short DSt[FA_WAVES][16][KV + 8];
+  __shared__ unsigned char Mtile[KV][FA_QBLK];   // [q in tile][key in block]
 
   const int bh = blockIdx.y;
   const int batch = bh / h;
@@ -523,7 +575,10 @@ void fa_bwd_dkv_kernel(
   const short* qp = q + (long)bh * nq * FA_D;
   const short* kp = k + (long)bh * nk * FA_D;
   const short* vp = v + (long)bh * nk * FA_D;
-  const short* dop = dout + (long)bh * nq * FA_D;
+  const long do_stride = do_bnhd ? (long)h * FA_D : FA_D;
+  const short* dop = do_bnhd
+      ? dout + ((long)batch * nq * h + (bh - batch * h)) * FA_D
+      : dout + (long)bh * nq * FA_D;
 
   bf16x8 kfrag[2], vfrag[2];
   {
@@ -566,17 +621,36 @@ void fa_bwd_dkv_kernel(
 
   const int srow0 = tid >> 3;
   const int sc8 = (tid & 7) * 8;
-  int4v qreg[2], doreg[2];
+  const int mrow = tid >> 2;
+  const int mc16 = (tid & 3) * 16;
+  int4v qreg[2], doreg[2], mreg;
   auto prefetch = [&](int t) {
     #pragma unroll
     for (int half = 0; half < 2; ++half) {
       const int qg = t * KV + srow0 + 32 * half;
       if (qg < nq) {
         qreg[half] = *reinterpret_cast<const int4v*>(qp + (long)qg * FA_D + sc8);
-        doreg[half] = *reinterpret_cast<const int4v*>(dop + (long)qg * FA_D + sc8);
+        doreg[half] = *reinterpret_cast<const int4v*>(dop + (long)qg * do_stride + sc8);
       } else {
         qreg[half] = int4v{0, 0, 0, 0};
         doreg[half] = int4v{0, 0, 0, 0};
+      }
+    }
+    if (static_mask != nullptr) {
+      const int mq = t * KV + mrow;            // q row of the streamed tile
+      const long base = (long)mq * nk + k0 + mc16;
+      if (mq < nq && k0 + mc16 + 16 <= nk && (base & 15) == 0) {
+        mreg = *reinterpret_cast<const int4v*>(
+            reinterpret_cast<const char*>(static_mask) + base);
+      } else {
+        unsigned char mb[16];
+        #pragma unroll
+        for (int e = 0; e < 16; ++e) {
+          const int kg = k0 + mc16 + e;
+          mb[e] = (mq < nq && kg < nk)
+              ? (unsigned char)static_mask[(long)mq * nk + kg] : 0;
+        }
+        mreg = *reinterpret_cast<const int4v*>(mb);
       }
     }
   };
@@ -602,6 +676,8 @@ void fa_bwd_dkv_kernel(
         dOtr[sc8 + e][row] = ds_[e];
       }
     }
+    if (static_mask != nullptr)
+      *reinterpret_cast<int4v*>(&Mtile[mrow][mc16]) = mreg;
     __syncthreads();
     if (qt_next < nqt) prefetch(qt_next);
 
@@ -634,7 +710,8 @@ void fa_bwd_dkv_kernel(
         bool ok = (key < nk) & (qg < nq);
         if (causal) ok &= key <= qg + diag;
         if (key_mask != nullptr && ok) ok &= key_mask[(long)batch * nk + key];
-        if (static_mask != nullptr && ok) ok &= static_mask[(long)qg * nk + key];
+        if (static_mask != nullptr && ok)
+          ok &= Mtile[mt * 16 + lq][key - k0] != 0;
         float p = 0.f, ds = 0.f;
         if (ok) {
           const float l = lse[(long)bh * nq + qg];
@@ -1011,14 +1088,17 @@ std::vector<torch::Tensor> fa_bwd(torch::Tensor q, torch::Tensor k, torch::Tenso
                                   std::optional<torch::Tensor> key_mask,
                                   std::optional<torch::Tensor> static_mask,
                                   std::optional<torch::Tensor> tile_map,
-                                  std::optional<torch::Tensor> tile_map_t) {
+                                  std::optional<torch::Tensor> tile_map_t,
+                                  bool out_bnhd) {
   CHK(q.is_cuda() && q.dtype() == torch::kBFloat16);
   CHK(q.is_contiguous() && k.is_contiguous() && v.is_contiguous());
   CHK(out.is_contiguous() && dout.is_contiguous());
   const int b = q.size(0), h = q.size(1), nq = q.size(2), nk = k.size(2);
 
-  // D = rowsum(dO * O), fp32 (single fused pass through ATen)
+  // D = rowsum(dO * O), fp32, in [b, h, nq] layout (single ATen pass;
+  // for bnhd inputs the permute is just a view — no copy)
   auto Dv = (dout.to(torch::kFloat32) * out.to(torch::kFloat32)).sum(-1);
+  if (out_bnhd) Dv = Dv.permute({0, 2, 1});
   Dv = Dv.contiguous();
 
   auto dq = torch::empty_like(q);
@@ -1042,7 +1122,8 @@ std::vector<torch::Tensor> fa_bwd(torch::Tensor q, torch::Tensor k, torch::Tenso
                      reinterpret_cast<const short*>(dout.data_ptr()),
                      lse.data_ptr<float>(), Dv.data_ptr<float>(),
                      reinterpret_cast<short*>(dq.data_ptr()),
-                     km, sm, tm, b, h, nq, nk, (float)scale, causal ? 1 : 0);
+                     km, sm, tm, b, h, nq, nk, (float)scale, causal ? 1 : 0,
+                     out_bnhd ? 1 : 0);
   dim3 grid_k((nk + FA_QBLK - 1) / FA_QBLK, b * h);
   hipLaunchKernelGGL(fa_bwd_dkv_kernel, grid_k, dim3(256), 0, cur_stream(),
                      reinterpret_cast<const short*>(q.data_ptr()),
@@ -1052,7 +1133,8 @@ std::vector<torch::Tensor> fa_bwd(torch::Tensor q, torch::Tensor k, torch::Tenso
                      lse.data_ptr<float>(), Dv.data_ptr<float>(),
                      reinterpret_cast<short*>(dk.data_ptr()),
                      reinterpret_cast<short*>(dv.data_ptr()),
-                     km, sm, tmt, b, h, nq, nk, (float)scale, causal ? 1 : 0);
+                     km, sm, tmt, b, h, nq, nk, (float)scale, causal ? 1 : 0,
+                     out_bnhd ? 1 : 0);
   return {dq, dk, dv};
 }
 
