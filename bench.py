@@ -33,6 +33,10 @@ CONFIGS = {
                    reversible=True, batch_size=8, vae='dvae'),
     'd': dict(dim=1024, depth=64, heads=16, attn_types=('axial_row', 'axial_col'),
               reversible=True, batch_size=4, vae='vqgan16k'),
+    # plumbing smoke (CPU-runnable, used by the distributed-launch test)
+    'tiny': dict(dim=64, depth=1, heads=1, attn_types=('full',),
+                 reversible=False, batch_size=2, vae='dvae',
+                 image_size=64, text_seq_len=16, vocab=100),
 }
 
 
@@ -64,13 +68,18 @@ def maybe_self_launch(args):
 def build_model(cfg, device):
     from dalle_pytorch_amd import DALLE, DiscreteVAE, VQGanVAE
     torch.manual_seed(1234)
+    image_size = cfg.get('image_size', 256)
     if cfg['vae'] == 'dvae':
-        vae = DiscreteVAE(image_size=256, num_layers=3, num_tokens=8192,
-                          codebook_dim=512, hidden_dim=64)
+        hidden = 8 if image_size < 256 else 64
+        tokens = 512 if image_size < 256 else 8192
+        vae = DiscreteVAE(image_size=image_size, num_layers=3, num_tokens=tokens,
+                          codebook_dim=64 if image_size < 256 else 512,
+                          hidden_dim=hidden)
     else:
-        vae = VQGanVAE(num_tokens=16384)  # f=16 -> 256 image tokens... see note
+        vae = VQGanVAE(num_tokens=16384)  # f=16 -> 256 image tokens (config D)
     dalle = DALLE(
-        dim=cfg['dim'], vae=vae, num_text_tokens=10000, text_seq_len=256,
+        dim=cfg['dim'], vae=vae, num_text_tokens=cfg.get('vocab', 10000),
+        text_seq_len=cfg.get('text_seq_len', 256),
         depth=cfg['depth'], heads=cfg['heads'], dim_head=64,
         attn_types=cfg['attn_types'], reversible=cfg['reversible'],
         shift_tokens=True, rotary_emb=True)
@@ -107,8 +116,11 @@ def main():
     # synthetic data pool: distinct per rank & step, generated once on device
     torch.manual_seed(4321 + rank)
     n_pool = 4
-    pool = [(torch.randint(1, 10000, (bsz, 256), device=device),
-             torch.rand(bsz, 3, 256, 256, device=device)) for _ in range(n_pool)]
+    vocab = cfg.get('vocab', 10000)
+    tlen = dalle.text_seq_len
+    isz = dalle.vae.image_size
+    pool = [(torch.randint(1, vocab, (bsz, tlen), device=device),
+             torch.rand(bsz, 3, isz, isz, device=device)) for _ in range(n_pool)]
 
     autocast = torch.autocast(device_type='cuda', dtype=torch.bfloat16,
                               enabled=use_cuda)

@@ -91,6 +91,7 @@ class OpenAIDiscreteVAE(nn.Module):
             mods = [cls(n_in if i == 0 else n_out, n_out) for i in range(blocks)]
             return nn.Sequential(*mods)
 
+        n_init = nh // 2  # dall_e n_init = 128 at n_hid = 256
         self.encoder = nn.Sequential(OrderedDict([
             ('input', nn.Conv2d(3, nh, 7, padding=3)),
             ('group_1', group(nh, nh, blocks_per_group)),
@@ -102,10 +103,12 @@ class OpenAIDiscreteVAE(nn.Module):
             ('group_4', group(4 * nh, 8 * nh, blocks_per_group)),
             ('output', nn.Sequential(nn.ReLU(), nn.Conv2d(8 * nh, vocab_size, 1))),
         ]))
-        self.codebook = nn.Embedding(vocab_size, nh // 2)  # dall_e uses n_init=128
+        # dall_e's decoder "input" is Conv2d(vocab, n_init, 1) applied to the
+        # one-hot codes — numerically an embedding table, stored here as one
+        # so decode() is a gather instead of an 8192-channel conv
+        self.codebook = nn.Embedding(vocab_size, n_init)
         self.decoder = nn.Sequential(OrderedDict([
-            ('input', nn.Conv2d(nh // 2, nh, 1)),
-            ('group_1', group(nh, 8 * nh, blocks_per_group, enc=False)),
+            ('group_1', group(n_init, 8 * nh, blocks_per_group, enc=False)),
             ('up_1', nn.Upsample(scale_factor=2, mode='nearest')),
             ('group_2', group(8 * nh, 4 * nh, blocks_per_group, enc=False)),
             ('up_2', nn.Upsample(scale_factor=2, mode='nearest')),
@@ -263,6 +266,21 @@ class VQGanVAE(nn.Module):
                  image_size=256, num_tokens=16384, embed_dim=256, ch=128,
                  ch_mult=(1, 1, 2, 2, 4), num_res_blocks=2, gumbel=False):
         super().__init__()
+        if vqgan_config_path is not None:
+            # taming-transformers yaml config (reference vae.py:148-158 uses
+            # OmegaConf; plain yaml suffices for the fields we need)
+            import yaml
+            with open(vqgan_config_path) as f:
+                cfg = yaml.safe_load(f)
+            params = cfg['model']['params']
+            gumbel = 'Gumbel' in cfg['model'].get('target', '')
+            dd = params['ddconfig']
+            ch = dd.get('ch', ch)
+            ch_mult = tuple(dd.get('ch_mult', ch_mult))
+            num_res_blocks = dd.get('num_res_blocks', num_res_blocks)
+            image_size = dd.get('resolution', image_size)
+            num_tokens = params.get('n_embed', num_tokens)
+            embed_dim = params.get('embed_dim', embed_dim)
         f = 2 ** (len(ch_mult) - 1)
         self.image_size = image_size
         self.num_tokens = num_tokens

@@ -24,8 +24,7 @@ class TextImageDataset(Dataset):
     def __init__(self, folder, text_len=256, image_size=128, truncate_captions=False,
                  resize_ratio=0.75, tokenizer=None, shuffle=False):
         super().__init__()
-        import torchvision.transforms as T
-        from PIL import Image  # noqa: F401 (validated lazily)
+        from dalle_pytorch_amd.utils.vision import random_resized_crop, to_tensor
 
         self.shuffle = shuffle
         path = Path(folder)
@@ -38,11 +37,14 @@ class TextImageDataset(Dataset):
         self.text_len = text_len
         self.truncate_captions = truncate_captions
         self.tokenizer = tokenizer
-        self.image_transform = T.Compose([
-            T.Lambda(lambda img: img.convert('RGB') if img.mode != 'RGB' else img),
-            T.RandomResizedCrop(image_size, scale=(resize_ratio, 1.), ratio=(1., 1.)),
-            T.ToTensor(),
-        ])
+
+        def transform(img):
+            if img.mode != 'RGB':
+                img = img.convert('RGB')
+            img = random_resized_crop(img, image_size, scale=(resize_ratio, 1.))
+            return to_tensor(img)
+
+        self.image_transform = transform
 
     def __len__(self):
         return len(self.keys)

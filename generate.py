@@ -89,12 +89,9 @@ def main(argv=None):
 
         subdir = out_root / raw_text.replace(' ', '_')[:100]
         subdir.mkdir(parents=True, exist_ok=True)
-        try:
-            from torchvision.utils import save_image
-            for j, img in enumerate(images):
-                save_image(img, subdir / f'{j}.png', normalize=False)
-        except ImportError:
-            torch.save(images, subdir / 'images.pt')
+        from dalle_pytorch_amd.utils.vision import save_image
+        for j, img in enumerate(images):
+            save_image(img.float().cpu(), subdir / f'{j}.png')
         print(f'created {images.shape[0]} images at "{subdir}"')
 
 

@@ -5,6 +5,8 @@ import torch
 def pytest_configure(config):
     config.addinivalue_line(
         'markers', 'gpu: needs a ROCm GPU (run on an MI355X box)')
+    config.addinivalue_line(
+        'markers', 'slow: long-running CPU test (minutes)')
 
 
 def pytest_collection_modifyitems(config, items):

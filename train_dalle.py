@@ -46,8 +46,8 @@ def parse_args(argv=None):
     p.add_argument('--synthetic', action='store_true',
                    help='train on synthetic random data (no dataset needed)')
     p.add_argument('--wds', type=str, default='',
-                   help='webdataset keys (accepted for CLI parity; tar input '
-                        'not yet wired, use folders)')
+                   help='tar shard spec for streaming input, e.g. '
+                        '"shards/train-{000..099}.tar" or a glob')
     p.add_argument('--truncate_captions', action='store_true')
     p.add_argument('--random_resize_crop_lower_ratio', dest='resize_ratio',
                    type=float, default=0.75)
@@ -66,7 +66,10 @@ def parse_args(argv=None):
     p.add_argument('--deepspeed', action='store_true', help=argparse.SUPPRESS)
     p.add_argument('--horovod', action='store_true', help=argparse.SUPPRESS)
     p.add_argument('--local_rank', type=int, default=0, help=argparse.SUPPRESS)
-    p.add_argument('--flops_profiler', action='store_true')
+    p.add_argument('--flops_profiler', action='store_true',
+                   help='profile FLOPs at --profile_step then stop '
+                        '(reference behavior)')
+    p.add_argument('--profile_step', type=int, default=200)
 
     t = p.add_argument_group('Training settings')
     t.add_argument('--epochs', default=20, type=int)
@@ -192,7 +195,16 @@ def main(argv=None):
 
     # --------------------------------------------------------- dataset
     text_seq_len = dalle_params['text_seq_len']  # checkpoint-authoritative on resume
-    if args.synthetic or not exists(args.image_text_folder):
+    streaming = bool(args.wds)
+    if streaming:
+        from dalle_pytorch_amd.utils.wds import TarImageTextDataset
+        ds = TarImageTextDataset(args.wds, tokenizer=tok, text_len=text_seq_len,
+                                 image_size=vae.image_size,
+                                 truncate_captions=args.truncate_captions,
+                                 resize_ratio=args.resize_ratio)
+        if is_root:
+            print(f'streaming from {len(ds.shards)} tar shard(s)')
+    elif args.synthetic or not exists(args.image_text_folder):
         ds = SyntheticTextImageDataset(
             length=max(args.batch_size * world * 64, 512),
             text_len=text_seq_len, image_size=vae.image_size,
@@ -203,12 +215,13 @@ def main(argv=None):
             image_size=vae.image_size, resize_ratio=args.resize_ratio,
             truncate_captions=args.truncate_captions, tokenizer=tok, shuffle=True)
         assert len(ds) > 0, 'dataset is empty'
-    if is_root:
+    if is_root and not streaming:
         print(f'{len(ds)} image-text pairs found for training')
 
     sampler = DistributedSampler(ds, num_replicas=world, rank=rank, shuffle=True) \
-        if world > 1 else None
-    dl = DataLoader(ds, batch_size=args.batch_size, shuffle=sampler is None,
+        if (world > 1 and not streaming) else None
+    dl = DataLoader(ds, batch_size=args.batch_size,
+                    shuffle=(sampler is None and not streaming),
                     drop_last=True, sampler=sampler, num_workers=2,
                     pin_memory=device.type == 'cuda')
 
@@ -298,6 +311,28 @@ def main(argv=None):
                         text[:1], filter_thres=0.9, use_cache=True)
                 logger.log_image('generated image', sample_img[0].float().cpu(),
                                  step=global_step)
+
+            if args.flops_profiler and global_step == args.profile_step:
+                # profile the next step's wall time, print, then abort
+                # (reference train_dalle.py:492-499,656-657)
+                import time as _t
+                if device.type == 'cuda':
+                    torch.cuda.synchronize()
+                t0 = _t.perf_counter()
+                with torch.autocast(device_type='cuda', dtype=torch.bfloat16,
+                                    enabled=autocast_enabled):
+                    ploss = dalle(text, images, return_loss=True)
+                ploss.backward()
+                engine.finish_gradient_sync()
+                opt.step()
+                engine.zero_grad()
+                if device.type == 'cuda':
+                    torch.cuda.synchronize()
+                if is_root:
+                    from dalle_pytorch_amd.utils.flops import profile_step
+                    profile_step(dalle, text.shape[0], _t.perf_counter() - t0,
+                                 reversible=dalle_params.get('reversible', False))
+                args.stop_after_steps = global_step  # abort after profiling
 
             if args.stop_after_steps and global_step >= args.stop_after_steps:
                 break

@@ -74,14 +74,9 @@ def main(argv=None):
 
     channels = 4 if args.transparent else 3
     if args.image_folder:
-        import torchvision.transforms as T
-        from torchvision.datasets import ImageFolder
-        ds = ImageFolder(args.image_folder, T.Compose([
-            T.Lambda(lambda img: img.convert('RGBA' if args.transparent else 'RGB')),
-            T.Resize(args.image_size),
-            T.CenterCrop(args.image_size),
-            T.ToTensor(),
-        ]))
+        from dalle_pytorch_amd.utils.vision import ImageFolderDataset
+        ds = ImageFolderDataset(args.image_folder, args.image_size,
+                                transparent=args.transparent)
         assert len(ds) > 0, 'folder does not contain any images'
     else:
         ds = _SyntheticImages(max(args.batch_size * world * 64, 256),
