@@ -136,14 +136,23 @@ def main():
         engine.zero_grad()
         return loss
 
-    def gen_step(_):
-        text = pool[0][0][:1].repeat(args.gen_batch, 1)
-        with autocast:
-            dalle.generate_images(text, use_cache=True, filter_thres=0.9)
-
-    step = train_step if args.mode == 'train' else gen_step
+    decoder = None
     if args.mode == 'generate':
         dalle.eval()
+        if not args.eager:
+            from dalle_pytorch_amd.engine import FastDecoder
+            decoder = FastDecoder(dalle, batch_size=args.gen_batch,
+                                  use_graph=use_cuda)
+
+    def gen_step(_):
+        text = pool[0][0][:1].repeat(args.gen_batch, 1)
+        if decoder is not None:
+            decoder.generate(text, filter_thres=0.9)
+        else:
+            with autocast:
+                dalle.generate_images(text, use_cache=True, filter_thres=0.9)
+
+    step = train_step if args.mode == 'train' else gen_step
 
     for i in range(args.warmup):
         step(i)

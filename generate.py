@@ -33,6 +33,8 @@ def parse_args(argv=None):
     p.add_argument('--gentxt', action='store_true',
                    help='complete the prompt with the model before generating')
     p.add_argument('--no_cache', action='store_true')
+    p.add_argument('--no_fast', action='store_true',
+                   help='disable the static-shape decode engine (HIP graphs)')
     p.add_argument('--cond_scale', type=float, default=1.0)
     return p.parse_args(argv)
 
@@ -79,12 +81,28 @@ def main(argv=None):
                                        truncate_text=True).to(device)
             text_tokens = text_tokens.repeat(args.num_images, 1)
 
+        use_fast = (not args.no_fast and not args.gentxt
+                    and args.cond_scale == 1.0)
+        decoder = None
+        if use_fast:
+            try:
+                from dalle_pytorch_amd.engine import FastDecoder
+                decoder = FastDecoder(dalle, batch_size=args.batch_size,
+                                      use_graph=device.type == 'cuda')
+            except (ValueError, AssertionError) as e:
+                print(f'fast decoder unavailable ({e}); using cached decode')
+
         images = []
         for i in range(0, text_tokens.shape[0], args.batch_size):
             chunk = text_tokens[i:i + args.batch_size]
-            images.append(dalle.generate_images(
-                chunk, filter_thres=args.top_k, temperature=args.temperature,
-                use_cache=not args.no_cache, cond_scale=args.cond_scale))
+            if decoder is not None and chunk.shape[0] == args.batch_size:
+                images.append(decoder.generate(
+                    chunk, filter_thres=args.top_k,
+                    temperature=args.temperature))
+            else:
+                images.append(dalle.generate_images(
+                    chunk, filter_thres=args.top_k, temperature=args.temperature,
+                    use_cache=not args.no_cache, cond_scale=args.cond_scale))
         images = torch.cat(images, dim=0)
 
         subdir = out_root / raw_text.replace(' ', '_')[:100]
