@@ -10,7 +10,7 @@ from pathlib import Path
 
 class RunLogger:
     def __init__(self, project, config=None, enabled=True, use_wandb=True,
-                 output_dir='.', run_name=None):
+                 output_dir='.', run_name=None, entity=None):
         self.enabled = enabled
         self.wandb = None
         self._jsonl = None
@@ -20,7 +20,7 @@ class RunLogger:
             try:
                 import wandb
                 self.wandb = wandb
-                wandb.init(project=project, name=run_name,
+                wandb.init(project=project, name=run_name, entity=entity,
                            config=config or {}, resume=False)
             except Exception:
                 self.wandb = None

@@ -66,6 +66,9 @@ def parse_args(argv=None):
     p.add_argument('--deepspeed', action='store_true', help=argparse.SUPPRESS)
     p.add_argument('--horovod', action='store_true', help=argparse.SUPPRESS)
     p.add_argument('--local_rank', type=int, default=0, help=argparse.SUPPRESS)
+    p.add_argument('--torch_profile', action='store_true',
+                   help='capture a torch.profiler chrome trace of steps '
+                        '10-13 into <output_dir>/trace.json')
     p.add_argument('--flops_profiler', action='store_true',
                    help='profile FLOPs at --profile_step then stop '
                         '(reference behavior)')
@@ -245,7 +248,20 @@ def main(argv=None):
         **{k: v for k, v in dalle_params.items() if k != 'shared_attn_ids'},
         'batch_size': args.batch_size, 'world_size': world,
         'learning_rate': args.learning_rate},
-        enabled=is_root, output_dir=args.output_dir, run_name=args.wandb_name)
+        enabled=is_root, output_dir=args.output_dir, run_name=args.wandb_name,
+        entity=args.wandb_entity)
+
+    profiler = None
+    if args.torch_profile and is_root:
+        from torch.profiler import profile, schedule, ProfilerActivity
+        acts = [ProfilerActivity.CPU]
+        if device.type == 'cuda':
+            acts.append(ProfilerActivity.CUDA)
+        profiler = profile(
+            activities=acts, schedule=schedule(wait=10, warmup=1, active=3),
+            on_trace_ready=lambda p: p.export_chrome_trace(
+                str(out_dir / 'trace.json')))
+        profiler.__enter__()
 
     autocast_enabled = (args.fp16 or args.amp) and device.type == 'cuda'
 
@@ -334,6 +350,8 @@ def main(argv=None):
                                  reversible=dalle_params.get('reversible', False))
                 args.stop_after_steps = global_step  # abort after profiling
 
+            if profiler is not None:
+                profiler.step()
             if args.stop_after_steps and global_step >= args.stop_after_steps:
                 break
         if scheduler is not None:
@@ -342,6 +360,8 @@ def main(argv=None):
         if args.stop_after_steps and global_step >= args.stop_after_steps:
             break
 
+    if profiler is not None:
+        profiler.__exit__(None, None, None)
     save(args.epochs)
     logger.finish()
     barrier()
