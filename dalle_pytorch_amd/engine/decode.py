@@ -25,8 +25,6 @@ the model's dict-cache generation (itself verified bitwise-equal to
 uncached recomputation) across attention types, shift, stable, reversible.
 """
 
-import contextlib
-
 import torch
 import torch.nn.functional as F
 from torch import nn
@@ -78,7 +76,7 @@ def _pattern_mask(leaf, seq_len, device):
 
 
 class _LayerState:
-    __slots__ = ('leaf', 'info', 'pattern', 'k', 'v', 'ring', 'is_attn')
+    __slots__ = ('leaf', 'info', 'pattern', 'k', 'v', 'ring', 'is_attn', 'w')
 
 
 class FastDecoder:
@@ -125,7 +123,17 @@ class FastDecoder:
                     dim = st.info['norm'].normalized_shape[0]
                     st.ring = torch.zeros(self.b, S, dim // 2,
                                           device=self.device, dtype=self.dtype)
+                st.w = self._materialize(st)
                 self.states.append(st)
+
+        # head weights, pre-cast once (the decode loop never re-casts)
+        cast = lambda t: None if t is None else \
+            t.detach().to(self.device, self.dtype)
+        head_ln, head_lin = dalle.to_logits[0], dalle.to_logits[1]
+        self.head_w = {
+            'ln_w': cast(head_ln.weight), 'ln_b': cast(head_ln.bias),
+            'w': cast(head_lin.weight), 'b': cast(head_lin.bias),
+        }
 
         self.offset_t = torch.zeros(1, dtype=torch.long, device=self.device)
         self.key_arange = torch.arange(self.N, device=self.device)
@@ -133,10 +141,29 @@ class FastDecoder:
         self._g_token = None
         self._g_logits = None
 
-    def _ac(self):
-        if self.device.type == 'cuda' and self.dtype == torch.bfloat16:
-            return torch.autocast('cuda', torch.bfloat16)
-        return contextlib.nullcontext()
+    def _materialize(self, st):
+        """Pre-cast this branch's weights to the engine dtype once: under
+        autocast the casts would otherwise be captured into the decode graph
+        and replayed for every token (measured ~30% of replay time)."""
+        cast = lambda t: None if t is None else \
+            t.detach().to(self.device, self.dtype)
+        w = {'ln_w': cast(st.info['norm'].weight),
+             'ln_b': cast(st.info['norm'].bias),
+             'scale': cast(st.info['scale'])}
+        if st.info['norm_out'] is not None:
+            w['lno_w'] = cast(st.info['norm_out'].weight)
+            w['lno_b'] = cast(st.info['norm_out'].bias)
+        if st.is_attn:
+            w['qkv'] = cast(st.leaf.to_qkv.weight)
+            w['out_w'] = cast(st.leaf.to_out[0].weight)
+            w['out_b'] = cast(st.leaf.to_out[0].bias)
+        else:
+            net = st.leaf.net
+            w['ff1_w'] = cast(net[0].weight)
+            w['ff1_b'] = cast(net[0].bias)
+            w['ff2_w'] = cast(net[3].weight)
+            w['ff2_b'] = cast(net[3].bias)
+        return w
 
     # ----------------------------------------------------------- branches
 
@@ -157,7 +184,7 @@ class FastDecoder:
         offset 0; decode has n == 1)."""
         leaf = st.leaf
         h, d = leaf.heads, leaf.dim_head
-        qkv = leaf.to_qkv(x)
+        qkv = F.linear(x, st.w['qkv'])
         q, k, v = (t.reshape(self.b, n, h, d).permute(0, 2, 1, 3)
                    for t in qkv.chunk(3, dim=-1))
         if self.rotary:
@@ -188,7 +215,7 @@ class FastDecoder:
             out = attention_core(q, k, v, leaf.scale, causal=True,
                                  static_mask=sm)
         out = out.permute(0, 2, 1, 3).reshape(self.b, n, h * d)
-        return leaf.to_out(out)
+        return F.linear(out, st.w['out_w'], st.w['out_b'])
 
     def _shift_prefill(self, st, x, n):
         """Training-style token shift over the prompt + ring seeding
@@ -243,15 +270,24 @@ class FastDecoder:
         return torch.cat((top_q, left_q, x[..., dim // 2:]), dim=-1)
 
     def _branch(self, st, x, offset_t, n):
-        y = st.info['norm'](x)
+        dim = x.shape[-1]
+        y = F.layer_norm(x, (dim,), st.w['ln_w'], st.w['ln_b'],
+                         st.info['norm'].eps)
         if st.info['shift'] is not None:
             y = self._shift_prefill(st, y, n) if n > 1 else \
                 self._shift_decode(st, y, offset_t)
-        y = self._attn(st, y, offset_t, n) if st.is_attn else st.leaf.net(y)
+        if st.is_attn:
+            y = self._attn(st, y, offset_t, n)
+        else:
+            from dalle_pytorch_amd.ops import geglu
+            y = F.linear(y, st.w['ff1_w'], st.w['ff1_b'])
+            y = geglu(y)
+            y = F.linear(y, st.w['ff2_w'], st.w['ff2_b'])
         if st.info['norm_out'] is not None:
-            y = st.info['norm_out'](y)
-        if st.info['scale'] is not None:
-            y = y * st.info['scale'].to(y.dtype)
+            y = F.layer_norm(y, (dim,), st.w['lno_w'], st.w['lno_b'],
+                             st.info['norm_out'].eps)
+        if st.w['scale'] is not None:
+            y = y * st.w['scale']
         return y
 
     def _run_stack(self, x, offset_t, n):
@@ -274,7 +310,9 @@ class FastDecoder:
         d = self.dalle
         if d.stable:
             x = x / x.amax(dim=-1, keepdim=True)
-        logits = d.to_logits(x)
+        x = F.layer_norm(x, (x.shape[-1],), self.head_w['ln_w'],
+                         self.head_w['ln_b'], d.to_logits[0].eps)
+        logits = F.linear(x, self.head_w['w'], self.head_w['b'])
         lm = d.logits_mask[0].to(self.device).index_select(0, position_mask_rows)
         return logits.masked_fill(lm.unsqueeze(0), -torch.finfo(logits.dtype).max)
 
@@ -298,10 +336,10 @@ class FastDecoder:
             iemb = iemb + d.image_pos_emb(iemb)
             tokens = torch.cat((tokens, iemb), dim=1)
         n = tokens.shape[1]
-        with self._ac():
-            x = self._run_stack(tokens, None, n)
-            rows = torch.arange(n, device=device)
-            logits = self._head(x, rows)
+        x = tokens.to(self.dtype)
+        x = self._run_stack(x, None, n)
+        rows = torch.arange(n, device=device)
+        logits = self._head(x, rows)
         self.offset_t.fill_(n)
         return logits[:, -1]
 
@@ -312,15 +350,14 @@ class FastDecoder:
         masked logits [b, total_tokens]; advances the offset."""
         d = self.dalle
         off = self.offset_t
-        with self._ac():
-            emb = d.image_emb(token).unsqueeze(1)
-            if not self.rotary:
-                g = (off - d.text_seq_len - 1).clamp(min=0)
-                full = (d.image_pos_emb.weights[0] + d.image_pos_emb.weights[1]) \
-                    .reshape(1, -1, emb.shape[-1])
-                emb = emb + full.index_select(1, g)
-            x = self._run_stack(emb, off, 1)
-            logits = self._head(x, off)[:, 0]
+        emb = d.image_emb(token).unsqueeze(1)
+        if not self.rotary:
+            g = (off - d.text_seq_len - 1).clamp(min=0)
+            full = (d.image_pos_emb.weights[0] + d.image_pos_emb.weights[1]) \
+                .reshape(1, -1, emb.shape[-1])
+            emb = emb + full.index_select(1, g)
+        x = self._run_stack(emb.to(self.dtype), off, 1)
+        logits = self._head(x, off)[:, 0]
         self.offset_t += 1
         return logits
 
