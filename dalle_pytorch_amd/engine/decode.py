@@ -135,8 +135,16 @@ class FastDecoder:
             'w': cast(head_lin.weight), 'b': cast(head_lin.bias),
         }
 
+        from dalle_pytorch_amd.ops.dispatch import hip_available
         self.offset_t = torch.zeros(1, dtype=torch.long, device=self.device)
         self.key_arange = torch.arange(self.N, device=self.device)
+        # fused single-token kernels need bf16 caches, d=64 and the extension
+        dims_ok = all((not st.is_attn) or st.leaf.dim_head == 64
+                      for st in self.states)
+        self._fused_decode = (self.device.type == 'cuda'
+                              and self.dtype == torch.bfloat16
+                              and self.N <= 4096 and dims_ok
+                              and hip_available())
         self._graph = None
         self._g_token = None
         self._g_logits = None
@@ -185,6 +193,14 @@ class FastDecoder:
         leaf = st.leaf
         h, d = leaf.heads, leaf.dim_head
         qkv = F.linear(x, st.w['qkv'])
+        if n == 1 and self._fused_decode:
+            from dalle_pytorch_amd.ops.dispatch import hip_module
+            out = hip_module().fa_decode(
+                qkv.view(self.b, -1), st.k, st.v,
+                self.cos if self.rotary else None,
+                self.sin if self.rotary else None,
+                offset_t, st.pattern, leaf.scale).view(self.b, 1, h * d)
+            return F.linear(out, st.w['out_w'], st.w['out_b'])
         q, k, v = (t.reshape(self.b, n, h, d).permute(0, 2, 1, 3)
                    for t in qkv.chunk(3, dim=-1))
         if self.rotary:
@@ -251,6 +267,11 @@ class FastDecoder:
     def _shift_decode(self, st, x, offset_t):
         shift = st.info['shift']
         S, text_len = shift.image_size, shift.text_len
+        if self._fused_decode:
+            from dalle_pytorch_amd.ops.dispatch import hip_module
+            return hip_module().shift_decode(
+                x.reshape(self.b, -1).contiguous(), offset_t, st.ring,
+                text_len).view(self.b, 1, -1)
         dim = x.shape[-1]
         qdim = dim // 4
         g = (offset_t - text_len).clamp(min=0)

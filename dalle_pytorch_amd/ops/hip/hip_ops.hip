@@ -1002,6 +1002,159 @@ __global__ void token_shift_kernel(
 }
 
 // ---------------------------------------------------------------------------
+// Fused single-token decode attention (guide App. B "attention decode").
+//
+// One launch per token per layer replaces the eager chain (rope gathers,
+// cache scatter, mask build, masked_fill, softmax, two batched GEMVs,
+// permute): block (bi, head) applies rotary to this token's q/k/v, writes
+// the K/V caches at *offset, computes the pattern+length-masked dots over
+// the cached keys (one wave per key, lane-per-element shuffle reduce),
+// block-wide softmax in LDS, and the P*V accumulation with lane = d so V
+// rows stream coalesced. Memory-bound by design: reads each cached K/V row
+// once.
+// ---------------------------------------------------------------------------
+
+constexpr int DEC_MAXN = 4096;
+
+__global__ __launch_bounds__(256)
+void fa_decode_kernel(
+    const short* __restrict__ qkv,    // [b, 3*h*64] (this token's projection)
+    short* __restrict__ kc,           // [b, h, N, 64]
+    short* __restrict__ vc,           // [b, h, N, 64]
+    const float* __restrict__ cosv,   // [Ncos, rot] or null
+    const float* __restrict__ sinv,
+    const long* __restrict__ offset,  // [1] current position
+    const bool* __restrict__ pattern, // [N, N] or null; row = *offset
+    short* __restrict__ out,          // [b, h*64]
+    int b, int h, int N, int rot, float scale) {
+
+  __shared__ float P[DEC_MAXN];
+  __shared__ float qs[64];
+  __shared__ float red[4 * 64];
+  __shared__ float stat[8];
+
+  const int head = blockIdx.x, bi = blockIdx.y;
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6, lane = tid & 63;
+  const long off = *offset;
+
+  // ---- rope + cache write for this token (threads 0..63, d = tid)
+  if (tid < 64) {
+    const int d = tid;
+    const long base = (long)bi * 3 * h * 64 + (long)head * 64 + d;
+    float qv = bf2f(qkv[base]);
+    float kv = bf2f(qkv[base + (long)h * 64]);
+    float vv = bf2f(qkv[base + 2L * h * 64]);
+    if (d < rot && cosv != nullptr) {
+      const int prt = d ^ 1;                        // pair partner
+      const float cs = cosv[off * rot + d];
+      const float sn = sinv[off * rot + d];
+      const float sgn = (d & 1) ? 1.f : -1.f;       // even: -x_{d+1}, odd: +x_{d-1}
+      const long pbase = (long)bi * 3 * h * 64 + (long)head * 64 + prt;
+      qv = qv * cs + sgn * bf2f(qkv[pbase]) * sn;
+      kv = kv * cs + sgn * bf2f(qkv[pbase + (long)h * 64]) * sn;
+      vv = vv * cs + sgn * bf2f(qkv[pbase + 2L * h * 64]) * sn;
+    }
+    const long cbase = (((long)bi * h + head) * N + off) * 64 + d;
+    kc[cbase] = f2bf(kv);
+    vc[cbase] = f2bf(vv);
+    qs[d] = qv * scale;
+  }
+  __syncthreads();
+
+  // ---- dots: one wave per key (4 keys in flight), lane = d
+  const bool* prow = pattern ? pattern + off * N : nullptr;
+  const short* krow0 = kc + ((long)bi * h + head) * N * 64;
+  for (long key = wave; key <= off; key += 4) {
+    const bool allowed = prow == nullptr || prow[key];
+    float dot = NEG_INF;
+    if (allowed) {
+      float p = qs[lane] * bf2f(krow0[key * 64 + lane]);
+      #pragma unroll
+      for (int s = 32; s > 0; s >>= 1) p += __shfl_xor(p, s);
+      dot = p;
+    }
+    if (lane == 0) P[key] = dot;
+  }
+  __syncthreads();
+
+  // ---- block softmax over P[0..off]
+  float m = NEG_INF;
+  for (long i = tid; i <= off; i += 256) m = fmaxf(m, P[i]);
+  #pragma unroll
+  for (int s = 32; s > 0; s >>= 1) m = fmaxf(m, __shfl_xor(m, s));
+  if (lane == 0) stat[wave] = m;
+  __syncthreads();
+  m = fmaxf(fmaxf(stat[0], stat[1]), fmaxf(stat[2], stat[3]));
+
+  float lsum = 0.f;
+  for (long i = tid; i <= off; i += 256) {
+    const float p = (P[i] == NEG_INF) ? 0.f : __expf(P[i] - m);
+    P[i] = p;
+    lsum += p;
+  }
+  #pragma unroll
+  for (int s = 32; s > 0; s >>= 1) lsum += __shfl_xor(lsum, s);
+  if (lane == 0) stat[4 + wave] = lsum;
+  __syncthreads();
+  const float denom = stat[4] + stat[5] + stat[6] + stat[7];
+
+  // ---- P*V: lane = d, wave-strided keys, coalesced V rows
+  const short* vrow0 = vc + ((long)bi * h + head) * N * 64;
+  float acc = 0.f;
+  for (long key = wave; key <= off; key += 4) {
+    const float p = P[key];
+    if (p != 0.f) acc += p * bf2f(vrow0[key * 64 + lane]);
+  }
+  red[wave * 64 + lane] = acc;
+  __syncthreads();
+  if (wave == 0) {
+    const float total = red[lane] + red[64 + lane] + red[128 + lane] +
+        red[192 + lane];
+    out[(long)bi * h * 64 + (long)head * 64 + lane] =
+        f2bf(denom > 0.f ? total / denom : 0.f);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Fused single-token token-shift (ring-buffer form): out token gets its
+// first quarter from ring[g % S] (the grid row above), second quarter from
+// ring[(g-1) % S] (left neighbor, zeroed at column 0), then the ring slot
+// is overwritten with this token's first half. One launch replaces ~8.
+// ---------------------------------------------------------------------------
+
+__global__ void shift_decode_kernel(
+    const short* __restrict__ x,     // [b, dim]
+    const long* __restrict__ offset, // [1]
+    short* __restrict__ ring,        // [b, S, dim/2]
+    short* __restrict__ out,         // [b, dim]
+    int b, int dim, int S, int text_len) {
+  const int bi = blockIdx.y;
+  const int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= dim) return;
+  const long g0 = *offset - text_len;
+  const long g = g0 < 0 ? 0 : g0;
+  const int pos = (int)(g % S);
+  const int prev = (int)(((g - 1) % S + S) % S);
+  const int half = dim / 2, quarter = dim / 4;
+
+  short val;
+  if (i < quarter) {
+    val = ring[((long)bi * S + pos) * half + i];              // top quarter
+  } else if (i < half) {
+    val = (g % S == 0) ? (short)0
+        : ring[((long)bi * S + prev) * half + i];             // left quarter
+  } else {
+    val = x[(long)bi * dim + i];                              // pass-through
+  }
+  // race-free without a barrier: the only slot written is `pos`; element
+  // ring[pos][i] is read and written by the SAME thread i (program order),
+  // and the quarter..half reads touch slot `prev` != `pos` for S >= 2
+  if (i < half) ring[((long)bi * S + pos) * half + i] = x[(long)bi * dim + i];
+  out[(long)bi * dim + i] = val;
+}
+
+// ---------------------------------------------------------------------------
 // MFMA layout probe (test support): one 16x16x32 bf16 MFMA with the exact
 // fragment mappings the attention kernel assumes. The GPU test compares
 // C against torch.matmul on asymmetric inputs (guide G9) so a wrong operand
@@ -1240,6 +1393,60 @@ torch::Tensor geglu_bwd(torch::Tensor x, torch::Tensor dout) {
   return dx;
 }
 
+torch::Tensor fa_decode(torch::Tensor qkv, torch::Tensor kc, torch::Tensor vc,
+                        std::optional<torch::Tensor> cosv,
+                        std::optional<torch::Tensor> sinv,
+                        torch::Tensor offset,
+                        std::optional<torch::Tensor> pattern,
+                        double scale) {
+  CHK(qkv.is_cuda() && qkv.dtype() == torch::kBFloat16 && qkv.is_contiguous());
+  CHK(kc.is_contiguous() && vc.is_contiguous());
+  const int b = kc.size(0), h = kc.size(1), N = kc.size(2);
+  CHK(kc.size(3) == 64 && N <= DEC_MAXN);
+  CHK(offset.dtype() == torch::kLong);
+  int rot = 0;
+  const float* cp = nullptr;
+  const float* sp = nullptr;
+  if (cosv.has_value()) {
+    rot = cosv->size(1);
+    cp = cosv->data_ptr<float>();
+    sp = sinv->data_ptr<float>();
+  }
+  const bool* pat = nullptr;
+  if (pattern.has_value()) {
+    CHK(pattern->dtype() == torch::kBool && pattern->is_contiguous());
+    CHK(pattern->size(1) == N);
+    pat = pattern->data_ptr<bool>();
+  }
+  auto out = torch::empty({b, (long)h * 64}, qkv.options());
+  dim3 grid(h, b);
+  hipLaunchKernelGGL(fa_decode_kernel, grid, dim3(256), 0, cur_stream(),
+                     reinterpret_cast<const short*>(qkv.data_ptr()),
+                     reinterpret_cast<short*>(kc.data_ptr()),
+                     reinterpret_cast<short*>(vc.data_ptr()),
+                     cp, sp, offset.data_ptr<long>(), pat,
+                     reinterpret_cast<short*>(out.data_ptr()),
+                     b, h, N, rot, (float)scale);
+  return out;
+}
+
+torch::Tensor shift_decode(torch::Tensor x, torch::Tensor offset,
+                           torch::Tensor ring, int64_t text_len) {
+  CHK(x.is_cuda() && x.is_contiguous() && ring.is_contiguous());
+  const int b = x.size(0), dim = x.size(-1);
+  const int S = ring.size(1);
+  CHK(ring.size(2) == dim / 2);
+  auto out = torch::empty_like(x);
+  dim3 grid((dim + 255) / 256, b);
+  hipLaunchKernelGGL(shift_decode_kernel, grid, dim3(256), 0, cur_stream(),
+                     reinterpret_cast<const short*>(x.data_ptr()),
+                     offset.data_ptr<long>(),
+                     reinterpret_cast<short*>(ring.data_ptr()),
+                     reinterpret_cast<short*>(out.data_ptr()),
+                     b, dim, S, (int)text_len);
+  return out;
+}
+
 torch::Tensor token_shift(torch::Tensor x, int64_t text_len, int64_t image_size,
                           bool backward) {
   CHK(x.is_cuda() && x.is_contiguous() && x.dim() == 3);
@@ -1276,6 +1483,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "fused qkv split + rotary (q,k,v all rotated)");
   m.def("rope_split_bwd", &rope_split_bwd, "rope_split backward");
   m.def("token_shift", &token_shift, "fused token shift (fwd/transpose)");
+  m.def("fa_decode", &fa_decode, "fused single-token decode attention");
+  m.def("shift_decode", &shift_decode, "fused single-token token shift");
   m.def("geglu_fwd", &geglu_fwd, "fused GEGLU forward");
   m.def("geglu_bwd", &geglu_bwd, "fused GEGLU backward");
   m.def("mfma_probe", &mfma_probe, "MFMA 16x16x32 bf16 layout probe");

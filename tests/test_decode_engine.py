@@ -80,3 +80,29 @@ def test_engine_gpu_graph_mode():
     torch.manual_seed(7)
     got = dec.generate(text, temperature=1e-8, filter_thres=0.99)
     assert torch.allclose(got, ref, atol=1e-3), (got - ref).abs().max().item()
+
+
+@pytest.mark.gpu
+def test_fused_decode_kernels_match_torch_path():
+    """bf16 fused fa_decode/shift_decode vs the torch-op decode path."""
+    torch.manual_seed(6)
+    d = tiny_dalle(attn_types=('axial_row', 'full'), depth=2, dim=256,
+                   heads=4, dim_head=64).cuda().eval()
+    text = torch.randint(1, 50, (2, 8), device='cuda')
+    token = torch.randint(0, 64, (2,), device='cuda')
+
+    dec_f = FastDecoder(d, batch_size=2, dtype=torch.bfloat16)
+    assert dec_f._fused_decode
+    dec_t = FastDecoder(d, batch_size=2, dtype=torch.bfloat16)
+    dec_t._fused_decode = False
+
+    with torch.no_grad():
+        a = dec_f.prefill(text)
+        b = dec_t.prefill(text)
+        for _ in range(5):
+            a = dec_f.step(token)
+            b = dec_t.step(token)
+    valid = b > -1e30
+    rel = (a[valid].float() - b[valid].float()).abs().max() / \
+        b[valid].float().abs().max()
+    assert rel < 5e-2, rel.item()
