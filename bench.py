@@ -175,6 +175,8 @@ def main():
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
     elapsed = float(t.item())
 
+    max_mem_gb = round(torch.cuda.max_memory_allocated() / 2**30, 2) \
+        if use_cuda else None
     if rank == 0:
         ms_per_step = elapsed / args.steps * 1000
         if args.mode == 'train':
@@ -199,6 +201,7 @@ def main():
                 'global_batch': global_batch, 'seq_len': seq_len,
                 'image_seq_len': image_seq_len,
                 'parallelism': f'dp{world}', 'mode': args.mode,
+                'max_mem_gb': max_mem_gb,
                 'eager': bool(args.eager)},
         }))
 

@@ -115,6 +115,14 @@ class DiscreteVAE(nn.Module):
     def get_codebook_indices(self, images):
         was_training = self.training
         self.eval()
+        if images.is_cuda:
+            # channels-last steers MIOpen to its tuned NHWC bf16 implicit-GEMM
+            # conv kernels; the NCHW bf16 path falls back to naive_conv
+            # (measured ~22% of the training step)
+            if not getattr(self, '_encoder_channels_last', False):
+                self.encoder.to(memory_format=torch.channels_last)
+                self._encoder_channels_last = True
+            images = images.contiguous(memory_format=torch.channels_last)
         logits = self(images, return_logits=True)
         self.train(was_training)
         return logits.argmax(dim=1).flatten(1)

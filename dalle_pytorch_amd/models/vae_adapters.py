@@ -121,6 +121,11 @@ class OpenAIDiscreteVAE(nn.Module):
     @torch.no_grad()
     def get_codebook_indices(self, img):
         img = map_pixels(img)
+        if img.is_cuda:
+            if not getattr(self, '_encoder_channels_last', False):
+                self.encoder.to(memory_format=torch.channels_last)
+                self._encoder_channels_last = True
+            img = img.contiguous(memory_format=torch.channels_last)
         logits = self.encoder(img)
         return logits.argmax(dim=1).flatten(1)
 
@@ -313,6 +318,12 @@ class VQGanVAE(nn.Module):
     @torch.no_grad()
     def get_codebook_indices(self, img):
         img = 2 * img - 1  # reference vae.py:212
+        if img.is_cuda:
+            if not getattr(self, '_encoder_channels_last', False):
+                self.encoder.to(memory_format=torch.channels_last)
+                self.quant_conv.to(memory_format=torch.channels_last)
+                self._encoder_channels_last = True
+            img = img.contiguous(memory_format=torch.channels_last)
         z = self.quant_conv(self.encoder(img))
         b, c, h, w = z.shape
         flat = z.permute(0, 2, 3, 1).reshape(-1, c)
