@@ -28,9 +28,9 @@ CONFIGS = {
     'b': dict(dim=512, depth=12, heads=8, attn_types=('full',),
               reversible=False, batch_size=8, vae='dvae'),
     'c': dict(dim=1024, depth=12, heads=16, attn_types=('axial_row', 'axial_col'),
-              reversible=True, batch_size=8, vae='dvae'),
+              reversible=True, batch_size=64, vae='dvae'),
     'c_full': dict(dim=1024, depth=12, heads=16, attn_types=('full',),
-                   reversible=True, batch_size=8, vae='dvae'),
+                   reversible=True, batch_size=64, vae='dvae'),
     'd': dict(dim=1024, depth=64, heads=16, attn_types=('axial_row', 'axial_col'),
               reversible=True, batch_size=4, vae='vqgan16k'),
     # plumbing smoke (CPU-runnable, used by the distributed-launch test)
