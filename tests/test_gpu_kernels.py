@@ -356,3 +356,20 @@ def test_generation_gpu_cached():
     imgs = d.generate_images(text, use_cache=True)
     assert imgs.shape == (2, 3, 64, 64)
     assert torch.isfinite(imgs).all()
+
+
+def test_dvae_gemm_encoder_matches_conv():
+    """The im2col+GEMM frozen-encoder path == the nn.Conv2d stack."""
+    from dalle_pytorch_amd import DiscreteVAE
+    torch.manual_seed(10)
+    vae = DiscreteVAE(image_size=64, num_layers=3, num_tokens=128,
+                      codebook_dim=64, hidden_dim=16,
+                      num_resnet_blocks=1).cuda().eval()
+    img = torch.rand(2, 3, 64, 64, device='cuda')
+    with torch.no_grad():
+        ref = vae.encoder(vae.norm(img))
+        got = vae._encode_as_gemms(vae.norm(img))
+    assert (got - ref).abs().max().item() < 1e-3
+    codes = vae.get_codebook_indices(img)
+    ref_codes = ref.argmax(dim=1).flatten(1)
+    assert (codes == ref_codes).float().mean().item() > 0.99
