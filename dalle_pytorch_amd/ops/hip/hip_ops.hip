@@ -67,6 +67,22 @@ DEVFN bf16x8 frag_from_lds(const short* base) {
   return *reinterpret_cast<const bf16x8*>(base);  // ds_read_b128
 }
 
+
+// XCD-aware block mapping (guide T1, bijective m204 form): the runtime
+// round-robins flat block ids across the 8 XCDs, so consecutive ids land on
+// different L2s. Remapping gives each XCD a CONTIGUOUS range of the
+// (bh-major) linearization — q-tiles that share one (batch,head)'s K/V
+// stream stay on one XCD's L2 instead of re-fetching from HBM 8x.
+DEVFN void xcd_chunked(int bid, int nwg, int tiles_per_bh,
+                       int* tile, int* bh) {
+  constexpr int NXCD = 8;
+  const int q = nwg / NXCD, r = nwg % NXCD;
+  const int xcd = bid % NXCD, pos = bid / NXCD;
+  const int lin = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + pos;
+  *tile = lin % tiles_per_bh;
+  *bh = lin / tiles_per_bh;
+}
+
 __global__ __launch_bounds__(256, 2)
 void fa_fwd_d64_kernel(
     const short* __restrict__ q,    // [bh, nq, 64] bf16 bits
@@ -89,9 +105,11 @@ void fa_fwd_d64_kernel(
   __shared__ short Pl[FA_WAVES][16][KV + 8];
   __shared__ unsigned char Mtile[FA_QBLK][KV];   // static-mask tile
 
-  const int bh = blockIdx.y;
+  const int n_qt = (nq + FA_QBLK - 1) / FA_QBLK;
+  int qtile, bh;
+  xcd_chunked(blockIdx.x, n_qt * b * h, n_qt, &qtile, &bh);
   const int batch = bh / h;
-  const int q0 = blockIdx.x * FA_QBLK;
+  const int q0 = qtile * FA_QBLK;
   const int tid = threadIdx.x;
   const int wave = tid >> 6;
   const int lane = tid & 63;
@@ -130,7 +148,7 @@ void fa_fwd_d64_kernel(
     ntiles = lim < 0 ? 0 : (lim / KV + 1);
   }
   const unsigned char* tmap_row =
-      tile_map ? tile_map + (long)blockIdx.x * ntk : nullptr;
+      tile_map ? tile_map + (long)qtile * ntk : nullptr;
 
   // block-sparse skip: the host precomputes, per (64q, 32k) granule,
   // whether any static-mask entry is set; a 64-key tile is live if either
@@ -363,9 +381,11 @@ void fa_bwd_dq_kernel(
   __shared__ short DSl[FA_WAVES][16][KV + 8];
   __shared__ unsigned char Mtile[FA_QBLK][KV];
 
-  const int bh = blockIdx.y;
+  const int n_qt = (nq + FA_QBLK - 1) / FA_QBLK;
+  int qtile, bh;
+  xcd_chunked(blockIdx.x, n_qt * b * h, n_qt, &qtile, &bh);
   const int batch = bh / h;
-  const int q0 = blockIdx.x * FA_QBLK;
+  const int q0 = qtile * FA_QBLK;
   const int tid = threadIdx.x;
   const int wave = tid >> 6, lane = tid & 63;
   const int lq = lane & 15, grp = lane >> 4;
@@ -407,7 +427,7 @@ void fa_bwd_dq_kernel(
     ntiles = lim < 0 ? 0 : (lim / KV + 1);
   }
   const unsigned char* tmap_row =
-      tile_map ? tile_map + (long)blockIdx.x * ntk : nullptr;
+      tile_map ? tile_map + (long)qtile * ntk : nullptr;
 
   auto tile_live = [&](int t) -> bool {
     if (!tmap_row) return true;
@@ -563,9 +583,11 @@ void fa_bwd_dkv_kernel(
   __shared__ short DSt[FA_WAVES][16][KV + 8];
   __shared__ unsigned char Mtile[KV][FA_QBLK];   // [q in tile][key in block]
 
-  const int bh = blockIdx.y;
+  const int n_kt = (nk + FA_QBLK - 1) / FA_QBLK;
+  int ktile, bh;
+  xcd_chunked(blockIdx.x, n_kt * b * h, n_kt, &ktile, &bh);
   const int batch = bh / h;
-  const int k0 = blockIdx.x * FA_QBLK;    // 64 keys per block
+  const int k0 = ktile * FA_QBLK;         // 64 keys per block
   const int tid = threadIdx.x;
   const int wave = tid >> 6, lane = tid & 63;
   const int lq = lane & 15, grp = lane >> 4;
@@ -605,7 +627,7 @@ void fa_bwd_dkv_kernel(
     qt_start = qmin <= 0 ? 0 : qmin / KV;
   }
   const unsigned char* tmap_row =
-      tile_map_t ? tile_map_t + (long)blockIdx.x * nqg : nullptr;
+      tile_map_t ? tile_map_t + (long)ktile * nqg : nullptr;
 
   auto tile_live = [&](int t) -> bool {
     if (!tmap_row) return true;
@@ -1222,7 +1244,7 @@ std::vector<torch::Tensor> fa_fwd(torch::Tensor q, torch::Tensor k, torch::Tenso
     tm = tile_map->data_ptr<uint8_t>();
   }
 
-  dim3 grid((nq + FA_QBLK - 1) / FA_QBLK, b * h);
+  dim3 grid(((nq + FA_QBLK - 1) / FA_QBLK) * b * h);
   hipLaunchKernelGGL(fa_fwd_d64_kernel, grid, dim3(256), 0, cur_stream(),
                      reinterpret_cast<const short*>(q.data_ptr()),
                      reinterpret_cast<const short*>(k.data_ptr()),
@@ -1267,7 +1289,7 @@ std::vector<torch::Tensor> fa_bwd(torch::Tensor q, torch::Tensor k, torch::Tenso
   if (tile_map.has_value()) tm = tile_map->data_ptr<uint8_t>();
   if (tile_map_t.has_value()) tmt = tile_map_t->data_ptr<uint8_t>();
 
-  dim3 grid_q((nq + FA_QBLK - 1) / FA_QBLK, b * h);
+  dim3 grid_q(((nq + FA_QBLK - 1) / FA_QBLK) * b * h);
   hipLaunchKernelGGL(fa_bwd_dq_kernel, grid_q, dim3(256), 0, cur_stream(),
                      reinterpret_cast<const short*>(q.data_ptr()),
                      reinterpret_cast<const short*>(k.data_ptr()),
@@ -1277,7 +1299,7 @@ std::vector<torch::Tensor> fa_bwd(torch::Tensor q, torch::Tensor k, torch::Tenso
                      reinterpret_cast<short*>(dq.data_ptr()),
                      km, sm, tm, b, h, nq, nk, (float)scale, causal ? 1 : 0,
                      out_bnhd ? 1 : 0);
-  dim3 grid_k((nk + FA_QBLK - 1) / FA_QBLK, b * h);
+  dim3 grid_k(((nk + FA_QBLK - 1) / FA_QBLK) * b * h);
   hipLaunchKernelGGL(fa_bwd_dkv_kernel, grid_k, dim3(256), 0, cur_stream(),
                      reinterpret_cast<const short*>(q.data_ptr()),
                      reinterpret_cast<const short*>(k.data_ptr()),
