@@ -25,7 +25,6 @@ Semantics notes carried over from the reference (kept bit-for-bit):
 from math import ceil
 
 import torch
-import torch.nn.functional as F
 from torch import nn
 
 from dalle_pytorch_amd.models.positional import apply_rotary_to_qkv

@@ -13,7 +13,6 @@ MI355X-native precision; --deepspeed/--horovod -> the RCCL engine).
 """
 
 import argparse
-import os
 import time
 from pathlib import Path
 

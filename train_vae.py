@@ -7,7 +7,6 @@
 
 import argparse
 import math
-import time
 from pathlib import Path
 
 import torch
