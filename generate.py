@@ -30,6 +30,9 @@ def parse_args(argv=None):
     p.add_argument('--bpe_path', type=str, default=None)
     p.add_argument('--hug', action='store_true')
     p.add_argument('--chinese', action='store_true')
+    p.add_argument('--taming', action='store_true',
+                   help='use the native VQGAN (with --vqgan_model_path/'
+                        '--vqgan_config_path) as the decoder VAE')
     p.add_argument('--gentxt', action='store_true',
                    help='complete the prompt with the model before generating')
     p.add_argument('--no_cache', action='store_true')
@@ -61,7 +64,8 @@ def main(argv=None):
 
     ckpt = load_dalle_checkpoint(args.dalle_path)
     vae = None
-    if ckpt.get('vae_class_name') == 'VQGanVAE' and args.vqgan_model_path:
+    if args.taming or (ckpt.get('vae_class_name') == 'VQGanVAE'
+                       and (args.vqgan_model_path or args.vqgan_config_path)):
         from dalle_pytorch_amd import VQGanVAE
         vae = VQGanVAE(args.vqgan_model_path, args.vqgan_config_path)
     dalle, vae = build_dalle_from_checkpoint(ckpt, vae=vae)

@@ -118,3 +118,29 @@ def test_single_process_noop():
     m2(x).sum().backward()
     for a, b in zip(g0, (p.grad for p in m2.parameters())):
         assert torch.allclose(a, b)
+
+
+def test_distributed_utils_facade():
+    """Reference-surface facade maps onto the RCCL engine."""
+    import argparse
+    from dalle_pytorch_amd import distributed_utils as du
+
+    parser = argparse.ArgumentParser()
+    parser = du.wrap_arg_parser(parser)
+    args = parser.parse_args([])
+    backend = du.set_backend_from_args(args)
+    backend.initialize()
+    assert backend.get_world_size() == 1 and backend.is_root_worker()
+    assert du.using_backend(type(backend))
+    backend.check_batch_size(4)
+
+    model = torch.nn.Linear(4, 4)
+    opt = torch.optim.SGD(model.parameters(), lr=0.1)
+    m2, o2, _, _ = backend.distribute(model=model, optimizer=opt)
+    assert m2 is model and o2 is opt
+    v = backend.average_all(torch.tensor(3.0))
+    assert float(v) == 3.0
+
+    args = parser.parse_args(['--distributed_backend', 'deepspeed'])
+    backend = du.set_backend_from_args(args)
+    assert du.using_backend(du.DeepSpeedBackend)

@@ -94,6 +94,8 @@ def parse_args(argv=None):
     m.add_argument('--heads', default=8, type=int)
     m.add_argument('--dim_head', default=64, type=int)
     m.add_argument('--reversible', dest='reversible', action='store_true')
+    m.add_argument('--attn_dropout', default=0.0, type=float)
+    m.add_argument('--ff_dropout', default=0.0, type=float)
     m.add_argument('--loss_img_weight', default=7, type=int)
     m.add_argument('--attn_types', default='full', type=str,
                    help='comma separated: full, sparse, axial_row, axial_col, conv_like')
@@ -178,6 +180,8 @@ def main(argv=None):
             heads=args.heads,
             dim_head=args.dim_head,
             reversible=args.reversible,
+            attn_dropout=args.attn_dropout,
+            ff_dropout=args.ff_dropout,
             loss_img_weight=args.loss_img_weight,
             attn_types=tuple(args.attn_types.split(',')),
             shift_tokens=args.shift_tokens,
