@@ -1,30 +1,44 @@
 #!/usr/bin/env python3
-"""Summarize a rocprofv3 rocpd database's PMC counters per kernel (run on
-the GPU box; keeps only a small text summary)."""
+"""Summarize a rocprofv3 rocpd database's PMC counters per kernel (runs on
+the GPU box; emits a small text summary instead of shipping the db)."""
 import sqlite3
 import sys
 from collections import defaultdict
 
 db = sqlite3.connect(sys.argv[1])
 cur = db.cursor()
-try:
-    rows = list(cur.execute("""
-        SELECT k.name, p.name, SUM(e.value), COUNT(*)
-        FROM pmc_events e
-        JOIN kernels k ON e.dispatch_id = k.dispatch_id
-        JOIN pmc_info p ON e.pmc_id = p.id
-        GROUP BY k.name, p.name"""))
-except Exception as exc:
-    # schema fallback: dump table names
-    print('query failed:', exc)
-    for r in cur.execute("SELECT name FROM sqlite_master WHERE type IN ('table','view')"):
-        print(r[0])
-    sys.exit(0)
-per = defaultdict(dict)
-for kname, cname, val, cnt in rows:
-    per[kname][cname] = (val, cnt)
-for kname, counters in sorted(per.items()):
-    short = kname.split('(')[0][:70]
-    print(f'== {short}')
-    for cname, (val, cnt) in sorted(counters.items()):
-        print(f'   {cname:28s} sum={val:.3e} dispatches={cnt}')
+
+
+def cols(table):
+    try:
+        return [r[1] for r in cur.execute(f'PRAGMA table_info({table})')]
+    except sqlite3.Error:
+        return []
+
+
+candidates = ['counters_collection', 'pmc_events', 'rocpd_pmc_event']
+for t in candidates:
+    c = cols(t)
+    if c:
+        print(f'# table {t}: {c}', file=sys.stderr)
+
+# preferred: the convenience view joining everything
+cc = cols('counters_collection')
+if cc:
+    name_col = next((c for c in cc if 'kernel' in c.lower() and 'name' in c.lower()), None)
+    ctr_col = next((c for c in cc if c.lower() in ('counter_name', 'name')), None)
+    val_col = next((c for c in cc if 'value' in c.lower()), None)
+    if name_col and ctr_col and val_col:
+        per = defaultdict(lambda: defaultdict(lambda: [0.0, 0]))
+        for kname, cname, val in cur.execute(
+                f'SELECT {name_col}, {ctr_col}, {val_col} FROM counters_collection'):
+            a = per[kname][cname]
+            a[0] += val or 0.0
+            a[1] += 1
+        for kname in sorted(per):
+            print(f'== {str(kname).split("(")[0][:70]}')
+            for cname, (val, cnt) in sorted(per[kname].items()):
+                print(f'   {cname:28s} sum={val:.4e} n={cnt}')
+        sys.exit(0)
+
+print('counters_collection view not usable; columns dumped to stderr')
