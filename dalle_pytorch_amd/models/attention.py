@@ -173,9 +173,12 @@ class Attention(nn.Module):
         sm = self.static_mask[offset:offset + n_q, :n_k]
         tiles = tiles_t = None
         if sm.is_cuda and offset == 0 and n_k == self.static_mask.shape[1]:
-            _, tiles, tiles_t = _cached_mask(
-                ('attn-static', id(self.static_mask), n_q, n_k),
-                device, lambda: sm)
+            hit = getattr(self.static_mask, '_dalle_amd_tiles', None)
+            if hit is None or hit[0] != (n_q, n_k):
+                hit = ((n_q, n_k), build_tile_map(sm),
+                       build_tile_map(sm.t()))
+                self.static_mask._dalle_amd_tiles = hit
+            _, tiles, tiles_t = hit
         return sm, tiles, tiles_t
 
     def forward(self, x, mask=None, rotary_pos_emb=None, cache=None, cache_key=None):

@@ -10,18 +10,15 @@ import torch
 
 from dalle_pytorch_amd.ops.dispatch import hip_module
 
-_TRIG_CACHE = {}
-
-
 def trig_tables(angles: torch.Tensor):
-    """cos/sin fp32 [N, rot] for a rotary angle table [1, N, rot]; cached
-    per (tensor, device) since the table is a fixed model buffer."""
-    key = (id(angles), angles.device, angles.shape[-2])
-    hit = _TRIG_CACHE.get(key)
+    """cos/sin fp32 [N, rot] for a rotary angle table [1, N, rot]; cached on
+    the tensor object itself (the table is a fixed model buffer, so the
+    cache's lifetime tracks the model's)."""
+    hit = getattr(angles, '_dalle_amd_trig', None)
     if hit is None:
         a = angles.squeeze(0).float()
         hit = (a.cos().contiguous(), a.sin().contiguous())
-        _TRIG_CACHE[key] = hit
+        angles._dalle_amd_trig = hit
     return hit
 
 
