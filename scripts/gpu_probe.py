@@ -20,11 +20,6 @@ def main():
     C = ext.mfma_probe(A, B)
     ref = A.float() @ B.float()
     print('probe err', (C - ref).abs().max().item())
-    refT = A.float().t() @ B.float() if False else None
-    alt = {
-        'C^T': (C - ref.t()).abs().max().item(),
-        'A^T.B': (C - (A.float().t()[:16, :16] if False else torch.zeros(16, 16, device='cuda'))).abs().max().item() if False else None,
-    }
     print('C^T err', (C - ref.t()).abs().max().item())
     print('C[0,:4]', C[0, :4].tolist())
     print('ref[0,:4]', ref[0, :4].tolist())
