@@ -25,9 +25,9 @@ for t in candidates:
 # preferred: the convenience view joining everything
 cc = cols('counters_collection')
 if cc:
-    name_col = next((c for c in cc if 'kernel' in c.lower() and 'name' in c.lower()), None)
-    ctr_col = next((c for c in cc if c.lower() in ('counter_name', 'name')), None)
-    val_col = next((c for c in cc if 'value' in c.lower()), None)
+    name_col = 'kernel_name' if 'kernel_name' in cc else None
+    ctr_col = 'counter_name' if 'counter_name' in cc else None
+    val_col = 'value' if 'value' in cc else None
     if name_col and ctr_col and val_col:
         per = defaultdict(lambda: defaultdict(lambda: [0.0, 0]))
         for kname, cname, val in cur.execute(
