@@ -67,6 +67,17 @@ DEVFN bf16x8 frag_from_lds(const short* base) {
   return *reinterpret_cast<const bf16x8*>(base);  // ds_read_b128
 }
 
+// Block-XOR swizzle for the transposed LDS tiles ([64 d][64 keys(+pad)]):
+// XOR the key's 8-block index with the d-row's block index. Without it,
+// every 16-byte-aligned row stride puts all d-blocks of one key group in
+// the same banks, so the 8-element scalar transpose-stores serialize
+// 8-way (PMC: 88M SQ_LDS_BANK_CONFLICT per fa_fwd dispatch). The XOR
+// flips only key bits 3..5, so 8-element groups stay contiguous and the
+// ds_read_b128 fragments remain 16B-aligned.
+DEVFN int swz_key(int d, int key) {
+  return key ^ (((d >> 3) & 7) << 3);
+}
+
 
 // XCD-aware block mapping (guide T1, bijective m204 form): the runtime
 // round-robins flat block ids across the 8 XCDs, so consecutive ids land on
@@ -219,7 +230,8 @@ void fa_fwd_d64_kernel(
       *reinterpret_cast<int4v*>(&Kt[row][sc8]) = kreg[half];
       const short* vs = reinterpret_cast<const short*>(&vreg[half]);
       #pragma unroll
-      for (int e = 0; e < 8; ++e) Vt[sc8 + e][row] = vs[e];
+      for (int e = 0; e < 8; ++e)
+        Vt[sc8 + e][swz_key(sc8 + e, row)] = vs[e];
     }
     if (static_mask != nullptr)
       *reinterpret_cast<int4v*>(&Mtile[mrow][mc16]) = mreg;
@@ -312,8 +324,9 @@ void fa_fwd_d64_kernel(
     __builtin_amdgcn_s_setprio(1);
     #pragma unroll
     for (int nt = 0; nt < 4; ++nt) {
-      bf16x8 vf0 = frag_from_lds(&Vt[lq + 16 * nt][8 * grp]);
-      bf16x8 vf1 = frag_from_lds(&Vt[lq + 16 * nt][32 + 8 * grp]);
+      const int d = lq + 16 * nt;
+      bf16x8 vf0 = frag_from_lds(&Vt[d][swz_key(d, 8 * grp)]);
+      bf16x8 vf1 = frag_from_lds(&Vt[d][swz_key(d, 32 + 8 * grp)]);
       acc[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pf0, vf0, acc[nt], 0, 0, 0);
       acc[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pf1, vf1, acc[nt], 0, 0, 0);
     }
@@ -493,7 +506,8 @@ void fa_bwd_dq_kernel(
       *reinterpret_cast<int4v*>(&Vr[row][sc8]) = vreg[half];
       const short* ks = reinterpret_cast<const short*>(&kreg[half]);
       #pragma unroll
-      for (int e = 0; e < 8; ++e) Ktr[sc8 + e][row] = ks[e];
+      for (int e = 0; e < 8; ++e)
+        Ktr[sc8 + e][swz_key(sc8 + e, row)] = ks[e];
     }
     if (static_mask != nullptr)
       *reinterpret_cast<int4v*>(&Mtile[mrow][mc16]) = mreg;
@@ -537,8 +551,9 @@ void fa_bwd_dq_kernel(
     __builtin_amdgcn_s_setprio(1);
     #pragma unroll
     for (int nt = 0; nt < 4; ++nt) {
-      bf16x8 kf0 = frag_from_lds(&Ktr[lq + 16 * nt][8 * grp]);
-      bf16x8 kf1 = frag_from_lds(&Ktr[lq + 16 * nt][32 + 8 * grp]);
+      const int d = lq + 16 * nt;
+      bf16x8 kf0 = frag_from_lds(&Ktr[d][swz_key(d, 8 * grp)]);
+      bf16x8 kf1 = frag_from_lds(&Ktr[d][swz_key(d, 32 + 8 * grp)]);
       acc[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(dsf0, kf0, acc[nt], 0, 0, 0);
       acc[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(dsf1, kf1, acc[nt], 0, 0, 0);
     }
@@ -694,8 +709,9 @@ void fa_bwd_dkv_kernel(
       const short* ds_ = reinterpret_cast<const short*>(&doreg[half]);
       #pragma unroll
       for (int e = 0; e < 8; ++e) {
-        Qtr[sc8 + e][row] = qs[e];
-        dOtr[sc8 + e][row] = ds_[e];
+        const int kk = swz_key(sc8 + e, row);
+        Qtr[sc8 + e][kk] = qs[e];
+        dOtr[sc8 + e][kk] = ds_[e];
       }
     }
     if (static_mask != nullptr)
@@ -753,10 +769,11 @@ void fa_bwd_dkv_kernel(
     __builtin_amdgcn_s_setprio(1);
     #pragma unroll
     for (int nt = 0; nt < 4; ++nt) {
-      bf16x8 dof0 = frag_from_lds(&dOtr[lq + 16 * nt][8 * grp]);
-      bf16x8 dof1 = frag_from_lds(&dOtr[lq + 16 * nt][32 + 8 * grp]);
-      bf16x8 qf0 = frag_from_lds(&Qtr[lq + 16 * nt][8 * grp]);
-      bf16x8 qf1 = frag_from_lds(&Qtr[lq + 16 * nt][32 + 8 * grp]);
+      const int d = lq + 16 * nt;
+      bf16x8 dof0 = frag_from_lds(&dOtr[d][swz_key(d, 8 * grp)]);
+      bf16x8 dof1 = frag_from_lds(&dOtr[d][swz_key(d, 32 + 8 * grp)]);
+      bf16x8 qf0 = frag_from_lds(&Qtr[d][swz_key(d, 8 * grp)]);
+      bf16x8 qf1 = frag_from_lds(&Qtr[d][swz_key(d, 32 + 8 * grp)]);
       acc_dv[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pf0, dof0, acc_dv[nt], 0, 0, 0);
       acc_dv[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pf1, dof1, acc_dv[nt], 0, 0, 0);
       acc_dk[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(dsf0, qf0, acc_dk[nt], 0, 0, 0);
