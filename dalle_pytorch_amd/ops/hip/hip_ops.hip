@@ -114,7 +114,7 @@ void fa_fwd_d64_kernel(
   __shared__ short Kt[KV][KPAD];
   __shared__ short Vt[FA_D][KV + 8];
   __shared__ short Pl[FA_WAVES][16][KV + 8];
-  __shared__ unsigned char Mtile[FA_QBLK][KV];   // static-mask tile
+  __shared__ unsigned char Mtile[FA_QBLK][KV + 16];   // static-mask tile (80B stride: 2-way reads)
 
   const int n_qt = (nq + FA_QBLK - 1) / FA_QBLK;
   int qtile, bh;
@@ -392,7 +392,7 @@ void fa_bwd_dq_kernel(
   __shared__ short Vr[KV][KPAD];           // V row-major (A-operand of dP^T)
   __shared__ short Ktr[FA_D][KV + 8];      // K transposed (B-operand of dS*K)
   __shared__ short DSl[FA_WAVES][16][KV + 8];
-  __shared__ unsigned char Mtile[FA_QBLK][KV];
+  __shared__ unsigned char Mtile[FA_QBLK][KV + 16];
 
   const int n_qt = (nq + FA_QBLK - 1) / FA_QBLK;
   int qtile, bh;
@@ -596,7 +596,7 @@ void fa_bwd_dkv_kernel(
   __shared__ short dOtr[FA_D][KV + 8];     // dO transposed (dV = P^T dO)
   __shared__ short Pt[FA_WAVES][16][KV + 8];
   __shared__ short DSt[FA_WAVES][16][KV + 8];
-  __shared__ unsigned char Mtile[KV][FA_QBLK];   // [q in tile][key in block]
+  __shared__ unsigned char Mtile[KV][FA_QBLK + 16];   // [q in tile][key in block] (80B stride)
 
   const int n_kt = (nk + FA_QBLK - 1) / FA_QBLK;
   int ktile, bh;
