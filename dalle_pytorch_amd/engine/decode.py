@@ -307,24 +307,28 @@ class FastDecoder:
         if st.info['norm_out'] is not None:
             y = F.layer_norm(y, (dim,), st.w['lno_w'], st.w['lno_b'],
                              st.info['norm_out'].eps)
-        if st.w['scale'] is not None:
-            y = y * st.w['scale']
         return y
+
+    def _residual(self, x, st, y):
+        # x + y*scale in one fused kernel (addcmul) instead of mul + add
+        if st.w['scale'] is not None:
+            return torch.addcmul(x, y, st.w['scale'])
+        return x + y
 
     def _run_stack(self, x, offset_t, n):
         if not self.reversible:
             it = iter(self.states)
             for attn_st in it:
                 ff_st = next(it)
-                x = x + self._branch(attn_st, x, offset_t, n)
-                x = x + self._branch(ff_st, x, offset_t, n)
+                x = self._residual(x, attn_st, self._branch(attn_st, x, offset_t, n))
+                x = self._residual(x, ff_st, self._branch(ff_st, x, offset_t, n))
             return x
         x1, x2 = x, x.clone()
         it = iter(self.states)
         for f_st in it:
             g_st = next(it)
-            x1 = x1 + self._branch(f_st, x2, offset_t, n)
-            x2 = x2 + self._branch(g_st, x1, offset_t, n)
+            x1 = self._residual(x1, f_st, self._branch(f_st, x2, offset_t, n))
+            x2 = self._residual(x2, g_st, self._branch(g_st, x1, offset_t, n))
         return (x1 + x2) / 2
 
     def _head(self, x, position_mask_rows):
