@@ -187,13 +187,16 @@ def attention_core(q, k, v, scale, causal=True, key_mask=None, static_mask=None,
 
 
 def build_tile_map(static_mask, causal=False):
-    """uint8 [ceil(nq/64), ceil(nk/32)] block map: 1 where any mask entry in
-    the (64, 32) tile is True (conservative — kernel still applies the exact
-    element mask)."""
+    """uint8 [ceil(nq/64), ceil(nk/32)] block map of the static mask:
+    0 = no entry set (kernel skips the tile), 1 = partially set (kernel
+    applies the exact element mask), 2 = fully set (kernel can skip the
+    per-element mask loop entirely when causality also allows it)."""
     nq, nk = static_mask.shape
     tq, tk = (nq + 63) // 64, (nk + 31) // 32
     padded = torch.zeros(tq * 64, tk * 32, dtype=torch.bool,
                          device=static_mask.device)
     padded[:nq, :nk] = static_mask
-    tiles = padded.reshape(tq, 64, tk, 32).any(dim=3).any(dim=1)
-    return tiles.to(torch.uint8).contiguous()
+    blocks = padded.reshape(tq, 64, tk, 32)
+    any_set = blocks.any(dim=3).any(dim=1)
+    all_set = blocks.all(dim=3).all(dim=1)
+    return (any_set.to(torch.uint8) + all_set.to(torch.uint8)).contiguous()

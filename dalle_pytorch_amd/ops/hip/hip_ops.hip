@@ -175,6 +175,15 @@ void fa_fwd_d64_kernel(
     while (t < ntiles && !tile_live(t)) ++t;
     return t;
   };
+  // a tile whose two map granules are both 2 has every mask entry set:
+  // combined with causal-interior + bounds checks the per-element mask
+  // loop can be skipped (the fa kernels are VALU-bound on exactly that)
+  auto tile_full = [&](int t) -> bool {
+    if (!tmap_row) return true;
+    const int g0 = 2 * t;
+    if (g0 + 1 >= ntk) return false;
+    return tmap_row[g0] == 2 && tmap_row[g0 + 1] == 2;
+  };
 
   // staging geometry: 64 rows x 64 cols = 512 16B chunks, 2 per thread
   const int srow0 = tid >> 3;           // chunk-0 row (0..31)
@@ -259,16 +268,26 @@ void fa_fwd_d64_kernel(
     // After the swap: C col = lane&15 = q row; C row = grp*4 + r = key.
     // s16[i] = S[qrow][kbase + (i>>2)*16 + grp*4 + (i&3)].
 
-    // ---- scale + masks
-    #pragma unroll
-    for (int i = 0; i < 16; ++i) {
-      const int kg = kbase + (i >> 2) * 16 + grp * 4 + (i & 3);
-      bool ok = (kg < nk) & (qrow < nq);
-      if (causal) ok &= kg <= qrow + diag;
-      if (key_mask != nullptr && ok) ok &= key_mask[(long)batch * nk + kg];
-      if (static_mask != nullptr && ok)
-        ok &= Mtile[wave * 16 + lq][kg - kbase] != 0;
-      s16[i] = ok ? s16[i] * scale : NEG_INF;
+    // ---- scale + masks (uniform fast path for fully-unmasked interior)
+    const bool interior =
+        (kbase + KV <= nk) && (q0 + FA_QBLK <= nq) &&
+        (!causal || (kbase + KV - 1 <= q0 + diag)) &&
+        key_mask == nullptr &&
+        (static_mask == nullptr || tile_full(kt));
+    if (interior) {
+      #pragma unroll
+      for (int i = 0; i < 16; ++i) s16[i] *= scale;
+    } else {
+      #pragma unroll
+      for (int i = 0; i < 16; ++i) {
+        const int kg = kbase + (i >> 2) * 16 + grp * 4 + (i & 3);
+        bool ok = (kg < nk) & (qrow < nq);
+        if (causal) ok &= kg <= qrow + diag;
+        if (key_mask != nullptr && ok) ok &= key_mask[(long)batch * nk + kg];
+        if (static_mask != nullptr && ok)
+          ok &= Mtile[wave * 16 + lq][kg - kbase] != 0;
+        s16[i] = ok ? s16[i] * scale : NEG_INF;
+      }
     }
 
     // ---- online softmax update (lane-local + 4-lane reduce per q row)
@@ -453,6 +472,12 @@ void fa_bwd_dq_kernel(
     while (t < ntiles && !tile_live(t)) ++t;
     return t;
   };
+  auto tile_full = [&](int t) -> bool {
+    if (!tmap_row) return true;
+    const int g0 = 2 * t;
+    if (g0 + 1 >= ntk) return false;
+    return tmap_row[g0] == 2 && tmap_row[g0 + 1] == 2;
+  };
 
   const int srow0 = tid >> 3;
   const int sc8 = (tid & 7) * 8;
@@ -532,18 +557,33 @@ void fa_bwd_dq_kernel(
     }
     __builtin_amdgcn_s_setprio(0);
 
-    #pragma unroll
-    for (int i = 0; i < 16; ++i) {
-      const int kg = kbase + (i >> 2) * 16 + grp * 4 + (i & 3);
-      bool ok = (kg < nk) & (qrow < nq);
-      if (causal) ok &= kg <= qrow + diag;
-      if (key_mask != nullptr && ok) ok &= key_mask[(long)batch * nk + kg];
-      if (static_mask != nullptr && ok)
-        ok &= Mtile[wave * 16 + lq][kg - kbase] != 0;
-      const float p = ok ? __expf(s16[i] * scale - lse_q) : 0.f;
-      const float ds = p * (dp16[i] - D_q) * scale;
-      const int kk = (i >> 2) * 16 + grp * 4 + (i & 3);
-      DSl[wave][lq][kk] = f2bf(ds);
+    const bool interior =
+        (kbase + KV <= nk) && (q0 + FA_QBLK <= nq) &&
+        (!causal || (kbase + KV - 1 <= q0 + diag)) &&
+        key_mask == nullptr &&
+        (static_mask == nullptr || tile_full(kt));
+    if (interior) {
+      #pragma unroll
+      for (int i = 0; i < 16; ++i) {
+        const float p = __expf(s16[i] * scale - lse_q);
+        const float ds = p * (dp16[i] - D_q) * scale;
+        const int kk = (i >> 2) * 16 + grp * 4 + (i & 3);
+        DSl[wave][lq][kk] = f2bf(ds);
+      }
+    } else {
+      #pragma unroll
+      for (int i = 0; i < 16; ++i) {
+        const int kg = kbase + (i >> 2) * 16 + grp * 4 + (i & 3);
+        bool ok = (kg < nk) & (qrow < nq);
+        if (causal) ok &= kg <= qrow + diag;
+        if (key_mask != nullptr && ok) ok &= key_mask[(long)batch * nk + kg];
+        if (static_mask != nullptr && ok)
+          ok &= Mtile[wave * 16 + lq][kg - kbase] != 0;
+        const float p = ok ? __expf(s16[i] * scale - lse_q) : 0.f;
+        const float ds = p * (dp16[i] - D_q) * scale;
+        const int kk = (i >> 2) * 16 + grp * 4 + (i & 3);
+        DSl[wave][lq][kk] = f2bf(ds);
+      }
     }
 
     bf16x8 dsf0 = frag_from_lds(&DSl[wave][lq][8 * grp]);
@@ -655,6 +695,12 @@ void fa_bwd_dkv_kernel(
     while (t < nqt && !tile_live(t)) ++t;
     return t;
   };
+  auto tile_full = [&](int t) -> bool {
+    if (!tmap_row) return true;
+    const int g0 = 2 * t;
+    if (g0 + 1 >= nqg) return false;
+    return tmap_row[g0] == 2 && tmap_row[g0 + 1] == 2;
+  };
 
   const int srow0 = tid >> 3;
   const int sc8 = (tid & 7) * 8;
@@ -739,26 +785,47 @@ void fa_bwd_dkv_kernel(
     }
     __builtin_amdgcn_s_setprio(0);
 
-    #pragma unroll
-    for (int mt = 0; mt < 4; ++mt) {
+    const bool interior =
+        (qbase + KV <= nq) && (k0 + FA_QBLK <= nk) &&
+        (!causal || (k0 + FA_QBLK - 1 <= qbase + diag)) &&
+        key_mask == nullptr &&
+        (static_mask == nullptr || tile_full(qt));
+    if (interior) {
       #pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        const int key = k0 + wave * 16 + grp * 4 + r;
+      for (int mt = 0; mt < 4; ++mt) {
         const int qg = qbase + mt * 16 + lq;
-        bool ok = (key < nk) & (qg < nq);
-        if (causal) ok &= key <= qg + diag;
-        if (key_mask != nullptr && ok) ok &= key_mask[(long)batch * nk + key];
-        if (static_mask != nullptr && ok)
-          ok &= Mtile[mt * 16 + lq][key - k0] != 0;
-        float p = 0.f, ds = 0.f;
-        if (ok) {
-          const float l = lse[(long)bh * nq + qg];
-          const float Dq = Dv[(long)bh * nq + qg];
-          p = __expf(st4[mt][r] * scale - l);
-          ds = p * (dpt4[mt][r] - Dq) * scale;
+        const float l = lse[(long)bh * nq + qg];
+        const float Dq = Dv[(long)bh * nq + qg];
+        #pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const float p = __expf(st4[mt][r] * scale - l);
+          const float ds = p * (dpt4[mt][r] - Dq) * scale;
+          Pt[wave][grp * 4 + r][mt * 16 + lq] = f2bf(p);
+          DSt[wave][grp * 4 + r][mt * 16 + lq] = f2bf(ds);
         }
-        Pt[wave][grp * 4 + r][mt * 16 + lq] = f2bf(p);
-        DSt[wave][grp * 4 + r][mt * 16 + lq] = f2bf(ds);
+      }
+    } else {
+      #pragma unroll
+      for (int mt = 0; mt < 4; ++mt) {
+        #pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const int key = k0 + wave * 16 + grp * 4 + r;
+          const int qg = qbase + mt * 16 + lq;
+          bool ok = (key < nk) & (qg < nq);
+          if (causal) ok &= key <= qg + diag;
+          if (key_mask != nullptr && ok) ok &= key_mask[(long)batch * nk + key];
+          if (static_mask != nullptr && ok)
+            ok &= Mtile[mt * 16 + lq][key - k0] != 0;
+          float p = 0.f, ds = 0.f;
+          if (ok) {
+            const float l = lse[(long)bh * nq + qg];
+            const float Dq = Dv[(long)bh * nq + qg];
+            p = __expf(st4[mt][r] * scale - l);
+            ds = p * (dpt4[mt][r] - Dq) * scale;
+          }
+          Pt[wave][grp * 4 + r][mt * 16 + lq] = f2bf(p);
+          DSt[wave][grp * 4 + r][mt * 16 + lq] = f2bf(ds);
+        }
       }
     }
 
