@@ -273,7 +273,7 @@ void fa_fwd_d64_kernel(
         (kbase + KV <= nk) && (q0 + FA_QBLK <= nq) &&
         (!causal || (kbase + KV - 1 <= q0 + diag)) &&
         key_mask == nullptr &&
-        (static_mask == nullptr || tile_full(kt));
+        (static_mask == nullptr || (tmap_row != nullptr && tile_full(kt)));
     if (interior) {
       #pragma unroll
       for (int i = 0; i < 16; ++i) s16[i] *= scale;
@@ -561,7 +561,7 @@ void fa_bwd_dq_kernel(
         (kbase + KV <= nk) && (q0 + FA_QBLK <= nq) &&
         (!causal || (kbase + KV - 1 <= q0 + diag)) &&
         key_mask == nullptr &&
-        (static_mask == nullptr || tile_full(kt));
+        (static_mask == nullptr || (tmap_row != nullptr && tile_full(kt)));
     if (interior) {
       #pragma unroll
       for (int i = 0; i < 16; ++i) {
@@ -789,7 +789,7 @@ void fa_bwd_dkv_kernel(
         (qbase + KV <= nq) && (k0 + FA_QBLK <= nk) &&
         (!causal || (k0 + FA_QBLK - 1 <= qbase + diag)) &&
         key_mask == nullptr &&
-        (static_mask == nullptr || tile_full(qt));
+        (static_mask == nullptr || (tmap_row != nullptr && tile_full(qt)));
     if (interior) {
       #pragma unroll
       for (int mt = 0; mt < 4; ++mt) {
