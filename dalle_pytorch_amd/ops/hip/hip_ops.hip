@@ -23,6 +23,7 @@
 #define DEVFN __device__ __forceinline__
 
 using bf16x8 = __attribute__((ext_vector_type(8))) short;   // MFMA A/B frag
+using bf16x4 = __attribute__((ext_vector_type(4))) short;    // packed P stores
 using f32x4 = __attribute__((ext_vector_type(4))) float;    // MFMA C/D frag
 using int4v = __attribute__((ext_vector_type(4))) int;      // 16B copies
 
@@ -319,11 +320,14 @@ void fa_fwd_d64_kernel(
     }
     l_run = l_run * alpha + lsum;
 
-    // ---- P -> LDS (bf16) to reshape into the PV A-fragment
+    // ---- P -> LDS (bf16) to reshape into the PV A-fragment; each quarter
+    // lands at 4 consecutive kk, so pack into one 8-byte store
     #pragma unroll
-    for (int i = 0; i < 16; ++i) {
-      const int kk = (i >> 2) * 16 + grp * 4 + (i & 3);
-      Pl[wave][lq][kk] = f2bf(p16[i]);
+    for (int a = 0; a < 4; ++a) {
+      bf16x4 pk;
+      #pragma unroll
+      for (int e = 0; e < 4; ++e) pk[e] = f2bf(p16[a * 4 + e]);
+      *reinterpret_cast<bf16x4*>(&Pl[wave][lq][16 * a + 4 * grp]) = pk;
     }
 
     if (!defer) {
@@ -562,13 +566,12 @@ void fa_bwd_dq_kernel(
         (!causal || (kbase + KV - 1 <= q0 + diag)) &&
         key_mask == nullptr &&
         (static_mask == nullptr || (tmap_row != nullptr && tile_full(kt)));
+    float ds16[16];
     if (interior) {
       #pragma unroll
       for (int i = 0; i < 16; ++i) {
         const float p = __expf(s16[i] * scale - lse_q);
-        const float ds = p * (dp16[i] - D_q) * scale;
-        const int kk = (i >> 2) * 16 + grp * 4 + (i & 3);
-        DSl[wave][lq][kk] = f2bf(ds);
+        ds16[i] = p * (dp16[i] - D_q) * scale;
       }
     } else {
       #pragma unroll
@@ -580,10 +583,15 @@ void fa_bwd_dq_kernel(
         if (static_mask != nullptr && ok)
           ok &= Mtile[wave * 16 + lq][kg - kbase] != 0;
         const float p = ok ? __expf(s16[i] * scale - lse_q) : 0.f;
-        const float ds = p * (dp16[i] - D_q) * scale;
-        const int kk = (i >> 2) * 16 + grp * 4 + (i & 3);
-        DSl[wave][lq][kk] = f2bf(ds);
+        ds16[i] = p * (dp16[i] - D_q) * scale;
       }
+    }
+    #pragma unroll
+    for (int a = 0; a < 4; ++a) {
+      bf16x4 pk;
+      #pragma unroll
+      for (int e = 0; e < 4; ++e) pk[e] = f2bf(ds16[a * 4 + e]);
+      *reinterpret_cast<bf16x4*>(&DSl[wave][lq][16 * a + 4 * grp]) = pk;
     }
 
     bf16x8 dsf0 = frag_from_lds(&DSl[wave][lq][8 * grp]);
