@@ -91,3 +91,17 @@ def rotate_checkpoints(directory, pattern, keep_n):
             old.unlink()
         except OSError:
             pass
+
+
+def save_clip_checkpoint(path, clip, clip_params):
+    torch.save({'hparams': clip_params, 'weights': clip.state_dict()}, path)
+
+
+def load_clip_checkpoint(path, map_location='cpu'):
+    path = Path(path)
+    assert path.exists(), f'CLIP checkpoint {path} does not exist'
+    ckpt = torch.load(str(path), map_location=map_location, weights_only=False)
+    from dalle_pytorch_amd import CLIP
+    clip = CLIP(**ckpt['hparams'])
+    clip.load_state_dict(ckpt['weights'])
+    return clip, ckpt['hparams']

@@ -39,6 +39,10 @@ def parse_args(argv=None):
     p.add_argument('--no_fast', action='store_true',
                    help='disable the static-shape decode engine (HIP graphs)')
     p.add_argument('--cond_scale', type=float, default=1.0)
+    p.add_argument('--clip_path', type=str, default=None,
+                   help='trained CLIP checkpoint: re-rank generations and '
+                        'keep the best --num_images of an oversampled batch')
+    p.add_argument('--clip_oversample', type=int, default=2)
     return p.parse_args(argv)
 
 
@@ -61,6 +65,12 @@ def main(argv=None):
     args = parse_args(argv)
     device = torch.device('cuda:0') if torch.cuda.is_available() else torch.device('cpu')
     tok = get_tokenizer(args)
+
+    clip = None
+    if args.clip_path:
+        from dalle_pytorch_amd.utils.checkpoint import load_clip_checkpoint
+        clip, _ = load_clip_checkpoint(args.clip_path)
+        clip = clip.to(device).eval()
 
     ckpt = load_dalle_checkpoint(args.dalle_path)
     vae = None
@@ -108,6 +118,14 @@ def main(argv=None):
                     chunk, filter_thres=args.top_k, temperature=args.temperature,
                     use_cache=not args.no_cache, cond_scale=args.cond_scale))
         images = torch.cat(images, dim=0)
+
+        if clip is not None:
+            # CLIP re-rank (reference generate path: dalle_pytorch.py:558-560)
+            with torch.no_grad():
+                scores = clip(text_tokens[:images.shape[0], :clip.text_pos_emb.num_embeddings],
+                              images, return_loss=False)
+            keep = scores.argsort(descending=True)[:args.num_images]
+            images = images[keep]
 
         subdir = out_root / raw_text.replace(' ', '_')[:100]
         subdir.mkdir(parents=True, exist_ok=True)

@@ -158,3 +158,21 @@ def test_bench_distributed_launch_cpu():
     assert rec['n_gpus'] == 2
     assert rec['config']['parallelism'] == 'dp2'
     assert rec['value'] > 0
+
+
+def test_clip_train_and_rerank(tmp_path):
+    """CLIP example trains (loss falls) and the saved checkpoint loads for
+    re-ranking."""
+    sys.path.insert(0, str(REPO / 'examples'))
+    import train_clip
+    out = tmp_path / 'clip.pt'
+    first, last = train_clip.main(['--steps', '40', '--dim', '64',
+                                   '--depth', '1', '--batch_size', '16',
+                                   '--out', str(out)])
+    assert last < first
+    from dalle_pytorch_amd.utils.checkpoint import load_clip_checkpoint
+    clip, _ = load_clip_checkpoint(out)
+    text = torch.randint(0, 100, (4, 16))
+    imgs = torch.rand(4, 3, 32, 32)
+    scores = clip(text, imgs)
+    assert scores.shape == (4,)
