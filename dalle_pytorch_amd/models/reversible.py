@@ -77,26 +77,23 @@ class Deterministic(nn.Module):
 
 class ReversibleBlock(nn.Module):
     """y1 = x1 + f(x2); y2 = x2 + g(y1). Backward reconstructs x from y and
-    re-runs f,g once each (reference reversible.py:54-106)."""
+    re-runs f,g once each (reference reversible.py:54-106). The two streams
+    stay separate tensors end to end — the reference's per-block
+    cat/chunk round-trips are pure copy traffic (~12 GB/step at the
+    flagship shape)."""
 
     def __init__(self, f, g):
         super().__init__()
         self.f = Deterministic(f)
         self.g = Deterministic(g)
 
-    def forward(self, x, f_args={}, g_args={}):
-        x1, x2 = torch.chunk(x, 2, dim=2)
+    def forward(self, x1, x2, f_args={}, g_args={}):
         with torch.no_grad():
             y1 = x1 + self.f(x2, record_rng=self.training, **f_args)
             y2 = x2 + self.g(y1, record_rng=self.training, **g_args)
-        return torch.cat((y1, y2), dim=2)
+        return y1, y2
 
-    def backward_pass(self, y, dy, f_args={}, g_args={}):
-        y1, y2 = torch.chunk(y, 2, dim=2)
-        del y
-        dy1, dy2 = torch.chunk(dy, 2, dim=2)
-        del dy
-
+    def backward_pass(self, y1, y2, dy1, dy2, f_args={}, g_args={}):
         with torch.enable_grad():
             y1.requires_grad = True
             gy1 = self.g(y1, set_rng=True, **g_args)
@@ -120,27 +117,26 @@ class ReversibleBlock(nn.Module):
             dx2 = dy2 + x2.grad
             del dy2
             x2.grad = None
-            x = torch.cat((x1, x2.detach()), dim=2)
-            dx = torch.cat((dx1, dx2), dim=2)
-        return x, dx
+        return x1, x2.detach(), dx1, dx2
 
 
 class _ReversibleFunction(Function):
     @staticmethod
-    def forward(ctx, x, blocks, args):
+    def forward(ctx, x1, x2, blocks, args):
         ctx.args = args
         for block, kw in zip(blocks, args):
-            x = block(x, **kw)
-        ctx.y = x.detach()
+            x1, x2 = block(x1, x2, **kw)
+        ctx.y1 = x1.detach()
+        ctx.y2 = x2.detach()
         ctx.blocks = blocks
-        return x
+        return x1, x2
 
     @staticmethod
-    def backward(ctx, dy):
-        y = ctx.y
+    def backward(ctx, dy1, dy2):
+        y1, y2 = ctx.y1, ctx.y2
         for block, kw in zip(ctx.blocks[::-1], ctx.args[::-1]):
-            y, dy = block.backward_pass(y, dy, **kw)
-        return dy, None, None
+            y1, y2, dy1, dy2 = block.backward_pass(y1, y2, dy1, dy2, **kw)
+        return dy1, dy2, None, None
 
 
 class SequentialSequence(nn.Module):
@@ -170,9 +166,9 @@ class ReversibleSequence(nn.Module):
         self.blocks = nn.ModuleList([ReversibleBlock(f=f, g=g) for f, g in blocks])
 
     def forward(self, x, **kwargs):
-        x = torch.cat((x, x), dim=-1)
         args = route_args(self.args_route, kwargs, len(self.blocks))
         args = [{'f_args': fa, 'g_args': ga} for fa, ga in args]
-        out = _ReversibleFunction.apply(x, self.blocks, args)
-        y1, y2 = out.chunk(2, dim=-1)
+        # duplicated input stream (reference reversible.py:150) without the
+        # cat: the two streams are independent tensors throughout
+        y1, y2 = _ReversibleFunction.apply(x, x.clone(), self.blocks, args)
         return (y1 + y2) / 2
