@@ -21,7 +21,8 @@ from dalle_pytorch_amd.models.attention import (
     Attention, SparseAttention, SparseConvCausalAttention, SparseAxialCausalAttention)
 from dalle_pytorch_amd.models.positional import build_dalle_rotary_table
 from dalle_pytorch_amd.ops import geglu
-from dalle_pytorch_amd.ops.fused import token_shift, token_shift_supported
+from dalle_pytorch_amd.ops.fused import (token_shift, token_shift_supported,
+                                          layer_norm)
 
 
 def _as_tuple(val, depth=1):
@@ -100,7 +101,9 @@ class LayerScale(nn.Module):
 
 
 class PreNorm(nn.Module):
-    """Pre-LN (optionally sandwich) around a layer (reference transformer.py:92-102)."""
+    """Pre-LN (optionally sandwich) around a layer (reference
+    transformer.py:92-102); the norm runs through the fused bf16 kernel on
+    GPU (ops/fused.py)."""
 
     def __init__(self, dim, fn, sandwich=False):
         super().__init__()
@@ -109,7 +112,12 @@ class PreNorm(nn.Module):
         self.fn = fn
 
     def forward(self, x, **kwargs):
-        return self.norm_out(self.fn(self.norm(x), **kwargs))
+        y = layer_norm(x, self.norm.weight, self.norm.bias, self.norm.eps)
+        y = self.fn(y, **kwargs)
+        if isinstance(self.norm_out, nn.LayerNorm):
+            y = layer_norm(y, self.norm_out.weight, self.norm_out.bias,
+                           self.norm_out.eps)
+        return y
 
 
 class GEGLU(nn.Module):

@@ -61,3 +61,33 @@ def token_shift(x, text_len, image_size):
 def token_shift_supported(x):
     return (x.is_cuda and hip_module() is not None
             and x.shape[-1] * x.element_size() % 64 == 0)
+
+
+class _LayerNormFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, weight, bias, eps):
+        ext = hip_module()
+        y, mean, rstd = ext.ln_fwd(x.contiguous(), weight, bias, eps)
+        ctx.save_for_backward(x, weight, mean, rstd)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        ext = hip_module()
+        x, weight, mean, rstd = ctx.saved_tensors
+        dx, dgamma, dbeta = ext.ln_bwd(x, dy.contiguous(), weight, mean, rstd)
+        return dx, dgamma.to(weight.dtype), dbeta.to(weight.dtype), None
+
+
+_LN_DIMS = (512, 1024, 1536, 2048)
+
+
+def layer_norm(x, weight, bias, eps=1e-5):
+    """LayerNorm over the last dim; fused bf16 kernel on GPU for the
+    transformer widths, F.layer_norm elsewhere. Matches autocast semantics
+    (fp32 statistics and affine) with a single bf16 rounding at the output.
+    """
+    if (x.is_cuda and x.dtype == torch.bfloat16 and x.shape[-1] in _LN_DIMS
+            and not using_eager_fallback(x)):
+        return _LayerNormFn.apply(x, weight, bias, eps)
+    return F.layer_norm(x, (x.shape[-1],), weight, bias, eps)

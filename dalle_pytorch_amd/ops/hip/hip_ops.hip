@@ -1301,6 +1301,146 @@ __global__ void shift_decode_kernel(
   out[(long)bi * dim + i] = val;
 }
 
+
+// ---------------------------------------------------------------------------
+// Fused LayerNorm over the last dim (rows of `dim` bf16 values, fp32
+// gamma/beta — the PreNorm hot path, kernel K7). One wave per row,
+// short8-vectorized; saves mean/rstd for the backward. The backward
+// computes dx in one pass and accumulates per-block dgamma/dbeta partials
+// into a [cap, dim] buffer summed by ATen (no atomic contention).
+// ---------------------------------------------------------------------------
+
+template <int PER>
+__global__ __launch_bounds__(256)
+void ln_fwd_kernel(const short* __restrict__ x, const float* __restrict__ w,
+                   const float* __restrict__ bia, short* __restrict__ y,
+                   float* __restrict__ mean_out, float* __restrict__ rstd_out,
+                   long rows, float eps) {
+  constexpr int per = PER;
+  constexpr int dim = PER * 64;
+  const long row = blockIdx.x * 4 + (threadIdx.x >> 6);
+  const int lane = threadIdx.x & 63;
+  if (row >= rows) return;
+  const short* xr = x + row * (long)dim;
+
+  float sum = 0.f, sq = 0.f;
+  float vals[PER];
+  #pragma unroll
+  for (int i = 0; i < per; i += 8) {
+    int4v v = *reinterpret_cast<const int4v*>(xr + lane * per + i);
+    const short* vs = reinterpret_cast<const short*>(&v);
+    #pragma unroll
+    for (int e = 0; e < 8; ++e) {
+      const float f = bf2f(vs[e]);
+      vals[i + e] = f;
+      sum += f;
+      sq += f * f;
+    }
+  }
+  #pragma unroll
+  for (int sft = 32; sft > 0; sft >>= 1) {
+    sum += __shfl_xor(sum, sft);
+    sq += __shfl_xor(sq, sft);
+  }
+  const float mu = sum / dim;
+  const float var = sq / dim - mu * mu;
+  const float rstd = __frsqrt_rn(var + eps);
+  if (lane == 0) {
+    mean_out[row] = mu;
+    rstd_out[row] = rstd;
+  }
+  short* yr = y + row * (long)dim;
+  #pragma unroll
+  for (int i = 0; i < per; i += 8) {
+    short out8[8];
+    #pragma unroll
+    for (int e = 0; e < 8; ++e) {
+      const int d = lane * per + i + e;
+      out8[e] = f2bf((vals[i + e] - mu) * rstd * w[d] + bia[d]);
+    }
+    *reinterpret_cast<int4v*>(yr + lane * per + i) =
+        *reinterpret_cast<const int4v*>(out8);
+  }
+}
+
+template <int PER>
+__global__ __launch_bounds__(256)
+void ln_bwd_kernel(const short* __restrict__ x, const short* __restrict__ dy,
+                   const float* __restrict__ w,
+                   const float* __restrict__ mean_in,
+                   const float* __restrict__ rstd_in,
+                   short* __restrict__ dx,
+                   float* __restrict__ dgamma_part,   // [cap, dim]
+                   float* __restrict__ dbeta_part,
+                   long rows) {
+  constexpr int per = PER;
+  constexpr int dim = PER * 64;
+  const int block_row0 = blockIdx.x * 4;
+  const int wrow = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  const int gridrows = gridDim.x * 4;
+
+  // per-thread dgamma/dbeta partials over this block's strided rows
+  float dgp[PER], dbp[PER];
+  #pragma unroll
+  for (int i = 0; i < per; ++i) { dgp[i] = 0.f; dbp[i] = 0.f; }
+
+  for (long row = block_row0 + wrow; row < rows; row += gridrows) {
+    const short* xr = x + row * dim;
+    const short* dyr = dy + row * dim;
+    const float mu = mean_in[row];
+    const float rstd = rstd_in[row];
+
+    float xh[PER], g[PER];
+    float s1 = 0.f, s2 = 0.f;
+    #pragma unroll
+    for (int i = 0; i < per; i += 8) {
+      int4v xv = *reinterpret_cast<const int4v*>(xr + lane * per + i);
+      int4v dv = *reinterpret_cast<const int4v*>(dyr + lane * per + i);
+      const short* xs = reinterpret_cast<const short*>(&xv);
+      const short* ds_ = reinterpret_cast<const short*>(&dv);
+      #pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        const int d = lane * per + i + e;
+        const float xhat = (bf2f(xs[e]) - mu) * rstd;
+        const float dyv = bf2f(ds_[e]);
+        const float gv = dyv * w[d];
+        xh[i + e] = xhat;
+        g[i + e] = gv;
+        s1 += gv;
+        s2 += gv * xhat;
+        dgp[i + e] += dyv * xhat;
+        dbp[i + e] += dyv;
+      }
+    }
+    #pragma unroll
+    for (int sft = 32; sft > 0; sft >>= 1) {
+      s1 += __shfl_xor(s1, sft);
+      s2 += __shfl_xor(s2, sft);
+    }
+    s1 /= dim;
+    s2 /= dim;
+    short* dxr = dx + row * (long)dim;
+    #pragma unroll
+    for (int i = 0; i < per; i += 8) {
+      short out8[8];
+      #pragma unroll
+      for (int e = 0; e < 8; ++e)
+        out8[e] = f2bf((g[i + e] - s1 - xh[i + e] * s2) * rstd);
+      *reinterpret_cast<int4v*>(dxr + lane * per + i) =
+          *reinterpret_cast<const int4v*>(out8);
+    }
+  }
+  // one partial row per (block, wave-row slot): slot = blockIdx.x*4 + wrow
+  float* dgr = dgamma_part + ((long)blockIdx.x * 4 + wrow) * dim;
+  float* dbr = dbeta_part + ((long)blockIdx.x * 4 + wrow) * dim;
+  #pragma unroll
+  for (int i = 0; i < per; ++i) {
+    dgr[lane * per + i] = dgp[i];
+    dbr[lane * per + i] = dbp[i];
+  }
+}
+
 // ---------------------------------------------------------------------------
 // MFMA layout probe (test support): one 16x16x32 bf16 MFMA with the exact
 // fragment mappings the attention kernel assumes. The GPU test compares
@@ -1626,6 +1766,70 @@ torch::Tensor mfma_probe(torch::Tensor A, torch::Tensor B) {
   return C;
 }
 
+
+std::vector<torch::Tensor> ln_fwd(torch::Tensor x, torch::Tensor w,
+                                  torch::Tensor b, double eps) {
+  CHK(x.is_cuda() && x.dtype() == torch::kBFloat16 && x.is_contiguous());
+  const int dim = x.size(-1);
+
+  const long rows = x.numel() / dim;
+  auto y = torch::empty_like(x);
+  auto mean = torch::empty({rows}, x.options().dtype(torch::kFloat32));
+  auto rstd = torch::empty_like(mean);
+  auto wf = w.to(torch::kFloat32).contiguous();
+  auto bf_ = b.to(torch::kFloat32).contiguous();
+  dim3 grid((rows + 3) / 4);
+  auto launch = [&](auto kern) {
+    hipLaunchKernelGGL(kern, grid, dim3(256), 0, cur_stream(),
+                       reinterpret_cast<const short*>(x.data_ptr()),
+                       wf.data_ptr<float>(), bf_.data_ptr<float>(),
+                       reinterpret_cast<short*>(y.data_ptr()),
+                       mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                       rows, (float)eps);
+  };
+  switch (dim) {
+    case 512: launch(ln_fwd_kernel<8>); break;
+    case 1024: launch(ln_fwd_kernel<16>); break;
+    case 1536: launch(ln_fwd_kernel<24>); break;
+    case 2048: launch(ln_fwd_kernel<32>); break;
+    default: TORCH_CHECK(false, "ln_fwd: unsupported dim ", dim);
+  }
+  return {y, mean, rstd};
+}
+
+std::vector<torch::Tensor> ln_bwd(torch::Tensor x, torch::Tensor dy,
+                                  torch::Tensor w, torch::Tensor mean,
+                                  torch::Tensor rstd) {
+  CHK(x.is_cuda() && x.is_contiguous() && dy.is_contiguous());
+  const int dim = x.size(-1);
+  const long rows = x.numel() / dim;
+  auto dx = torch::empty_like(x);
+  const long cap = std::min<long>((rows + 3) / 4, 512);
+  auto dgp = torch::empty({cap * 4, (long)dim}, x.options().dtype(torch::kFloat32));
+  auto dbp = torch::empty_like(dgp);
+  auto wf = w.to(torch::kFloat32).contiguous();
+  dim3 grid(cap);
+  auto launch = [&](auto kern) {
+    hipLaunchKernelGGL(kern, grid, dim3(256), 0, cur_stream(),
+                       reinterpret_cast<const short*>(x.data_ptr()),
+                       reinterpret_cast<const short*>(dy.data_ptr()),
+                       wf.data_ptr<float>(), mean.data_ptr<float>(),
+                       rstd.data_ptr<float>(),
+                       reinterpret_cast<short*>(dx.data_ptr()),
+                       dgp.data_ptr<float>(), dbp.data_ptr<float>(), rows);
+  };
+  switch (dim) {
+    case 512: launch(ln_bwd_kernel<8>); break;
+    case 1024: launch(ln_bwd_kernel<16>); break;
+    case 1536: launch(ln_bwd_kernel<24>); break;
+    case 2048: launch(ln_bwd_kernel<32>); break;
+    default: TORCH_CHECK(false, "ln_bwd: unsupported dim ", dim);
+  }
+  auto dgamma = dgp.sum(0);
+  auto dbeta = dbp.sum(0);
+  return {dx, dgamma, dbeta};
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("fa_fwd", &fa_fwd, "flash attention forward (gfx950, d=64)");
   m.def("fa_bwd", &fa_bwd, "flash attention backward (gfx950, d=64)");
@@ -1635,6 +1839,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("token_shift", &token_shift, "fused token shift (fwd/transpose)");
   m.def("fa_decode", &fa_decode, "fused single-token decode attention");
   m.def("shift_decode", &shift_decode, "fused single-token token shift");
+  m.def("ln_fwd", &ln_fwd, "fused LayerNorm forward (bf16 rows)");
+  m.def("ln_bwd", &ln_bwd, "fused LayerNorm backward");
   m.def("geglu_fwd", &geglu_fwd, "fused GEGLU forward");
   m.def("geglu_bwd", &geglu_bwd, "fused GEGLU backward");
   m.def("mfma_probe", &mfma_probe, "MFMA 16x16x32 bf16 layout probe");

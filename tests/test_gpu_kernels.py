@@ -373,3 +373,30 @@ def test_dvae_gemm_encoder_matches_conv():
     codes = vae.get_codebook_indices(img)
     ref_codes = ref.argmax(dim=1).flatten(1)
     assert (codes == ref_codes).float().mean().item() > 0.99
+
+
+def test_layer_norm_vs_oracle(ext):
+    torch.manual_seed(11)
+    from dalle_pytorch_amd.ops.fused import layer_norm
+    x0 = torch.randn(300, 1024, device='cuda') * 2 + 0.5
+    w0 = torch.randn(1024, device='cuda')
+    b0 = torch.randn(1024, device='cuda')
+    dy = torch.randn(300, 1024, device='cuda')
+
+    x = x0.bfloat16().requires_grad_()
+    w = w0.clone().requires_grad_()
+    b = b0.clone().requires_grad_()
+    y = layer_norm(x, w, b)
+    y.backward(dy.bfloat16())
+
+    xr = x0.clone().requires_grad_()
+    wr = w0.clone().requires_grad_()
+    br = b0.clone().requires_grad_()
+    yr = torch.nn.functional.layer_norm(xr, (1024,), wr, br)
+    yr.backward(dy)
+
+    assert (y.float() - yr).abs().max() / yr.abs().max() < 2e-2
+    for got, want, name in ((x.grad, xr.grad, 'dx'), (w.grad, wr.grad, 'dw'),
+                            (b.grad, br.grad, 'db')):
+        rel = (got.float() - want).abs().max() / want.abs().max().clamp(min=1e-6)
+        assert rel < 2e-2, f'{name} {rel.item()}'
