@@ -29,6 +29,8 @@ import torch
 import torch.nn.functional as F
 from torch import nn
 
+from dalle_pytorch_amd.ops.fused import layer_norm as fused_layer_norm
+
 from dalle_pytorch_amd.models import attention as attn_mod
 from dalle_pytorch_amd.models.transformer import CachedAs, NonCached, PreShiftToken
 from dalle_pytorch_amd.models.dalle import top_k, gumbel_sample
@@ -292,8 +294,8 @@ class FastDecoder:
 
     def _branch(self, st, x, offset_t, n):
         dim = x.shape[-1]
-        y = F.layer_norm(x, (dim,), st.w['ln_w'], st.w['ln_b'],
-                         st.info['norm'].eps)
+        y = fused_layer_norm(x, st.w['ln_w'], st.w['ln_b'],
+                             st.info['norm'].eps)
         if st.info['shift'] is not None:
             y = self._shift_prefill(st, y, n) if n > 1 else \
                 self._shift_decode(st, y, offset_t)
