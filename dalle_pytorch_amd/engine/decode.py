@@ -157,12 +157,16 @@ class FastDecoder:
         and replayed for every token (measured ~30% of replay time)."""
         cast = lambda t: None if t is None else \
             t.detach().to(self.device, self.dtype)
-        w = {'ln_w': cast(st.info['norm'].weight),
-             'ln_b': cast(st.info['norm'].bias),
+        castf = lambda t: None if t is None else \
+            t.detach().to(self.device, torch.float32)
+        # LN weights stay fp32: the fused kernel takes fp32 affine directly
+        # (a bf16 copy would be re-cast on every graph replay)
+        w = {'ln_w': castf(st.info['norm'].weight),
+             'ln_b': castf(st.info['norm'].bias),
              'scale': cast(st.info['scale'])}
         if st.info['norm_out'] is not None:
-            w['lno_w'] = cast(st.info['norm_out'].weight)
-            w['lno_b'] = cast(st.info['norm_out'].bias)
+            w['lno_w'] = castf(st.info['norm_out'].weight)
+            w['lno_b'] = castf(st.info['norm_out'].bias)
         if st.is_attn:
             w['qkv'] = cast(st.leaf.to_qkv.weight)
             w['out_w'] = cast(st.leaf.to_out[0].weight)
