@@ -90,4 +90,7 @@ def layer_norm(x, weight, bias, eps=1e-5):
     if (x.is_cuda and x.dtype == torch.bfloat16 and x.shape[-1] in _LN_DIMS
             and not using_eager_fallback(x)):
         return _LayerNormFn.apply(x, weight, bias, eps)
+    if weight is not None and weight.dtype != x.dtype and x.dtype != torch.float32:
+        weight = weight.to(x.dtype)
+        bias = bias.to(x.dtype) if bias is not None else None
     return F.layer_norm(x, (x.shape[-1],), weight, bias, eps)
