@@ -808,8 +808,9 @@ void fa_bwd_dkv_kernel(
         for (int r = 0; r < 4; ++r) {
           const float p = __expf(st4[mt][r] * scale - l);
           const float ds = p * (dpt4[mt][r] - Dq) * scale;
-          Pt[wave][grp * 4 + r][mt * 16 + lq] = f2bf(p);
-          DSt[wave][grp * 4 + r][mt * 16 + lq] = f2bf(ds);
+          const int cc = (mt * 16 + lq) ^ (grp << 3);   // swz by row>>2
+          Pt[wave][grp * 4 + r][cc] = f2bf(p);
+          DSt[wave][grp * 4 + r][cc] = f2bf(ds);
         }
       }
     } else {
@@ -831,16 +832,18 @@ void fa_bwd_dkv_kernel(
             p = __expf(st4[mt][r] * scale - l);
             ds = p * (dpt4[mt][r] - Dq) * scale;
           }
-          Pt[wave][grp * 4 + r][mt * 16 + lq] = f2bf(p);
-          DSt[wave][grp * 4 + r][mt * 16 + lq] = f2bf(ds);
+          const int cc = (mt * 16 + lq) ^ (grp << 3);   // swz by row>>2
+          Pt[wave][grp * 4 + r][cc] = f2bf(p);
+          DSt[wave][grp * 4 + r][cc] = f2bf(ds);
         }
       }
     }
 
-    bf16x8 pf0 = frag_from_lds(&Pt[wave][lq][8 * grp]);
-    bf16x8 pf1 = frag_from_lds(&Pt[wave][lq][32 + 8 * grp]);
-    bf16x8 dsf0 = frag_from_lds(&DSt[wave][lq][8 * grp]);
-    bf16x8 dsf1 = frag_from_lds(&DSt[wave][lq][32 + 8 * grp]);
+    const int rsw = ((lq >> 2) & 7) << 3;               // same swz, row = lq
+    bf16x8 pf0 = frag_from_lds(&Pt[wave][lq][(8 * grp) ^ rsw]);
+    bf16x8 pf1 = frag_from_lds(&Pt[wave][lq][(32 + 8 * grp) ^ rsw]);
+    bf16x8 dsf0 = frag_from_lds(&DSt[wave][lq][(8 * grp) ^ rsw]);
+    bf16x8 dsf1 = frag_from_lds(&DSt[wave][lq][(32 + 8 * grp) ^ rsw]);
     __builtin_amdgcn_s_setprio(1);
     #pragma unroll
     for (int nt = 0; nt < 4; ++nt) {
