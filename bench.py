@@ -131,7 +131,7 @@ def main():
             loss = dalle(text, images, return_loss=True)
         loss.backward()
         engine.finish_gradient_sync()
-        torch.nn.utils.clip_grad_norm_(dalle.parameters(), 0.5)
+        engine.clip_grad_norm_(0.5)
         opt.step()
         engine.zero_grad()
         return loss

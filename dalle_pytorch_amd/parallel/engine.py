@@ -191,6 +191,24 @@ class DataParallelEngine:
             if self.grad_average:
                 b.flat.div_(self.world_size)
 
+    def clip_grad_norm_(self, max_norm, eps=1e-6):
+        """Global grad-norm clip computed on the flat buckets: one norm and
+        one scale kernel per 64 MiB bucket instead of the per-parameter
+        reduce kernels ``torch.nn.utils.clip_grad_norm_`` launches (~60 per
+        step at the flagship param count). Semantics match torch's default
+        (L2 norm over all grads, coef = max_norm/(total+eps) clamped to 1).
+        """
+        if not self._buckets:
+            return torch.nn.utils.clip_grad_norm_(
+                self.model.parameters(), max_norm)
+        norms = torch.stack(
+            [torch.linalg.vector_norm(b.flat, 2) for b in self._buckets])
+        total = torch.linalg.vector_norm(norms, 2)
+        coef = (max_norm / (total + eps)).clamp(max=1.0)
+        for b in self._buckets:
+            b.flat.mul_(coef)
+        return total
+
     def zero_grad(self):
         for b in self._buckets:
             b.flat.zero_()

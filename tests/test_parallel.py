@@ -120,6 +120,35 @@ def test_single_process_noop():
         assert torch.allclose(a, b)
 
 
+def test_bucket_clip_grad_norm_matches_torch():
+    """engine.clip_grad_norm_ (per-bucket norms) == torch's per-param clip."""
+    torch.manual_seed(7)
+    x = torch.randn(4, 8)
+
+    m1 = _model(seed=3)
+    eng = DataParallelEngine(m1, bucket_bytes=1 << 10)
+    (m1(x) ** 2).sum().backward()
+    eng.finish_gradient_sync()
+    total = eng.clip_grad_norm_(0.05)
+
+    m2 = _model(seed=3)
+    (m2(x) ** 2).sum().backward()
+    ref_total = torch.nn.utils.clip_grad_norm_(m2.parameters(), 0.05)
+
+    assert torch.allclose(total, ref_total, rtol=1e-6)
+    for a, b in zip(m1.parameters(), m2.parameters()):
+        assert torch.allclose(a.grad, b.grad, rtol=1e-6, atol=1e-7)
+
+    # above the max norm nothing is scaled
+    m3 = _model(seed=3)
+    eng3 = DataParallelEngine(m3, bucket_bytes=1 << 10)
+    (m3(x) ** 2).sum().backward()
+    g0 = [p.grad.clone() for p in m3.parameters()]
+    eng3.clip_grad_norm_(1e9)
+    for a, p in zip(g0, m3.parameters()):
+        assert torch.allclose(a, p.grad)
+
+
 def test_distributed_utils_facade():
     """Reference-surface facade maps onto the RCCL engine."""
     import argparse

@@ -17,7 +17,6 @@ import time
 from pathlib import Path
 
 import torch
-from torch.nn.utils import clip_grad_norm_
 from torch.optim import Adam
 from torch.optim.lr_scheduler import ReduceLROnPlateau
 from torch.utils.data import DataLoader
@@ -304,7 +303,7 @@ def main(argv=None):
             if accum_boundary:
                 engine.finish_gradient_sync()
                 if args.clip_grad_norm:
-                    clip_grad_norm_(dalle.parameters(), args.clip_grad_norm)
+                    engine.clip_grad_norm_(args.clip_grad_norm)
                 opt.step()
                 engine.zero_grad()
 
