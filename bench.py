@@ -96,6 +96,8 @@ def main():
 
     from dalle_pytorch_amd.parallel import init_distributed, barrier
     from dalle_pytorch_amd.parallel import DataParallelEngine
+    from dalle_pytorch_amd.utils.tunable import maybe_enable_tunableop
+    maybe_enable_tunableop()
 
     rank, world, local_rank = init_distributed()
     use_cuda = torch.cuda.is_available()

@@ -273,6 +273,65 @@ class SparseAxialCausalAttention(_StaticMaskSparseAttention):
     def _build_mask(self):
         return axial_mask(self.seq_len, self.text_len, self.image_size, self.axis)
 
+    # -- column-major permuted evaluation (axis=1 full-length sequences) ----
+    #
+    # Column keys sit 32 positions apart in sequence order, so under the
+    # tiled kernel EVERY 32-key image tile of an axial_col layer is
+    # "partial" with ~1 live key — the tile scan, LDS staging and MFMA all
+    # run at ~3% utility (measured: full dense attention beat axial at the
+    # flagship shape). Re-ordering the image tokens column-major makes each
+    # query's live keys contiguous, so the pattern tiles exactly like
+    # axial_row: ~10 live tiles per query block instead of ~40. Causality is
+    # folded into the permuted static mask (causal=False in the kernel);
+    # rotary embedding is applied before permuting, so every q·k pair is
+    # identical to the unpermuted computation.
+
+    def _perm(self, n, device):
+        key = ('axial_colperm_idx', self.seq_len, self.image_size)
+        hit = _MASK_CACHE.get((*key, str(device)))
+        if hit is None:
+            S, t = self.image_size, self.text_len
+            img = torch.arange(S * S).reshape(S, S).t().reshape(-1)
+            # the transformer sequence drops the final token, so the image
+            # part is S*S-1 long (or shorter) — keep only present slots
+            img = img[img < n - t]
+            perm = torch.cat([torch.arange(t), t + img]).to(device)
+            inv = torch.empty_like(perm)
+            inv[perm] = torch.arange(n, device=device)
+            hit = (perm, inv)
+            _MASK_CACHE[(*key, str(device))] = hit
+        return hit
+
+    def _permuted_pattern(self, device):
+        def build():
+            m = axial_mask(self.seq_len, self.text_len, self.image_size, 1)
+            m &= torch.ones(self.seq_len, self.seq_len, dtype=torch.bool).tril_()
+            perm, _ = self._perm(self.seq_len, torch.device('cpu'))
+            return m[perm][:, perm]
+        return _cached_mask(('axial_colperm', self.seq_len, self.image_size),
+                            device, build)
+
+    def forward(self, x, mask=None, rotary_pos_emb=None):
+        b, n, _ = x.shape
+        if self.axis != 1 or n != self.seq_len:
+            return super().forward(x, mask=mask, rotary_pos_emb=rotary_pos_emb)
+        q, k, v = _qkv_heads(x, self.to_qkv, self.heads, self.dim_head,
+                             rotary_pos_emb, 0)
+        perm, inv = self._perm(n, x.device)
+        q = q.index_select(2, perm)
+        k = k.index_select(2, perm)
+        v = v.index_select(2, perm)
+        static, tiles, tiles_t = self._permuted_pattern(x.device)
+        km = self._key_mask(mask, b, n, self.text_len, x.device)
+        if km is not None:
+            km = km.index_select(1, perm)
+        out = attention_core(q, k, v, self.scale, causal=False,
+                             key_mask=km, static_mask=static,
+                             static_tiles=tiles, static_tiles_t=tiles_t,
+                             fold_heads=True)
+        out = out.index_select(1, inv)
+        return self.to_out(out)
+
 
 class SparseConvCausalAttention(_StaticMaskSparseAttention):
     """Conv-like sparse attention: each image token attends to a causally

@@ -130,6 +130,8 @@ def get_tokenizer(args):
 
 def main(argv=None):
     args = parse_args(argv)
+    from dalle_pytorch_amd.utils.tunable import maybe_enable_tunableop
+    maybe_enable_tunableop()
     rank, world, local_rank = init_distributed()
     is_root = rank == 0
     device = torch.device(f'cuda:{local_rank}') if torch.cuda.is_available() \
