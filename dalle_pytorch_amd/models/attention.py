@@ -25,6 +25,7 @@ Semantics notes carried over from the reference (kept bit-for-bit):
 from math import ceil
 
 import torch
+import torch.nn.functional as F
 from torch import nn
 
 from dalle_pytorch_amd.models.positional import apply_rotary_to_qkv
@@ -286,27 +287,31 @@ class SparseAxialCausalAttention(_StaticMaskSparseAttention):
     # rotary embedding is applied before permuting, so every q·k pair is
     # identical to the unpermuted computation.
 
-    def _perm(self, n, device):
-        key = ('axial_colperm_idx', self.seq_len, self.image_size)
-        hit = _MASK_CACHE.get((*key, str(device)))
-        if hit is None:
-            S, t = self.image_size, self.text_len
-            img = torch.arange(S * S).reshape(S, S).t().reshape(-1)
-            # the transformer sequence drops the final token, so the image
-            # part is S*S-1 long (or shorter) — keep only present slots
-            img = img[img < n - t]
-            perm = torch.cat([torch.arange(t), t + img]).to(device)
-            inv = torch.empty_like(perm)
-            inv[perm] = torch.arange(n, device=device)
-            hit = (perm, inv)
-            _MASK_CACHE[(*key, str(device))] = hit
-        return hit
+    def _grid_transpose(self, z):
+        """Transpose the image part of a [..., n, d] tensor between row- and
+        column-major grid order (text prefix untouched). Self-inverse: the
+        grid transpose is an involution, and the sequence's dropped final
+        token maps to the final slot in both orders (r=c=S-1). Implemented
+        as pad+view+transpose+copy so the autograd backward is another
+        transpose copy — no scatter-add atomics."""
+        S, t = self.image_size, self.text_len
+        zt, zi = z[..., :t, :], z[..., t:, :]
+        n_img, d = zi.shape[-2], zi.shape[-1]
+        lead = zi.shape[:-2]
+        if n_img < S * S:
+            zi = F.pad(zi, (0, 0, 0, S * S - n_img))
+        zi = (zi.reshape(*lead, S, S, d).transpose(-3, -2)
+              .reshape(*lead, S * S, d)[..., :n_img, :])
+        return torch.cat((zt, zi), dim=-2)
 
     def _permuted_pattern(self, device):
         def build():
             m = axial_mask(self.seq_len, self.text_len, self.image_size, 1)
             m &= torch.ones(self.seq_len, self.seq_len, dtype=torch.bool).tril_()
-            perm, _ = self._perm(self.seq_len, torch.device('cpu'))
+            S, t = self.image_size, self.text_len
+            img = torch.arange(S * S).reshape(S, S).t().reshape(-1)
+            img = img[img < self.seq_len - t]
+            perm = torch.cat([torch.arange(t), t + img])
             return m[perm][:, perm]
         return _cached_mask(('axial_colperm', self.seq_len, self.image_size),
                             device, build)
@@ -317,19 +322,18 @@ class SparseAxialCausalAttention(_StaticMaskSparseAttention):
             return super().forward(x, mask=mask, rotary_pos_emb=rotary_pos_emb)
         q, k, v = _qkv_heads(x, self.to_qkv, self.heads, self.dim_head,
                              rotary_pos_emb, 0)
-        perm, inv = self._perm(n, x.device)
-        q = q.index_select(2, perm)
-        k = k.index_select(2, perm)
-        v = v.index_select(2, perm)
+        q = self._grid_transpose(q)
+        k = self._grid_transpose(k)
+        v = self._grid_transpose(v)
         static, tiles, tiles_t = self._permuted_pattern(x.device)
+        # text prefix stays in place and image key-mask entries are all True,
+        # so the key mask is invariant under the grid transpose
         km = self._key_mask(mask, b, n, self.text_len, x.device)
-        if km is not None:
-            km = km.index_select(1, perm)
         out = attention_core(q, k, v, self.scale, causal=False,
                              key_mask=km, static_mask=static,
                              static_tiles=tiles, static_tiles_t=tiles_t,
                              fold_heads=True)
-        out = out.index_select(1, inv)
+        out = self._grid_transpose(out)
         return self.to_out(out)
 
 
