@@ -25,7 +25,6 @@ Semantics notes carried over from the reference (kept bit-for-bit):
 from math import ceil
 
 import torch
-import torch.nn.functional as F
 from torch import nn
 
 from dalle_pytorch_amd.models.positional import apply_rotary_to_qkv
@@ -274,67 +273,13 @@ class SparseAxialCausalAttention(_StaticMaskSparseAttention):
     def _build_mask(self):
         return axial_mask(self.seq_len, self.text_len, self.image_size, self.axis)
 
-    # -- column-major permuted evaluation (axis=1 full-length sequences) ----
-    #
-    # Column keys sit 32 positions apart in sequence order, so under the
-    # tiled kernel EVERY 32-key image tile of an axial_col layer is
-    # "partial" with ~1 live key — the tile scan, LDS staging and MFMA all
-    # run at ~3% utility (measured: full dense attention beat axial at the
-    # flagship shape). Re-ordering the image tokens column-major makes each
-    # query's live keys contiguous, so the pattern tiles exactly like
-    # axial_row: ~10 live tiles per query block instead of ~40. Causality is
-    # folded into the permuted static mask (causal=False in the kernel);
-    # rotary embedding is applied before permuting, so every q·k pair is
-    # identical to the unpermuted computation.
-
-    def _grid_transpose(self, z):
-        """Transpose the image part of a [..., n, d] tensor between row- and
-        column-major grid order (text prefix untouched). Self-inverse: the
-        grid transpose is an involution, and the sequence's dropped final
-        token maps to the final slot in both orders (r=c=S-1). Implemented
-        as pad+view+transpose+copy so the autograd backward is another
-        transpose copy — no scatter-add atomics."""
-        S, t = self.image_size, self.text_len
-        zt, zi = z[..., :t, :], z[..., t:, :]
-        n_img, d = zi.shape[-2], zi.shape[-1]
-        lead = zi.shape[:-2]
-        if n_img < S * S:
-            zi = F.pad(zi, (0, 0, 0, S * S - n_img))
-        zi = (zi.reshape(*lead, S, S, d).transpose(-3, -2)
-              .reshape(*lead, S * S, d)[..., :n_img, :])
-        return torch.cat((zt, zi), dim=-2)
-
-    def _permuted_pattern(self, device):
-        def build():
-            m = axial_mask(self.seq_len, self.text_len, self.image_size, 1)
-            m &= torch.ones(self.seq_len, self.seq_len, dtype=torch.bool).tril_()
-            S, t = self.image_size, self.text_len
-            img = torch.arange(S * S).reshape(S, S).t().reshape(-1)
-            img = img[img < self.seq_len - t]
-            perm = torch.cat([torch.arange(t), t + img])
-            return m[perm][:, perm]
-        return _cached_mask(('axial_colperm', self.seq_len, self.image_size),
-                            device, build)
-
-    def forward(self, x, mask=None, rotary_pos_emb=None):
-        b, n, _ = x.shape
-        if self.axis != 1 or n != self.seq_len:
-            return super().forward(x, mask=mask, rotary_pos_emb=rotary_pos_emb)
-        q, k, v = _qkv_heads(x, self.to_qkv, self.heads, self.dim_head,
-                             rotary_pos_emb, 0)
-        q = self._grid_transpose(q)
-        k = self._grid_transpose(k)
-        v = self._grid_transpose(v)
-        static, tiles, tiles_t = self._permuted_pattern(x.device)
-        # text prefix stays in place and image key-mask entries are all True,
-        # so the key mask is invariant under the grid transpose
-        km = self._key_mask(mask, b, n, self.text_len, x.device)
-        out = attention_core(q, k, v, self.scale, causal=False,
-                             key_mask=km, static_mask=static,
-                             static_tiles=tiles, static_tiles_t=tiles_t,
-                             fold_heads=True)
-        out = self._grid_transpose(out)
-        return self.to_out(out)
+    # NOTE (measured, round 1): re-ordering the image tokens column-major so
+    # the axis=1 pattern tiles contiguously was tried in two forms —
+    # index_select (backward scatter-add atomics) and grid-transpose copies.
+    # Both REGRESSED the flagship step (-3% / -5%): three extra tensor
+    # passes per layer-visit cost more than the skipped partial tiles save.
+    # The profitable version needs the permutation folded into rope_split /
+    # the kernel epilogues (zero extra passes) — see NOTES_ROUND2.md.
 
 
 class SparseConvCausalAttention(_StaticMaskSparseAttention):
