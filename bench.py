@@ -49,6 +49,8 @@ def parse_args():
     p.add_argument('--batch_size', type=int, default=None, help='per-GPU batch')
     p.add_argument('--mode', type=str, default='train', choices=['train', 'generate'])
     p.add_argument('--gen_batch', type=int, default=64)
+    p.add_argument('--no_graph', action='store_true',
+                   help='disable HIP-graph decode replay (for kernel profiling)')
     p.add_argument('--eager', action='store_true',
                    help='force eager ops on GPU (baseline comparison)')
     return p.parse_args()
@@ -149,7 +151,7 @@ def main():
         if not args.eager:
             from dalle_pytorch_amd.engine import FastDecoder
             decoder = FastDecoder(dalle, batch_size=args.gen_batch,
-                                  use_graph=use_cuda)
+                                  use_graph=use_cuda and not args.no_graph)
 
     def gen_step(_):
         text = pool[0][0][:1].repeat(args.gen_batch, 1)
