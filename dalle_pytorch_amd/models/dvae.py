@@ -120,39 +120,7 @@ class DiscreteVAE(nn.Module):
         stride-2 4x4 convs as unfold+matmul and the 1x1 convs as plain
         matmuls pins the work to hipBLASLt MFMA GEMMs everywhere.
         """
-        def run(mod, x):
-            if isinstance(mod, nn.Conv2d):
-                b, c, h, w = x.shape
-                kh, kw = mod.kernel_size
-                if kh == 1 and kw == 1:
-                    out = torch.matmul(
-                        x.reshape(b, c, h * w).transpose(1, 2),
-                        mod.weight.reshape(mod.out_channels, c).t())
-                    out = out + mod.bias
-                    return out.transpose(1, 2).reshape(b, -1, h, w)
-                sh, sw = mod.stride
-                oh = (h + 2 * mod.padding[0] - kh) // sh + 1
-                ow = (w + 2 * mod.padding[1] - kw) // sw + 1
-                cols = F.unfold(x, (kh, kw), stride=(sh, sw),
-                                padding=mod.padding)        # [b, c*kh*kw, L]
-                out = torch.matmul(cols.transpose(1, 2),
-                                   mod.weight.reshape(mod.out_channels, -1).t())
-                out = out + mod.bias
-                return out.transpose(1, 2).reshape(b, -1, oh, ow)
-            if isinstance(mod, nn.ReLU):
-                return torch.relu(x)
-            if isinstance(mod, ResBlock):
-                y = x
-                for sub in mod.net:
-                    y = run(sub, y)
-                return y + x
-            if isinstance(mod, nn.Sequential):
-                for sub in mod:
-                    x = run(sub, x)
-                return x
-            return mod(x)
-
-        return run(self.encoder, x)
+        return _run_as_gemms(self.encoder, x)
 
     @torch.no_grad()
     def get_codebook_indices(self, images):

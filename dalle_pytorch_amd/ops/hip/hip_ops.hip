@@ -1152,22 +1152,30 @@ __global__ void token_shift_kernel(
 }
 
 // ---------------------------------------------------------------------------
-// Fused single-token decode attention (guide App. B "attention decode").
+// Fused single-token decode attention, key-split ("flash decoding"; guide
+// App. B "attention decode").
 //
-// One launch per token per layer replaces the eager chain (rope gathers,
-// cache scatter, mask build, masked_fill, softmax, two batched GEMVs,
-// permute): block (bi, head) applies rotary to this token's q/k/v, writes
-// the K/V caches at *offset, computes the pattern+length-masked dots over
-// the cached keys (one wave per key, lane-per-element shuffle reduce),
-// block-wide softmax in LDS, and the P*V accumulation with lane = d so V
-// rows stream coalesced. Memory-bound by design: reads each cached K/V row
-// once.
+// The single-block-per-(b,head) form measured 144 us/dispatch at N=1281 —
+// only b*h=1024 workgroups, each streaming its whole K/V slice through one
+// CU, fully latency-bound (51.5% of generation GPU time, see
+// profiles/decode_kernel_stats). Split the key range over KS extra grid
+// blocks instead: each block softmaxes its chunk locally (lane-per-key
+// dots on 16-byte K chunks — no shuffle reduces) and writes (m, l,
+// acc[64]) partials; a tiny combine kernel merges the KS partials with the
+// standard flash rescaling. b*h*KS blocks fill the chip and every K/V row
+// is still read exactly once.
+//
+// Two dispatches rather than one kernel with an atomic-counter tail: the
+// 8 XCDs have non-coherent L2s, so cross-block partial visibility would
+// need device-scope fences on the hot path; the kernel boundary gives the
+// same guarantee for one extra ~3 us launch.
 // ---------------------------------------------------------------------------
 
 constexpr int DEC_MAXN = 4096;
+constexpr int DEC_CHUNK_MAX = 2048;   // chunk <= N/KS aligned, KS >= 2
 
 __global__ __launch_bounds__(256)
-void fa_decode_kernel(
+void fa_decode_part_kernel(
     const short* __restrict__ qkv,    // [b, 3*h*64] (this token's projection)
     short* __restrict__ kc,           // [b, h, N, 64]
     short* __restrict__ vc,           // [b, h, N, 64]
@@ -1175,20 +1183,22 @@ void fa_decode_kernel(
     const float* __restrict__ sinv,
     const long* __restrict__ offset,  // [1] current position
     const bool* __restrict__ pattern, // [N, N] or null; row = *offset
-    short* __restrict__ out,          // [b, h*64]
-    int b, int h, int N, int rot, float scale) {
+    float* __restrict__ scratch,      // [b, h, KS, 66] = m, l, acc[64]
+    int b, int h, int N, int rot, float scale, int KS, int chunk) {
 
-  __shared__ float P[DEC_MAXN];
-  __shared__ float qs[64];
+  __shared__ float qs[64], ksn[64], vsn[64];
+  __shared__ float Pl[DEC_CHUNK_MAX];
   __shared__ float red[4 * 64];
   __shared__ float stat[8];
 
-  const int head = blockIdx.x, bi = blockIdx.y;
+  const int head = blockIdx.x, bi = blockIdx.y, z = blockIdx.z;
   const int tid = threadIdx.x;
   const int wave = tid >> 6, lane = tid & 63;
   const long off = *offset;
 
-  // ---- rope + cache write for this token (threads 0..63, d = tid)
+  // ---- rope for this token (threads 0..63, d = tid); every block computes
+  // it (cheap) so no block ever reads the fresh cache row another block
+  // wrote — block z==0 alone persists it to the caches.
   if (tid < 64) {
     const int d = tid;
     const long base = (long)bi * 3 * h * 64 + (long)head * 64 + d;
@@ -1205,32 +1215,49 @@ void fa_decode_kernel(
       kv = kv * cs + sgn * bf2f(qkv[pbase + (long)h * 64]) * sn;
       vv = vv * cs + sgn * bf2f(qkv[pbase + 2L * h * 64]) * sn;
     }
-    const long cbase = (((long)bi * h + head) * N + off) * 64 + d;
-    kc[cbase] = f2bf(kv);
-    vc[cbase] = f2bf(vv);
+    if (z == 0) {
+      const long cbase = (((long)bi * h + head) * N + off) * 64 + d;
+      kc[cbase] = f2bf(kv);
+      vc[cbase] = f2bf(vv);
+    }
     qs[d] = qv * scale;
+    ksn[d] = kv;
+    vsn[d] = vv;
   }
   __syncthreads();
 
-  // ---- dots: one wave per key (4 keys in flight), lane = d
+  // ---- dots over this block's key chunk: one LANE per key, K row read as
+  // 8x int4v (the full 128-byte row is consumed across the 8 chunks)
+  const long s0 = (long)z * chunk;
   const bool* prow = pattern ? pattern + off * N : nullptr;
   const short* krow0 = kc + ((long)bi * h + head) * N * 64;
-  for (long key = wave; key <= off; key += 4) {
-    const bool allowed = prow == nullptr || prow[key];
+  for (int base = wave * 64; base < chunk; base += 256) {
+    const long key = s0 + base + lane;
     float dot = NEG_INF;
-    if (allowed) {
-      float p = qs[lane] * bf2f(krow0[key * 64 + lane]);
-      #pragma unroll
-      for (int s = 32; s > 0; s >>= 1) p += __shfl_xor(p, s);
+    if (key <= off && (prow == nullptr || prow[key])) {
+      float p = 0.f;
+      if (key == off) {
+        #pragma unroll
+        for (int d = 0; d < 64; ++d) p += qs[d] * ksn[d];
+      } else {
+        const short* krow = krow0 + key * 64;
+        #pragma unroll
+        for (int c = 0; c < 8; ++c) {
+          int4v kk = *reinterpret_cast<const int4v*>(krow + c * 8);
+          const short* ks = reinterpret_cast<const short*>(&kk);
+          #pragma unroll
+          for (int e = 0; e < 8; ++e) p += qs[c * 8 + e] * bf2f(ks[e]);
+        }
+      }
       dot = p;
     }
-    if (lane == 0) P[key] = dot;
+    Pl[base + lane] = dot;
   }
   __syncthreads();
 
-  // ---- block softmax over P[0..off]
+  // ---- chunk-local softmax stats
   float m = NEG_INF;
-  for (long i = tid; i <= off; i += 256) m = fmaxf(m, P[i]);
+  for (int i = tid; i < chunk; i += 256) m = fmaxf(m, Pl[i]);
   #pragma unroll
   for (int s = 32; s > 0; s >>= 1) m = fmaxf(m, __shfl_xor(m, s));
   if (lane == 0) stat[wave] = m;
@@ -1238,32 +1265,62 @@ void fa_decode_kernel(
   m = fmaxf(fmaxf(stat[0], stat[1]), fmaxf(stat[2], stat[3]));
 
   float lsum = 0.f;
-  for (long i = tid; i <= off; i += 256) {
-    const float p = (P[i] == NEG_INF) ? 0.f : __expf(P[i] - m);
-    P[i] = p;
+  for (int i = tid; i < chunk; i += 256) {
+    const float p = (Pl[i] == NEG_INF || m == NEG_INF)
+        ? 0.f : __expf(Pl[i] - m);
+    Pl[i] = p;
     lsum += p;
   }
   #pragma unroll
   for (int s = 32; s > 0; s >>= 1) lsum += __shfl_xor(lsum, s);
   if (lane == 0) stat[4 + wave] = lsum;
   __syncthreads();
-  const float denom = stat[4] + stat[5] + stat[6] + stat[7];
+  const float l_loc = stat[4] + stat[5] + stat[6] + stat[7];
 
-  // ---- P*V: lane = d, wave-strided keys, coalesced V rows
+  // ---- P*V over the chunk: lane = d, wave-strided keys, coalesced V rows
   const short* vrow0 = vc + ((long)bi * h + head) * N * 64;
   float acc = 0.f;
-  for (long key = wave; key <= off; key += 4) {
-    const float p = P[key];
-    if (p != 0.f) acc += p * bf2f(vrow0[key * 64 + lane]);
+  for (int i = wave; i < chunk; i += 4) {
+    const float p = Pl[i];
+    if (p != 0.f) {
+      const long key = s0 + i;
+      const float vv = (key == off) ? vsn[lane]
+                                    : bf2f(vrow0[key * 64 + lane]);
+      acc += p * vv;
+    }
   }
   red[wave * 64 + lane] = acc;
   __syncthreads();
   if (wave == 0) {
-    const float total = red[lane] + red[64 + lane] + red[128 + lane] +
+    float* sl = scratch + (((long)bi * h + head) * KS + z) * 66;
+    sl[2 + lane] = red[lane] + red[64 + lane] + red[128 + lane] +
         red[192 + lane];
-    out[(long)bi * h * 64 + (long)head * 64 + lane] =
-        f2bf(denom > 0.f ? total / denom : 0.f);
+    if (lane == 0) { sl[0] = m; sl[1] = l_loc; }
   }
+}
+
+__global__ __launch_bounds__(64)
+void fa_decode_combine_kernel(
+    const float* __restrict__ scratch,  // [b, h, KS, 66]
+    short* __restrict__ out,            // [b, h*64]
+    int b, int h, int KS) {
+  const int head = blockIdx.x, bi = blockIdx.y;
+  const int lane = threadIdx.x;
+  const float* s0 = scratch + ((long)bi * h + head) * KS * 66;
+  float m = NEG_INF;
+  for (int z = 0; z < KS; ++z) m = fmaxf(m, s0[z * 66]);
+  float den = 0.f, acc = 0.f;
+  if (m != NEG_INF) {
+    for (int z = 0; z < KS; ++z) {
+      const float mz = s0[z * 66];
+      if (mz == NEG_INF) continue;
+      const float r = __expf(mz - m);
+      den += s0[z * 66 + 1] * r;
+      acc += s0[z * 66 + 2 + lane] * r;
+    }
+  }
+  out[(long)bi * h * 64 + (long)head * 64 + lane] =
+      f2bf(den > 0.f ? acc / den : 0.f);
 }
 
 // ---------------------------------------------------------------------------
@@ -1712,14 +1769,24 @@ torch::Tensor fa_decode(torch::Tensor qkv, torch::Tensor kc, torch::Tensor vc,
     pat = pattern->data_ptr<bool>();
   }
   auto out = torch::empty({b, (long)h * 64}, qkv.options());
-  dim3 grid(h, b);
-  hipLaunchKernelGGL(fa_decode_kernel, grid, dim3(256), 0, cur_stream(),
+  // split so each block owns ~192 keys: b*h*KS blocks fill the 256 CUs
+  int KS = std::min(8, std::max(2, (N + 191) / 192));
+  int chunk = ((N + KS - 1) / KS + 63) & ~63;
+  CHK(chunk <= DEC_CHUNK_MAX);
+  auto scratch = torch::empty({(long)b * h * KS * 66},
+                              qkv.options().dtype(torch::kFloat32));
+  dim3 grid(h, b, KS);
+  hipLaunchKernelGGL(fa_decode_part_kernel, grid, dim3(256), 0, cur_stream(),
                      reinterpret_cast<const short*>(qkv.data_ptr()),
                      reinterpret_cast<short*>(kc.data_ptr()),
                      reinterpret_cast<short*>(vc.data_ptr()),
                      cp, sp, offset.data_ptr<long>(), pat,
+                     scratch.data_ptr<float>(),
+                     b, h, N, rot, (float)scale, KS, chunk);
+  hipLaunchKernelGGL(fa_decode_combine_kernel, dim3(h, b), dim3(64), 0,
+                     cur_stream(), scratch.data_ptr<float>(),
                      reinterpret_cast<short*>(out.data_ptr()),
-                     b, h, N, rot, (float)scale);
+                     b, h, KS);
   return out;
 }
 
