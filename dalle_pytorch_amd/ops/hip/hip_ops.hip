@@ -1231,29 +1231,27 @@ void fa_decode_part_kernel(
   const long s0 = (long)z * chunk;
   const bool* prow = pattern ? pattern + off * N : nullptr;
   const short* krow0 = kc + ((long)bi * h + head) * N * 64;
-  // branch-free: load unconditionally from a clamped in-bounds address and
-  // select afterwards, so the 8 row chunks pipeline instead of serializing
-  // behind a mask test per key (the caches are zero-filled at construction,
-  // dead lanes multiply into a finite discarded value)
   for (int base = wave * 64; base < chunk; base += 256) {
     const long key = s0 + base + lane;
-    const long kcl = key < (long)N ? key : (long)N - 1;
-    const bool ok = key <= off && (prow == nullptr || prow[kcl]);
-    float p = 0.f;
-    if (key == off) {
-      #pragma unroll
-      for (int d = 0; d < 64; ++d) p += qs[d] * ksn[d];
-    } else {
-      const short* krow = krow0 + kcl * 64;
-      #pragma unroll
-      for (int c = 0; c < 8; ++c) {
-        int4v kk = *reinterpret_cast<const int4v*>(krow + c * 8);
-        const short* ks = reinterpret_cast<const short*>(&kk);
+    float dot = NEG_INF;
+    if (key <= off && (prow == nullptr || prow[key])) {
+      float p = 0.f;
+      if (key == off) {
         #pragma unroll
-        for (int e = 0; e < 8; ++e) p += qs[c * 8 + e] * bf2f(ks[e]);
+        for (int d = 0; d < 64; ++d) p += qs[d] * ksn[d];
+      } else {
+        const short* krow = krow0 + key * 64;
+        #pragma unroll
+        for (int c = 0; c < 8; ++c) {
+          int4v kk = *reinterpret_cast<const int4v*>(krow + c * 8);
+          const short* ks = reinterpret_cast<const short*>(&kk);
+          #pragma unroll
+          for (int e = 0; e < 8; ++e) p += qs[c * 8 + e] * bf2f(ks[e]);
+        }
       }
+      dot = p;
     }
-    Pl[base + lane] = ok ? p : NEG_INF;
+    Pl[base + lane] = dot;
   }
   __syncthreads();
 
@@ -1284,13 +1282,12 @@ void fa_decode_part_kernel(
   float acc = 0.f;
   for (int i = wave; i < chunk; i += 4) {
     const float p = Pl[i];
-    const long key = s0 + i;
-    const long kcl = key < (long)N ? key : (long)N - 1;
-    // branch-free as above: p is 0 for dead keys, the fresh row comes from
-    // LDS since other blocks never see block 0's cache write this dispatch
-    const float vv = (key == off) ? vsn[lane]
-                                  : bf2f(vrow0[kcl * 64 + lane]);
-    acc += p * vv;
+    if (p != 0.f) {
+      const long key = s0 + i;
+      const float vv = (key == off) ? vsn[lane]
+                                    : bf2f(vrow0[key * 64 + lane]);
+      acc += p * vv;
+    }
   }
   red[wave * 64 + lane] = acc;
   __syncthreads();
