@@ -1299,6 +1299,137 @@ void fa_decode_part_kernel(
   }
 }
 
+// Live-list variant: the decoder precomputes, per attention layer and per
+// offset row, the indices of the keys the pattern+causality actually allow
+// (~288 of 1281 under the flagship axial patterns). Blocks then iterate
+// listed keys only — every K/V row loaded is a live one — instead of
+// scanning the whole slot range and discarding ~75% of it.
+__global__ __launch_bounds__(256)
+void fa_decode_part_list_kernel(
+    const short* __restrict__ qkv,    // [b, 3*h*64]
+    short* __restrict__ kc,           // [b, h, N, 64]
+    short* __restrict__ vc,
+    const float* __restrict__ cosv,
+    const float* __restrict__ sinv,
+    const long* __restrict__ offset,  // [1]
+    const int* __restrict__ live,     // [N, Lmax] key indices per offset row
+    const int* __restrict__ live_cnt, // [N]
+    float* __restrict__ scratch,      // [b, h, KS, 66]
+    int b, int h, int N, int rot, float scale, int KS, int chunk, int Lmax) {
+
+  __shared__ float qs[64], ksn[64], vsn[64];
+  __shared__ float Pl[DEC_CHUNK_MAX];
+  __shared__ int   Ki[DEC_CHUNK_MAX];
+  __shared__ float red[4 * 64];
+  __shared__ float stat[8];
+
+  const int head = blockIdx.x, bi = blockIdx.y, z = blockIdx.z;
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6, lane = tid & 63;
+  const long off = *offset;
+
+  if (tid < 64) {
+    const int d = tid;
+    const long base = (long)bi * 3 * h * 64 + (long)head * 64 + d;
+    float qv = bf2f(qkv[base]);
+    float kv = bf2f(qkv[base + (long)h * 64]);
+    float vv = bf2f(qkv[base + 2L * h * 64]);
+    if (d < rot && cosv != nullptr) {
+      const int prt = d ^ 1;
+      const float cs = cosv[off * rot + d];
+      const float sn = sinv[off * rot + d];
+      const float sgn = (d & 1) ? 1.f : -1.f;
+      const long pbase = (long)bi * 3 * h * 64 + (long)head * 64 + prt;
+      qv = qv * cs + sgn * bf2f(qkv[pbase]) * sn;
+      kv = kv * cs + sgn * bf2f(qkv[pbase + (long)h * 64]) * sn;
+      vv = vv * cs + sgn * bf2f(qkv[pbase + 2L * h * 64]) * sn;
+    }
+    if (z == 0) {
+      const long cbase = (((long)bi * h + head) * N + off) * 64 + d;
+      kc[cbase] = f2bf(kv);
+      vc[cbase] = f2bf(vv);
+    }
+    qs[d] = qv * scale;
+    ksn[d] = kv;
+    vsn[d] = vv;
+  }
+  __syncthreads();
+
+  const int cnt = live_cnt[off];
+  const int j0 = z * chunk;
+  const int jn = min(chunk, cnt - j0);          // live entries in this block
+  const int* lrow = live + off * (long)Lmax;
+  const short* krow0 = kc + ((long)bi * h + head) * N * 64;
+
+  // dots: one lane per listed key; the index load doubles as the mask
+  for (int base = wave * 64; base < chunk; base += 256) {
+    const int j = base + lane;
+    float dot = NEG_INF;
+    int key = -1;
+    if (j < jn) key = lrow[j0 + j];
+    if (key >= 0) {
+      float p = 0.f;
+      if (key == (int)off) {
+        #pragma unroll
+        for (int d = 0; d < 64; ++d) p += qs[d] * ksn[d];
+      } else {
+        const short* krow = krow0 + (long)key * 64;
+        #pragma unroll
+        for (int c = 0; c < 8; ++c) {
+          int4v kk = *reinterpret_cast<const int4v*>(krow + c * 8);
+          const short* ks = reinterpret_cast<const short*>(&kk);
+          #pragma unroll
+          for (int e = 0; e < 8; ++e) p += qs[c * 8 + e] * bf2f(ks[e]);
+        }
+      }
+      dot = p;
+    }
+    if (base + lane < chunk) { Pl[j] = dot; Ki[j] = key; }
+  }
+  __syncthreads();
+
+  float m = NEG_INF;
+  for (int i = tid; i < chunk; i += 256) m = fmaxf(m, Pl[i]);
+  #pragma unroll
+  for (int s = 32; s > 0; s >>= 1) m = fmaxf(m, __shfl_xor(m, s));
+  if (lane == 0) stat[wave] = m;
+  __syncthreads();
+  m = fmaxf(fmaxf(stat[0], stat[1]), fmaxf(stat[2], stat[3]));
+
+  float lsum = 0.f;
+  for (int i = tid; i < chunk; i += 256) {
+    const float p = (Pl[i] == NEG_INF || m == NEG_INF)
+        ? 0.f : __expf(Pl[i] - m);
+    Pl[i] = p;
+    lsum += p;
+  }
+  #pragma unroll
+  for (int s = 32; s > 0; s >>= 1) lsum += __shfl_xor(lsum, s);
+  if (lane == 0) stat[4 + wave] = lsum;
+  __syncthreads();
+  const float l_loc = stat[4] + stat[5] + stat[6] + stat[7];
+
+  const short* vrow0 = vc + ((long)bi * h + head) * N * 64;
+  float acc = 0.f;
+  for (int i = wave; i < jn; i += 4) {
+    const float p = Pl[i];
+    if (p != 0.f) {
+      const int key = Ki[i];
+      const float vv = (key == (int)off) ? vsn[lane]
+                                         : bf2f(vrow0[(long)key * 64 + lane]);
+      acc += p * vv;
+    }
+  }
+  red[wave * 64 + lane] = acc;
+  __syncthreads();
+  if (wave == 0) {
+    float* sl = scratch + (((long)bi * h + head) * KS + z) * 66;
+    sl[2 + lane] = red[lane] + red[64 + lane] + red[128 + lane] +
+        red[192 + lane];
+    if (lane == 0) { sl[0] = m; sl[1] = l_loc; }
+  }
+}
+
 __global__ __launch_bounds__(64)
 void fa_decode_combine_kernel(
     const float* __restrict__ scratch,  // [b, h, KS, 66]
@@ -1748,7 +1879,9 @@ torch::Tensor fa_decode(torch::Tensor qkv, torch::Tensor kc, torch::Tensor vc,
                         std::optional<torch::Tensor> sinv,
                         torch::Tensor offset,
                         std::optional<torch::Tensor> pattern,
-                        double scale) {
+                        double scale,
+                        std::optional<torch::Tensor> live,
+                        std::optional<torch::Tensor> live_cnt) {
   CHK(qkv.is_cuda() && qkv.dtype() == torch::kBFloat16 && qkv.is_contiguous());
   CHK(kc.is_contiguous() && vc.is_contiguous());
   const int b = kc.size(0), h = kc.size(1), N = kc.size(2);
@@ -1769,13 +1902,27 @@ torch::Tensor fa_decode(torch::Tensor qkv, torch::Tensor kc, torch::Tensor vc,
     pat = pattern->data_ptr<bool>();
   }
   auto out = torch::empty({b, (long)h * 64}, qkv.options());
-  // split so each block owns ~192 keys: b*h*KS blocks fill the 256 CUs
-  int KS = std::min(8, std::max(2, (N + 191) / 192));
-  int chunk = ((N + KS - 1) / KS + 63) & ~63;
+  // split so each block owns ~192 (live) keys: b*h*KS blocks fill the chip
+  const int span = live.has_value() ? (int)live->size(1) : N;
+  int KS = std::min(8, std::max(live.has_value() ? 1 : 2, (span + 191) / 192));
+  int chunk = ((span + KS - 1) / KS + 63) & ~63;
   CHK(chunk <= DEC_CHUNK_MAX);
   auto scratch = torch::empty({(long)b * h * KS * 66},
                               qkv.options().dtype(torch::kFloat32));
   dim3 grid(h, b, KS);
+  if (live.has_value()) {
+    CHK(live->dtype() == torch::kInt32 && live->is_contiguous());
+    CHK(live_cnt.has_value() && live_cnt->dtype() == torch::kInt32);
+    hipLaunchKernelGGL(fa_decode_part_list_kernel, grid, dim3(256), 0,
+                       cur_stream(),
+                       reinterpret_cast<const short*>(qkv.data_ptr()),
+                       reinterpret_cast<short*>(kc.data_ptr()),
+                       reinterpret_cast<short*>(vc.data_ptr()),
+                       cp, sp, offset.data_ptr<long>(),
+                       live->data_ptr<int>(), live_cnt->data_ptr<int>(),
+                       scratch.data_ptr<float>(),
+                       b, h, N, rot, (float)scale, KS, chunk, span);
+  } else {
   hipLaunchKernelGGL(fa_decode_part_kernel, grid, dim3(256), 0, cur_stream(),
                      reinterpret_cast<const short*>(qkv.data_ptr()),
                      reinterpret_cast<short*>(kc.data_ptr()),
@@ -1783,6 +1930,7 @@ torch::Tensor fa_decode(torch::Tensor qkv, torch::Tensor kc, torch::Tensor vc,
                      cp, sp, offset.data_ptr<long>(), pat,
                      scratch.data_ptr<float>(),
                      b, h, N, rot, (float)scale, KS, chunk);
+  }
   hipLaunchKernelGGL(fa_decode_combine_kernel, dim3(h, b), dim3(64), 0,
                      cur_stream(), scratch.data_ptr<float>(),
                      reinterpret_cast<short*>(out.data_ptr()),
