@@ -1301,10 +1301,15 @@ void fa_decode_part_kernel(
 
 // Live-list variant: the decoder precomputes, per attention layer and per
 // offset row, the indices of the keys the pattern+causality actually allow
-// (~288 of 1281 under the flagship axial patterns). Blocks then iterate
-// listed keys only — every K/V row loaded is a live one — instead of
-// scanning the whole slot range and discarding ~75% of it.
-__global__ __launch_bounds__(256)
+// (~288 of 1281 under the flagship axial patterns). Blocks iterate listed
+// keys only — every K/V row loaded is a live one.
+//
+// ONE wave per block, 64 listed keys per block: at ~288 live keys the
+// 4-wave form spent its time in barriers and cross-wave staging for tiny
+// per-phase work (59 us/dispatch measured vs ~5 us of live traffic).
+// Wave-synchronous execution needs no barriers — softmax stats reduce by
+// shuffle, the only LDS is the q/k/v broadcast and the 64-entry p row.
+__global__ __launch_bounds__(64)
 void fa_decode_part_list_kernel(
     const short* __restrict__ qkv,    // [b, 3*h*64]
     short* __restrict__ kc,           // [b, h, N, 64]
@@ -1318,18 +1323,15 @@ void fa_decode_part_list_kernel(
     int b, int h, int N, int rot, float scale, int KS, int chunk, int Lmax) {
 
   __shared__ float qs[64], ksn[64], vsn[64];
-  __shared__ float Pl[DEC_CHUNK_MAX];
-  __shared__ int   Ki[DEC_CHUNK_MAX];
-  __shared__ float red[4 * 64];
-  __shared__ float stat[8];
+  __shared__ float Pl[64];
+  __shared__ int   Ki[64];
 
   const int head = blockIdx.x, bi = blockIdx.y, z = blockIdx.z;
-  const int tid = threadIdx.x;
-  const int wave = tid >> 6, lane = tid & 63;
+  const int lane = threadIdx.x;
   const long off = *offset;
 
-  if (tid < 64) {
-    const int d = tid;
+  {
+    const int d = lane;
     const long base = (long)bi * 3 * h * 64 + (long)head * 64 + d;
     float qv = bf2f(qkv[base]);
     float kv = bf2f(qkv[base + (long)h * 64]);
@@ -1338,7 +1340,7 @@ void fa_decode_part_list_kernel(
       const int prt = d ^ 1;
       const float cs = cosv[off * rot + d];
       const float sn = sinv[off * rot + d];
-      const float sgn = (d & 1) ? 1.f : -1.f;
+      const float sgn = (d & 1) ? 1.f : -1.f;  // even: -x_{d+1}, odd: +x_{d-1}
       const long pbase = (long)bi * 3 * h * 64 + (long)head * 64 + prt;
       qv = qv * cs + sgn * bf2f(qkv[pbase]) * sn;
       kv = kv * cs + sgn * bf2f(qkv[pbase + (long)h * 64]) * sn;
@@ -1353,81 +1355,64 @@ void fa_decode_part_list_kernel(
     ksn[d] = kv;
     vsn[d] = vv;
   }
-  __syncthreads();
+  __syncthreads();   // single wave: compiles to a waitcnt, no barrier cost
 
   const int cnt = live_cnt[off];
-  const int j0 = z * chunk;
-  const int jn = min(chunk, cnt - j0);          // live entries in this block
+  const int j0 = z * 64;
+  const int jn = min(64, cnt - j0);             // live entries in this block
   const int* lrow = live + off * (long)Lmax;
   const short* krow0 = kc + ((long)bi * h + head) * N * 64;
 
   // dots: one lane per listed key; the index load doubles as the mask
-  for (int base = wave * 64; base < chunk; base += 256) {
-    const int j = base + lane;
-    float dot = NEG_INF;
-    int key = -1;
-    if (j < jn) key = lrow[j0 + j];
-    if (key >= 0) {
-      float p = 0.f;
-      if (key == (int)off) {
+  float dot = NEG_INF;
+  int key = -1;
+  if (lane < jn) key = lrow[j0 + lane];
+  if (key >= 0) {
+    float p = 0.f;
+    if (key == (int)off) {
+      #pragma unroll
+      for (int d = 0; d < 64; ++d) p += qs[d] * ksn[d];
+    } else {
+      const short* krow = krow0 + (long)key * 64;
+      #pragma unroll
+      for (int c = 0; c < 8; ++c) {
+        int4v kk = *reinterpret_cast<const int4v*>(krow + c * 8);
+        const short* ks = reinterpret_cast<const short*>(&kk);
         #pragma unroll
-        for (int d = 0; d < 64; ++d) p += qs[d] * ksn[d];
-      } else {
-        const short* krow = krow0 + (long)key * 64;
-        #pragma unroll
-        for (int c = 0; c < 8; ++c) {
-          int4v kk = *reinterpret_cast<const int4v*>(krow + c * 8);
-          const short* ks = reinterpret_cast<const short*>(&kk);
-          #pragma unroll
-          for (int e = 0; e < 8; ++e) p += qs[c * 8 + e] * bf2f(ks[e]);
-        }
+        for (int e = 0; e < 8; ++e) p += qs[c * 8 + e] * bf2f(ks[e]);
       }
-      dot = p;
     }
-    if (base + lane < chunk) { Pl[j] = dot; Ki[j] = key; }
+    dot = p;
   }
-  __syncthreads();
 
-  float m = NEG_INF;
-  for (int i = tid; i < chunk; i += 256) m = fmaxf(m, Pl[i]);
+  // wave-local softmax stats by shuffle
+  float m = dot;
   #pragma unroll
   for (int s = 32; s > 0; s >>= 1) m = fmaxf(m, __shfl_xor(m, s));
-  if (lane == 0) stat[wave] = m;
-  __syncthreads();
-  m = fmaxf(fmaxf(stat[0], stat[1]), fmaxf(stat[2], stat[3]));
-
-  float lsum = 0.f;
-  for (int i = tid; i < chunk; i += 256) {
-    const float p = (Pl[i] == NEG_INF || m == NEG_INF)
-        ? 0.f : __expf(Pl[i] - m);
-    Pl[i] = p;
-    lsum += p;
-  }
+  const float p = (dot == NEG_INF || m == NEG_INF) ? 0.f : __expf(dot - m);
+  float lsum = p;
   #pragma unroll
   for (int s = 32; s > 0; s >>= 1) lsum += __shfl_xor(lsum, s);
-  if (lane == 0) stat[4 + wave] = lsum;
-  __syncthreads();
-  const float l_loc = stat[4] + stat[5] + stat[6] + stat[7];
 
+  Pl[lane] = p;
+  Ki[lane] = key;
+  __syncthreads();
+
+  // P*V: lane = d, serial over this block's listed keys
   const short* vrow0 = vc + ((long)bi * h + head) * N * 64;
   float acc = 0.f;
-  for (int i = wave; i < jn; i += 4) {
-    const float p = Pl[i];
-    if (p != 0.f) {
-      const int key = Ki[i];
-      const float vv = (key == (int)off) ? vsn[lane]
-                                         : bf2f(vrow0[(long)key * 64 + lane]);
-      acc += p * vv;
+  for (int i = 0; i < jn; ++i) {
+    const float pi = Pl[i];
+    if (pi != 0.f) {
+      const int ki = Ki[i];
+      const float vv = (ki == (int)off) ? vsn[lane]
+                                        : bf2f(vrow0[(long)ki * 64 + lane]);
+      acc += pi * vv;
     }
   }
-  red[wave * 64 + lane] = acc;
-  __syncthreads();
-  if (wave == 0) {
-    float* sl = scratch + (((long)bi * h + head) * KS + z) * 66;
-    sl[2 + lane] = red[lane] + red[64 + lane] + red[128 + lane] +
-        red[192 + lane];
-    if (lane == 0) { sl[0] = m; sl[1] = l_loc; }
-  }
+  float* sl = scratch + (((long)bi * h + head) * KS + z) * 66;
+  sl[2 + lane] = acc;
+  if (lane == 0) { sl[0] = m; sl[1] = lsum; }
 }
 
 __global__ __launch_bounds__(64)
@@ -1902,18 +1887,25 @@ torch::Tensor fa_decode(torch::Tensor qkv, torch::Tensor kc, torch::Tensor vc,
     pat = pattern->data_ptr<bool>();
   }
   auto out = torch::empty({b, (long)h * 64}, qkv.options());
-  // split so each block owns ~192 (live) keys: b*h*KS blocks fill the chip
   const int span = live.has_value() ? (int)live->size(1) : N;
-  int KS = std::min(8, std::max(live.has_value() ? 1 : 2, (span + 191) / 192));
-  int chunk = ((span + KS - 1) / KS + 63) & ~63;
-  CHK(chunk <= DEC_CHUNK_MAX);
+  // live lists: one 64-thread wave per 64 listed keys (no barriers);
+  // scan path: 4-wave blocks over ~192-slot chunks
+  int KS, chunk;
+  if (live.has_value()) {
+    KS = (span + 63) / 64;
+    chunk = 64;
+  } else {
+    KS = std::min(8, std::max(2, (span + 191) / 192));
+    chunk = ((span + KS - 1) / KS + 63) & ~63;
+    CHK(chunk <= DEC_CHUNK_MAX);
+  }
   auto scratch = torch::empty({(long)b * h * KS * 66},
                               qkv.options().dtype(torch::kFloat32));
   dim3 grid(h, b, KS);
   if (live.has_value()) {
     CHK(live->dtype() == torch::kInt32 && live->is_contiguous());
     CHK(live_cnt.has_value() && live_cnt->dtype() == torch::kInt32);
-    hipLaunchKernelGGL(fa_decode_part_list_kernel, grid, dim3(256), 0,
+    hipLaunchKernelGGL(fa_decode_part_list_kernel, grid, dim3(64), 0,
                        cur_stream(),
                        reinterpret_cast<const short*>(qkv.data_ptr()),
                        reinterpret_cast<short*>(kc.data_ptr()),
