@@ -93,11 +93,39 @@ def _check_average_scalar(rank, world):
     assert torch.allclose(v, torch.tensor((world - 1) / 2))
 
 
+def _check_distributed_clip(rank, world):
+    """Clip after all-reduce == torch clip on the averaged-grad oracle."""
+    model = _model(seed=0)
+    engine = DataParallelEngine(model, bucket_bytes=1 << 10)
+    torch.manual_seed(300 + rank)
+    x = torch.randn(4, 8)
+    (model(x) ** 2).sum().backward()
+    engine.finish_gradient_sync()
+    total = engine.clip_grad_norm_(0.02)
+
+    ref = _model(seed=0)
+    acc = [torch.zeros_like(p) for p in ref.parameters()]
+    for r in range(world):
+        m = _model(seed=0)
+        torch.manual_seed(300 + r)
+        xr = torch.randn(4, 8)
+        (m(xr) ** 2).sum().backward()
+        for a, p in zip(acc, m.parameters()):
+            a += p.grad / world
+    for a, p in zip(acc, ref.parameters()):
+        p.grad = a
+    ref_total = torch.nn.utils.clip_grad_norm_(ref.parameters(), 0.02)
+    assert torch.allclose(total, ref_total, rtol=1e-5)
+    for p, q in zip(model.parameters(), ref.parameters()):
+        assert torch.allclose(p.grad, q.grad, rtol=1e-5, atol=1e-8)
+
+
 @pytest.mark.parametrize('fn,port', [
     ('_check_broadcast', 29611),
     ('_check_grad_allreduce', 29612),
     ('_check_no_sync_accumulation', 29613),
     ('_check_average_scalar', 29614),
+    ('_check_distributed_clip', 29615),
 ])
 def test_distributed_gloo(fn, port):
     _spawn(fn, world=2, port=port)
