@@ -1,11 +1,13 @@
 // dalle_pytorch_amd gfx950 (CDNA4) kernel library.
 //
 // First-party HIP kernels for the hot ops of the DALL-E stack
-// (SURVEY.md §2.5 K2-K7): flash-style causal attention forward with online
-// softmax (MFMA bf16 16x16x32, LDS-tiled K/V with transposed-V store and
-// padded rows for bank-conflict-free ds_read_b128), and fused GEGLU.
-// Written for wave64 / 8-XCD MI355X per the CDNA4 HIP guide —
-// NOT a port of any CUDA kernel.
+// (decode): flash attention forward + backward
+// (dQ, dKdV, Dv) with online softmax and tile-skipping sparsity (MFMA bf16
+// 16x16x32, LDS-tiled K/V with transposed-V store and swizzled rows for
+// bank-conflict-free ds_read_b128), fused rope+QKV split, GEGLU,
+// token-shift, LayerNorm, and the single-token decode family
+// (key-split attention + ring-buffer shift). Written for wave64 / 8-XCD
+// MI355X per the CDNA4 HIP guide — NOT a port of any CUDA kernel.
 //
 // bf16 values are carried as raw `short` bit patterns end to end; float
 // math goes through explicit bit casts (bf2f/f2bf) so no accidental
