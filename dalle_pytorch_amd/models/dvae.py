@@ -17,6 +17,7 @@ Known reference quirks kept on purpose:
   inside forward (reference :185-193,225).
 """
 
+import os
 from math import log2, sqrt
 
 import torch
@@ -211,7 +212,12 @@ class DiscreteVAE(nn.Module):
             f'input must have the correct image size {image_size}'
 
         img = self.norm(img)
-        logits = self.encoder(img)
+        # DALLE_AMD_CONV_GEMM=1 routes the TRAINABLE conv stacks through the
+        # differentiable unfold/fold+GEMM walker too — opt-in MIOpen
+        # independence for train_vae.py on boxes where MIOpen falls back to
+        # naive_conv (the frozen encode/decode paths always use it)
+        gemm = img.is_cuda and os.environ.get('DALLE_AMD_CONV_GEMM') == '1'
+        logits = _run_as_gemms(self.encoder, img) if gemm else self.encoder(img)
         if return_logits:
             return logits
 
@@ -229,7 +235,7 @@ class DiscreteVAE(nn.Module):
             one_hot = p2 - p2.detach() + one_hot
 
         sampled = torch.einsum('bnhw,nd->bdhw', one_hot, self.codebook.weight)
-        out = self.decoder(sampled)
+        out = _run_as_gemms(self.decoder, sampled) if gemm else self.decoder(sampled)
 
         if not return_loss:
             return out

@@ -7,6 +7,7 @@
 
 import argparse
 import math
+import os
 from pathlib import Path
 
 import torch
@@ -45,6 +46,9 @@ def parse_args(argv=None):
     p.add_argument('--transparent', action='store_true')
     p.add_argument('--straight_through', action='store_true')
     p.add_argument('--reinmax', action='store_true')
+    p.add_argument('--conv_gemm', action='store_true',
+                   help='run the trainable conv stacks as unfold/fold+GEMMs '
+                        '(MIOpen independence on untuned boxes)')
     p.add_argument('--fp16', action='store_true', help='bf16 autocast on MI355X')
     p.add_argument('--stop_after_steps', type=int, default=None)
     p.add_argument('--output_dir', default='.')
@@ -66,6 +70,8 @@ class _SyntheticImages(torch.utils.data.Dataset):
 
 def main(argv=None):
     args = parse_args(argv)
+    if args.conv_gemm:
+        os.environ['DALLE_AMD_CONV_GEMM'] = '1'
     from dalle_pytorch_amd.utils.tunable import maybe_enable_tunableop
     maybe_enable_tunableop()
     rank, world, local_rank = init_distributed()
