@@ -47,8 +47,9 @@ def _flash_bwd_composite(q, k, v, out, lse, dout, scale, causal, key_mask,
     """Flash-style backward at the torch level: recompute P per q-chunk from
     the saved logsumexp, never materializing the full n x n matrix. All
     GEMMs run in bf16 on rocBLAS MFMA paths (fp32 accumulation inside);
-    the P/dS elementwise math is fp32. A fully hand-written CDNA4 bwd
-    kernel replaces this incrementally."""
+    the P/dS elementwise math is fp32. The hand-written fa_bwd kernels are
+    the production path; this stays as the oracle-adjacent fallback for
+    extensions built without them."""
     b, h, nq, d = q.shape
     nk = k.shape[2]
     diag = nk - nq
@@ -134,8 +135,9 @@ def _hip_supported(q, k, causal, key_mask):
     if q.dtype != torch.bfloat16:
         return False
     if q.shape[-2] < 32:
-        # single-token decode: rocBLAS batched GEMV (memory-bound) until the
-        # dedicated decode kernel lands; not a training-path fallback
+        # the model's dict-cache decode path: rocBLAS batched GEMV is fine
+        # there (the production decode is FastDecoder's fa_decode kernels);
+        # not a training-path fallback
         return False
     return True
 
