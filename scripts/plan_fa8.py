@@ -19,6 +19,20 @@ candidate K-tile swizzles.
 
 If the A/B-fragment guesses turn out wrong on hardware, flip them here
 first: a one-line change re-runs the whole derivation.
+
+Planner conclusions (enumerated here, to implement in round 2):
+* pipeline: the swapped-QK^T -> half-split softmax -> single
+  permlane32_swap -> PV chain is index-exact at D=64 (fp64 error 4e-15);
+* K tile: [64 keys][64 d] bf16 at 128 B stride with byte ^= (row&3)<<5
+  reaches the b128 bandwidth floor (8) on both the 8x8-chunk store and
+  the B-fragment read — the guide's D=128 swizzle also floors the read
+  but was derived for 256 B rows;
+* V tile: direct [k][d] layout is structurally 4-way on the tr-read (the
+  two lane-halves read the same column range at rows 8 apart under ANY
+  b128-legal transform) — store V TRANSPOSED instead, as the shipped
+  16x16x32 kernel already does: the PV B-fragment then becomes one
+  contiguous b128 row read and round 1's swz_key store swizzle holds the
+  transpose stores at 2-way.
 """
 
 import numpy as np
@@ -258,6 +272,35 @@ def no_swizzle(row, byte):
 # A 64-lane instruction with 16 B per lane cannot beat 8 LDS cycles:
 # counts above this are conflicts, counts equal to it are the floor.
 B128_FLOOR = 8
+
+
+def v_tile_conflicts(swizzle, kv=64, cols=64, elem=2):
+    """V LDS tile [kv keys][cols d] bf16, consumed by PV's B fragments:
+    lane l needs V[16*ks + 8*(l>>5) + e][d0 + (l&31)] — a COLUMN walk
+    (the ladder's "tr-read"). Modeled per 8-element fragment read as 8
+    b16 accesses (ds_read_u16 / tr-read micro-ops); a swizzle can spread
+    the column across banks. Store is the same 16B-chunk row pattern as
+    the K tile. The b16 floor for 64 lanes x 2 B is 32 banks -> 2."""
+    row_bytes = cols * elem
+    worst_store = 0
+    for base_row in range(0, kv, 8):
+        addrs = []
+        for lane in range(LANES):
+            row = base_row + lane // 8
+            byte = row * row_bytes + (lane % 8) * 16
+            addrs.append(swizzle(row, byte))
+        worst_store = max(worst_store, bank_conflicts_store(addrs))
+    worst_read = 0
+    for ks in range(kv // 16):
+        for d0 in range(0, cols, 32):
+            for e in range(8):
+                addrs = []
+                for lane in range(LANES):
+                    row = 16 * ks + 8 * (lane >> 5) + e
+                    byte = row * row_bytes + (d0 + (lane & 31)) * elem
+                    addrs.append(swizzle(row, byte))
+                worst_read = max(worst_read, bank_conflicts_store(addrs))
+    return worst_store, worst_read
 
 
 if __name__ == '__main__':

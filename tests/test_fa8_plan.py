@@ -38,3 +38,11 @@ def test_d64_swizzle_hits_bandwidth_floor():
     # and the unswizzled layout genuinely conflicts on the read
     _, rd_none = plan.k_tile_conflicts(plan.no_swizzle)
     assert rd_none > plan.B128_FLOOR
+
+
+def test_v_tile_direct_layout_is_structurally_conflicted():
+    """Direct [k][d] V layout cannot floor the tr-read (justifies keeping
+    the transposed-V store in the 8-wave ladder)."""
+    for sw in (plan.no_swizzle, plan.d64_swizzle, plan.guide_swizzle):
+        _, rd = plan.v_tile_conflicts(sw)
+        assert rd >= 4
