@@ -174,3 +174,82 @@ def test_rotary_applied_to_v_quirk():
     out_rot = attn(x, rotary_pos_emb=table)
     out_plain = attn(x)
     assert not torch.allclose(out_rot, out_plain, atol=1e-4)
+
+
+def _dense_attn_with_lse(q, k, v, scale, causal):
+    """Plain attention returning (out_unnormalized_by_partition, lse):
+    out = softmax(qk^T)v and lse = logsumexp of the scores row."""
+    s = torch.matmul(q * scale, k.transpose(-1, -2))
+    if causal:
+        i, j = s.shape[-2:]
+        cm = torch.ones(i, j, dtype=torch.bool).triu_(j - i + 1)
+        s = s.masked_fill(cm, float('-inf'))
+    lse = torch.logsumexp(s, dim=-1)
+    p = torch.exp(s - lse.unsqueeze(-1))
+    return torch.matmul(p, v), lse
+
+
+def test_axial_attention_lse_decomposition():
+    """Axial attention == two DENSE flash-style attentions merged by lse:
+    (a) image queries over the text prefix (non-causal: every text key is
+    allowed), (b) image queries over their own grid row/col, causal.
+    Pins the round-2 decomposition (NOTES_ROUND2.md, ladder option 3) that
+    removes masks and partial tiles entirely."""
+    torch.manual_seed(4)
+    S, t = 4, 3
+    n_img_full = S * S
+    for axis in (0, 1):
+        for n_img in (n_img_full, n_img_full - 1):
+            n = t + n_img
+            b, h, d = 2, 2, 16
+            scale = d ** -0.5
+            q = torch.randn(b, h, n, d)
+            k = torch.randn(b, h, n, d)
+            v = torch.randn(b, h, n, d)
+
+            # oracle: dense attention under the axial mask + causality
+            from dalle_pytorch_amd.models.attention import axial_mask
+            m = axial_mask(n, t, S, axis)
+            m &= torch.ones(n, n, dtype=torch.bool).tril_()
+            s = torch.matmul(q * scale, k.transpose(-1, -2))
+            s = s.masked_fill(~m, -torch.finfo(s.dtype).max)
+            ref = torch.matmul(s.softmax(-1), v)
+
+            # text queries: plain causal over the text prefix
+            out_text, _ = _dense_attn_with_lse(
+                q[:, :, :t], k[:, :, :t], v[:, :, :t], scale, causal=True)
+
+            # image queries, part 1: all text keys (non-causal)
+            qi = q[:, :, t:]
+            o1, l1 = _dense_attn_with_lse(qi, k[:, :, :t], v[:, :, :t],
+                                          scale, causal=False)
+
+            # image queries, part 2: row-(or col-)local causal attention.
+            # pad the grid to S*S (the pad token is a key only for queries
+            # after it — none — and is sliced off as a query)
+            pad = n_img_full - n_img
+            def grid(z):
+                zi = torch.nn.functional.pad(z[:, :, t:], (0, 0, 0, pad))
+                g = zi.reshape(b, h, S, S, d)
+                if axis == 1:
+                    g = g.transpose(2, 3)
+                return g.reshape(b, h * S, S, d)
+            qg, kg, vg = grid(q), grid(k), grid(v)
+            o2g, l2g = _dense_attn_with_lse(qg, kg, vg, scale, causal=True)
+            def ungrid(z, last):
+                g = z.reshape(b, h, S, S, *z.shape[3:])
+                if axis == 1:
+                    g = g.transpose(2, 3)
+                return g.reshape(b, h, n_img_full, *z.shape[3:])[:, :, :last]
+            o2 = ungrid(o2g, n_img)
+            l2 = ungrid(l2g.unsqueeze(-1), n_img).squeeze(-1)
+
+            # lse merge
+            mx = torch.maximum(l1, l2)
+            w1 = torch.exp(l1 - mx).unsqueeze(-1)
+            w2 = torch.exp(l2 - mx).unsqueeze(-1)
+            out_img = (o1 * w1 + o2 * w2) / (w1 + w2)
+
+            out = torch.cat((out_text, out_img), dim=2)
+            assert torch.allclose(out, ref, atol=1e-5), \
+                (axis, n_img, (out - ref).abs().max())
