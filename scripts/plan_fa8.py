@@ -208,6 +208,85 @@ def simulate_attention_tile(D=64, KV=32, seed=0):
     return out, ref
 
 
+def simulate_online_attention(D=64, KV_TILES=3, seed=0):
+    """Full online-softmax flow across several 32-key tiles, at the
+    fragment level. The subtlety this pins down: the PV accumulator's C
+    layout spreads each lane's 16 registers over 16 DIFFERENT q rows
+    ((r&3)+8*(r>>2)+4*(lane>>5)), while the softmax state (m, l) lives in
+    the lane owning q = lane&31 — so the per-tile rescale factor alpha[q]
+    must be broadcast q->(lane, reg). Modeled here as a 32-float LDS
+    round-trip per tile (alpha written by lanes 0..31, read per register),
+    which is the cheapest option at 16 ds_read_b32 per lane per tile.
+    """
+    rng = np.random.default_rng(seed)
+    KV = 32 * KV_TILES
+    Q = rng.standard_normal((32, D))
+    K = rng.standard_normal((KV, D))
+    V = rng.standard_normal((KV, D))
+
+    m_run = np.full(LANES, -np.inf)          # per lane: q = lane&31
+    l_run = np.zeros(LANES)
+    out_acc = {d0: np.zeros((LANES, REGS)) for d0 in range(0, D, 32)}
+
+    for t in range(KV_TILES):
+        Kt, Vt = K[32 * t:32 * t + 32], V[32 * t:32 * t + 32]
+        acc = np.zeros((LANES, REGS))
+        for k0 in range(0, D, 16):
+            acc = mfma_32x32x16(pack_a(Kt[:, k0:k0 + 16]),
+                                pack_b(Q[:, k0:k0 + 16].T), acc)
+        m_half = acc.max(axis=1)
+        m_tile = np.maximum(m_half, permlane32_swap(m_half))
+        m_new = np.maximum(m_run, m_tile)
+        alpha = np.exp(m_run - m_new)
+        alpha[np.isnan(alpha)] = 0.0          # first tile: -inf - -inf
+        p = np.exp(acc - m_new[:, None])
+        l_half = p.sum(axis=1)
+        l_run = l_run * alpha + l_half + permlane32_swap(l_half)
+
+        # alpha broadcast q -> (lane, reg) through a 32-slot LDS row
+        alpha_lds = alpha[:32]                # lane&31 owns q; halves agree
+        assert np.allclose(alpha[:32], alpha[32:])
+        for d0, acc_o in out_acc.items():
+            for l in range(LANES):
+                for r in range(REGS):
+                    qrow, _ = c_frag_index(l, r)
+                    acc_o[l, r] *= alpha_lds[qrow]
+
+        swapped = permlane32_swap(p)
+
+        def p_value(lane, key):
+            own_half = (lane >> 5)
+            key_half = (key >> 2) & 1
+            r = (key & 3) + 4 * ((key >> 3) & 3)
+            return p[lane, r] if key_half == own_half else swapped[lane, r]
+
+        pa = np.zeros((LANES, 8, 2))
+        for l in range(LANES):
+            for ks in range(2):
+                for e in range(8):
+                    pa[l, e, ks] = p_value(l, 16 * ks + 8 * (l >> 5) + e)
+        for d0 in out_acc:
+            for ks in range(2):
+                out_acc[d0] = mfma_32x32x16(
+                    pa[:, :, ks], pack_b(Vt[16 * ks:16 * ks + 16,
+                                            d0:d0 + 32]), out_acc[d0])
+        m_run = m_new
+
+    out = np.zeros((32, D))
+    for d0, acc_o in out_acc.items():
+        tile = unpack_c(acc_o)
+        for l in range(LANES):
+            for r in range(REGS):
+                qrow, col = c_frag_index(l, r)
+                tile[qrow, col] = acc_o[l, r] / l_run[qrow]
+        out[:, d0:d0 + 32] = tile
+
+    S = Q @ K.T
+    P = np.exp(S - S.max(axis=1, keepdims=True))
+    ref = (P / P.sum(axis=1, keepdims=True)) @ V
+    return out, ref
+
+
 # ------------------------------------------------------- LDS bank conflicts
 
 def bank_conflicts_store(addr_bytes):

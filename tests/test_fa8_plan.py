@@ -46,3 +46,10 @@ def test_v_tile_direct_layout_is_structurally_conflicted():
     for sw in (plan.no_swizzle, plan.d64_swizzle, plan.guide_swizzle):
         _, rd = plan.v_tile_conflicts(sw)
         assert rd >= 4
+
+
+def test_online_rescale_across_tiles():
+    """Multi-tile online softmax at the fragment level, including the
+    alpha[q] -> (lane, reg) broadcast the PV accumulator layout forces."""
+    out, ref = plan.simulate_online_attention(D=64, KV_TILES=3, seed=2)
+    assert np.abs(out - ref).max() < 1e-12
