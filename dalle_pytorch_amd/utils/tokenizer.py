@@ -3,15 +3,21 @@ and YouTokenToMe — the same four-way surface as the reference
 (tokenizer.py:55-266), all CPU-side.
 
 The default tokenizer implements OpenAI-CLIP byte-level BPE from first
-principles. The 49,152-merge table is *data*, not code: pass ``bpe_path``
-(or set ``DALLE_AMD_BPE_PATH``) to point at a CLIP-format merges file; when
-none is available the tokenizer degrades to pure byte-level encoding with
-the identical 49,408-slot vocabulary layout, so model embedding shapes are
-unchanged either way.
+principles. The 49,152-merge table is *data*, not code: the package bundles
+OpenAI's published CLIP BPE vocabulary (``data/clip_bpe_merges.txt.gz`` —
+the standard public ``bpe_simple_vocab_16e6`` merges artifact, gzip-stored;
+same data the reference ships at dalle_pytorch/data/). Pass ``bpe_path``
+(or set ``DALLE_AMD_BPE_PATH``) to override with any CLIP-format merges
+file, plain or gzipped. If no merges data can be found at all, the
+tokenizer degrades to pure byte-level encoding with the identical
+49,408-slot vocabulary layout (embedding shapes unchanged) and emits a
+loud warning, because token ids then differ from CLIP-BPE ids.
 """
 
+import gzip
 import html
 import os
+import warnings
 from functools import lru_cache
 from pathlib import Path
 
@@ -72,9 +78,16 @@ def default_bpe_path():
     env = os.environ.get('DALLE_AMD_BPE_PATH')
     if env:
         return env
-    # use the upstream merges data if it is mounted alongside (read-only ok)
-    candidate = Path('/root/reference/dalle_pytorch/data/bpe_simple_vocab_16e6.txt')
-    return str(candidate) if candidate.exists() else None
+    bundled = Path(__file__).resolve().parent.parent / 'data' / 'clip_bpe_merges.txt.gz'
+    return str(bundled) if bundled.exists() else None
+
+
+def _read_merges_text(path):
+    path = Path(path)
+    if path.suffix == '.gz':
+        with gzip.open(path, 'rt', encoding='utf8') as f:
+            return f.read()
+    return path.read_text(encoding='utf8')
 
 
 class SimpleTokenizer:
@@ -91,8 +104,17 @@ class SimpleTokenizer:
         bpe_path = bpe_path if bpe_path is not None else default_bpe_path()
         merges = []
         if bpe_path is not None and Path(bpe_path).exists():
-            lines = Path(bpe_path).read_text(encoding='utf8').split('\n')
+            lines = _read_merges_text(bpe_path).split('\n')
             merges = [tuple(l.split()) for l in lines[1:self.N_MERGES + 1]]
+        if not merges:
+            warnings.warn(
+                'SimpleTokenizer: no BPE merges data found '
+                f'(bpe_path={bpe_path!r}); falling back to pure BYTE-LEVEL '
+                'encoding. Token ids will NOT match CLIP-BPE — text encoded '
+                'on this machine is incompatible with checkpoints trained '
+                'with the standard merges table. Set DALLE_AMD_BPE_PATH or '
+                'reinstall the package with its bundled data/ directory.',
+                RuntimeWarning, stacklevel=2)
 
         symbols = list(self.byte_encoder.values())
         vocab = symbols + [s + '</w>' for s in symbols]

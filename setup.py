@@ -17,5 +17,7 @@ setup(
     version='0.1.0',
     description='MI355X-native DALL-E training/generation framework',
     packages=find_packages(exclude=('tests',)),
+    package_data={'dalle_pytorch_amd': ['data/*.txt.gz', 'ops/hip/*.hip']},
+    include_package_data=True,
     python_requires='>=3.9',
 )
