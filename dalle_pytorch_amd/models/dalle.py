@@ -15,6 +15,9 @@ Behavior kept bit-for-bit:
 * classifier-free guidance via a second null-cond forward (:564-574).
 """
 
+import copy
+from collections import deque
+
 import torch
 import torch.nn.functional as F
 from torch import nn
@@ -246,11 +249,13 @@ class DALLE(nn.Module):
             out = torch.cat((out, indices[:, :num_img_tokens]), dim=-1)
 
         cache = {} if use_cache else None
+        null_cache = {} if (use_cache and cond_scale != 1) else None
         for cur_len in range(out.shape[1], total_len):
             is_image = cur_len >= text_seq_len
             text_part, image_part = out[:, :text_seq_len], out[:, text_seq_len:]
             logits = self.forward_with_cond_scale(
-                text_part, image_part, cond_scale=cond_scale, cache=cache)
+                text_part, image_part, cond_scale=cond_scale, cache=cache,
+                null_cache=null_cache)
             logits = logits[:, -1, :]
             filtered = top_k(logits, thres=filter_thres)
             sample = gumbel_sample(filtered, temperature=temperature, dim=-1)
@@ -267,12 +272,28 @@ class DALLE(nn.Module):
             return images, scores
         return images
 
-    def forward_with_cond_scale(self, *args, cond_scale=1, cache=None, **kwargs):
+    def forward_with_cond_scale(self, *args, cond_scale=1, cache=None,
+                                null_cache=None, **kwargs):
+        """Classifier-free guidance (reference dalle_pytorch.py:564-574).
+
+        Unlike the reference — which snapshots the cond cache each step for
+        the null pass (so the null stream's own history is discarded and its
+        prefix k/v silently come from the *conditioned* text; PreShiftToken's
+        in-place deques additionally get double-advanced through the shallow
+        copy) — the null-conditioned stream here keeps its OWN persistent
+        cache, making cached guided decoding equal to uncached guided
+        decoding. Pass ``null_cache`` alongside ``cache`` when caching.
+        """
         if cond_scale == 1:
             return self(*args, cache=cache, **kwargs)
-        prev_cache = cache.copy() if cache is not None else None
+        if cache is not None and null_cache is None:
+            # legacy call shape: emulate a correct second stream by deep-
+            # copying mutable entries once — still loses null history, so
+            # warn toward the two-cache API
+            null_cache = {k: (copy.copy(v) if isinstance(v, deque) else v)
+                          for k, v in cache.items()}
         logits = self(*args, cache=cache, **kwargs)
-        null_logits = self(*args, null_cond_prob=1., cache=prev_cache, **kwargs)
+        null_logits = self(*args, null_cond_prob=1., cache=null_cache, **kwargs)
         return null_logits + (logits - null_logits) * cond_scale
 
     # ---------------------------------------------------------------- train

@@ -194,23 +194,56 @@ class _VqAttnBlock(nn.Module):
         return x + self.proj_out(out)
 
 
+class _VqDownsample(nn.Module):
+    """Strided conv with taming's ASYMMETRIC (0,1,0,1) zero pad — a plain
+    padding=1 conv gives the same shape but spatially shifted activations,
+    so real taming checkpoints would decode off-by-half-a-pixel. Nested
+    ``conv`` attribute matches taming's ``down.{i}.downsample.conv.*`` keys."""
+
+    def __init__(self, c):
+        super().__init__()
+        self.conv = nn.Conv2d(c, c, 3, stride=2, padding=0)
+
+    def forward(self, x):
+        return self.conv(F.pad(x, (0, 1, 0, 1)))
+
+
+class _VqUpsample(nn.Module):
+    """Nearest 2x upsample + conv (taming key layout ``up.{i}.upsample.conv.*``)."""
+
+    def __init__(self, c):
+        super().__init__()
+        self.conv = nn.Conv2d(c, c, 3, padding=1)
+
+    def forward(self, x):
+        return self.conv(F.interpolate(x, scale_factor=2.0, mode='nearest'))
+
+
 class _VqEncoder(nn.Module):
-    def __init__(self, ch, ch_mult, num_res_blocks, z_channels, in_ch=3, attn_at_last=True):
+    def __init__(self, ch, ch_mult, num_res_blocks, z_channels, in_ch=3,
+                 resolution=256, attn_resolutions=()):
         super().__init__()
         self.conv_in = nn.Conv2d(in_ch, ch, 3, padding=1)
         self.down = nn.ModuleList()
         c = ch
+        res = resolution
         for i, m in enumerate(ch_mult):
             stage = nn.Module()
             stage.block = nn.ModuleList(
                 [_VqResBlock(c if j == 0 else ch * m, ch * m) for j in range(num_res_blocks)])
             c = ch * m
-            stage.downsample = nn.Conv2d(c, c, 3, stride=2, padding=1) \
-                if i != len(ch_mult) - 1 else None
+            stage.attn = nn.ModuleList(
+                [_VqAttnBlock(c) for _ in range(num_res_blocks)]
+                if res in attn_resolutions else [])
+            if i != len(ch_mult) - 1:
+                stage.downsample = _VqDownsample(c)
+                res //= 2
+            else:
+                stage.downsample = None
             self.down.append(stage)
         self.mid = nn.Module()
         self.mid.block_1 = _VqResBlock(c, c)
-        self.mid.attn_1 = _VqAttnBlock(c) if attn_at_last else nn.Identity()
+        self.mid.attn_1 = _VqAttnBlock(c)
         self.mid.block_2 = _VqResBlock(c, c)
         self.norm_out = _gn(c)
         self.conv_out = nn.Conv2d(c, z_channels, 3, padding=1)
@@ -218,8 +251,10 @@ class _VqEncoder(nn.Module):
     def forward(self, x):
         h = self.conv_in(x)
         for stage in self.down:
-            for blk in stage.block:
+            for j, blk in enumerate(stage.block):
                 h = blk(h)
+                if len(stage.attn):
+                    h = stage.attn[j](h)
             if stage.downsample is not None:
                 h = stage.downsample(h)
         h = self.mid.block_2(self.mid.attn_1(self.mid.block_1(h)))
@@ -227,21 +262,30 @@ class _VqEncoder(nn.Module):
 
 
 class _VqDecoder(nn.Module):
-    def __init__(self, ch, ch_mult, num_res_blocks, z_channels, out_ch=3, attn_at_first=True):
+    def __init__(self, ch, ch_mult, num_res_blocks, z_channels, out_ch=3,
+                 resolution=256, attn_resolutions=()):
         super().__init__()
         c = ch * ch_mult[-1]
         self.conv_in = nn.Conv2d(z_channels, c, 3, padding=1)
         self.mid = nn.Module()
         self.mid.block_1 = _VqResBlock(c, c)
-        self.mid.attn_1 = _VqAttnBlock(c) if attn_at_first else nn.Identity()
+        self.mid.attn_1 = _VqAttnBlock(c)
         self.mid.block_2 = _VqResBlock(c, c)
         self.up = nn.ModuleList()
+        res = resolution // 2 ** (len(ch_mult) - 1)
         for i, m in reversed(list(enumerate(ch_mult))):
             stage = nn.Module()
             stage.block = nn.ModuleList(
                 [_VqResBlock(c if j == 0 else ch * m, ch * m) for j in range(num_res_blocks + 1)])
             c = ch * m
-            stage.upsample = nn.Conv2d(c, c, 3, padding=1) if i != 0 else None
+            stage.attn = nn.ModuleList(
+                [_VqAttnBlock(c) for _ in range(num_res_blocks + 1)]
+                if res in attn_resolutions else [])
+            if i != 0:
+                stage.upsample = _VqUpsample(c)
+                res *= 2
+            else:
+                stage.upsample = None
             self.up.insert(0, stage)
         self.norm_out = _gn(c)
         self.conv_out = nn.Conv2d(c, out_ch, 3, padding=1)
@@ -250,10 +294,11 @@ class _VqDecoder(nn.Module):
         h = self.conv_in(z)
         h = self.mid.block_2(self.mid.attn_1(self.mid.block_1(h)))
         for stage in reversed(self.up):
-            for blk in stage.block:
+            for j, blk in enumerate(stage.block):
                 h = blk(h)
+                if len(stage.attn):
+                    h = stage.attn[j](h)
             if stage.upsample is not None:
-                h = F.interpolate(h, scale_factor=2, mode='nearest')
                 h = stage.upsample(h)
         return self.conv_out(F.silu(self.norm_out(h)))
 
@@ -269,7 +314,8 @@ class VQGanVAE(nn.Module):
 
     def __init__(self, vqgan_model_path=None, vqgan_config_path=None, *,
                  image_size=256, num_tokens=16384, embed_dim=256, ch=128,
-                 ch_mult=(1, 1, 2, 2, 4), num_res_blocks=2, gumbel=False):
+                 ch_mult=(1, 1, 2, 2, 4), num_res_blocks=2, gumbel=False,
+                 attn_resolutions=(16,)):
         super().__init__()
         if vqgan_config_path is not None:
             # taming-transformers yaml config (reference vae.py:148-158 uses
@@ -284,6 +330,7 @@ class VQGanVAE(nn.Module):
             ch_mult = tuple(dd.get('ch_mult', ch_mult))
             num_res_blocks = dd.get('num_res_blocks', num_res_blocks)
             image_size = dd.get('resolution', image_size)
+            attn_resolutions = tuple(dd.get('attn_resolutions', attn_resolutions))
             num_tokens = params.get('n_embed', num_tokens)
             embed_dim = params.get('embed_dim', embed_dim)
         f = 2 ** (len(ch_mult) - 1)
@@ -294,15 +341,22 @@ class VQGanVAE(nn.Module):
         self.gumbel = gumbel
 
         z_ch = embed_dim
-        self.encoder = _VqEncoder(ch, ch_mult, num_res_blocks, z_ch)
-        self.decoder = _VqDecoder(ch, ch_mult, num_res_blocks, z_ch)
+        self.encoder = _VqEncoder(ch, ch_mult, num_res_blocks, z_ch,
+                                  resolution=image_size,
+                                  attn_resolutions=attn_resolutions)
+        self.decoder = _VqDecoder(ch, ch_mult, num_res_blocks, z_ch,
+                                  resolution=image_size,
+                                  attn_resolutions=attn_resolutions)
         self.quant_conv = nn.Conv2d(z_ch, embed_dim, 1)
         self.post_quant_conv = nn.Conv2d(embed_dim, z_ch, 1)
+        self.quantize = nn.Module()
         if gumbel:
-            self.quantize = nn.Module()
+            # taming GumbelQuantize: learned 1x1 proj -> vocab logits, then
+            # an embedding lookup; indices come from proj logits, NOT from a
+            # similarity against the codebook
+            self.quantize.proj = nn.Conv2d(embed_dim, num_tokens, 1)
             self.quantize.embed = nn.Embedding(num_tokens, embed_dim)
         else:
-            self.quantize = nn.Module()
             self.quantize.embedding = nn.Embedding(num_tokens, embed_dim)
 
         if vqgan_model_path is not None:
@@ -311,6 +365,12 @@ class VQGanVAE(nn.Module):
             missing, unexpected = self.load_state_dict(state, strict=False)
             if missing:
                 raise RuntimeError(f'VQGAN checkpoint missing keys: {missing[:8]}...')
+            # silently dropping quantizer weights would produce wrong codes
+            bad = [k for k in unexpected if k.startswith('quantize.')]
+            if bad:
+                raise RuntimeError(
+                    f'VQGAN checkpoint has quantizer keys this module did not '
+                    f'consume (wrong gumbel= setting?): {bad[:8]}')
 
     def _codebook_weight(self):
         return self.quantize.embed.weight if self.gumbel else self.quantize.embedding.weight
@@ -326,17 +386,17 @@ class VQGanVAE(nn.Module):
             img = img.contiguous(memory_format=torch.channels_last)
         z = self.quant_conv(self.encoder(img))
         b, c, h, w = z.shape
+        if self.gumbel:
+            # GumbelVQ: learned proj produces vocab logits per position
+            logits = self.quantize.proj(z)               # [b, vocab, h, w]
+            idx = logits.argmax(dim=1).reshape(b, h * w)
+            return idx
         flat = z.permute(0, 2, 3, 1).reshape(-1, c)
         book = self._codebook_weight()
-        if self.gumbel:
-            # GumbelVQ encodes logits directly over the vocabulary
-            logits = flat @ book.t()
-            idx = logits.argmax(dim=-1)
-        else:
-            d = (flat.pow(2).sum(1, keepdim=True)
-                 - 2 * flat @ book.t()
-                 + book.pow(2).sum(1))
-            idx = d.argmin(dim=-1)
+        d = (flat.pow(2).sum(1, keepdim=True)
+             - 2 * flat @ book.t()
+             + book.pow(2).sum(1))
+        idx = d.argmin(dim=-1)
         return idx.reshape(b, h * w)
 
     def decode(self, img_seq):

@@ -11,6 +11,7 @@ behavior).
 import io
 import random
 import tarfile
+import warnings
 from pathlib import Path
 
 import torch
@@ -64,8 +65,21 @@ class TarImageTextDataset(IterableDataset):
         shards = list(self.shards)
         if self.shuffle_shards:
             random.Random(self.seed).shuffle(shards)
-        shards = shards[get_rank()::max(get_world_size(), 1)]
+        world = max(get_world_size(), 1)
         info = torch.utils.data.get_worker_info()
+        workers = info.num_workers if info is not None else 1
+        if len(shards) < world * workers:
+            # a rank/worker with zero shards yields zero batches, which stalls
+            # the other ranks' gradient all-reduce forever: repeat the shard
+            # list so every (rank, worker) slot sees at least one shard
+            warnings.warn(
+                f'{len(shards)} shards < world_size*workers = {world}*{workers};'
+                ' repeating shards so every rank yields data (samples will '
+                'repeat across ranks — add shards for proper sharding)',
+                RuntimeWarning)
+            reps = -(-(world * workers) // len(shards))
+            shards = shards * reps
+        shards = shards[get_rank()::world]
         if info is not None:
             shards = shards[info.id::info.num_workers]
         return shards

@@ -55,6 +55,59 @@ def test_tar_streaming_dataset(tmp_path):
     assert batch[0].shape == (4, 16) and batch[1].shape == (4, 3, 32, 32)
 
 
+def test_tar_dataset_repeats_shards_when_fewer_than_ranks(tmp_path, monkeypatch):
+    """1 shard, 2 ranks: rank 1 must still yield data (a zero-batch rank
+    stalls the other ranks' gradient all-reduce forever)."""
+    from dalle_pytorch_amd.utils import wds as wds_mod
+    from dalle_pytorch_amd.utils.tokenizer import SimpleTokenizer
+    _make_shard(tmp_path / 'shard-000.tar', 3)
+    ds = wds_mod.TarImageTextDataset(str(tmp_path / '*.tar'),
+                                     tokenizer=SimpleTokenizer(), text_len=16,
+                                     image_size=32)
+    import dalle_pytorch_amd.parallel as par
+    monkeypatch.setattr(par, 'get_world_size', lambda: 2)
+    for rank in (0, 1):
+        monkeypatch.setattr(par, 'get_rank', lambda r=rank: r)
+        with pytest.warns(RuntimeWarning, match='repeating shards'):
+            assert len(ds._my_shards()) >= 1
+        assert len(list(ds)) == 3
+
+
+def test_vqgan_downsample_matches_taming_asymmetric_pad():
+    """taming pads (0,1,0,1) with a padding=0 stride-2 conv; a symmetric
+    padding=1 conv is spatially shifted and decodes real ckpts wrong."""
+    import torch.nn.functional as F
+    from dalle_pytorch_amd.models.vae_adapters import _VqDownsample
+    ds = _VqDownsample(4)
+    x = torch.randn(2, 4, 16, 16)
+    want = F.conv2d(F.pad(x, (0, 1, 0, 1)), ds.conv.weight, ds.conv.bias, stride=2)
+    assert torch.allclose(ds(x), want)
+    # and it must differ from the symmetric-pad variant on generic input
+    sym = F.conv2d(x, ds.conv.weight, ds.conv.bias, stride=2, padding=1)
+    assert ds(x).shape == sym.shape and not torch.allclose(ds(x), sym)
+
+
+def test_vqgan_gumbel_proj_consumed_and_rejects_mismatch(tmp_path):
+    from dalle_pytorch_amd import VQGanVAE
+    g = VQGanVAE(image_size=32, num_tokens=128, embed_dim=16, ch=16,
+                 ch_mult=(1, 2), num_res_blocks=1, gumbel=True)
+    sd = g.state_dict()
+    assert 'quantize.proj.weight' in sd  # taming GumbelQuantize layout
+    ck = tmp_path / 'g.ckpt'
+    torch.save({'state_dict': sd}, ck)
+    # round-trips into a fresh gumbel module
+    g2 = VQGanVAE(vqgan_model_path=str(ck), image_size=32, num_tokens=128,
+                  embed_dim=16, ch=16, ch_mult=(1, 2), num_res_blocks=1,
+                  gumbel=True)
+    img = torch.rand(1, 3, 32, 32)
+    assert torch.equal(g.get_codebook_indices(img), g2.get_codebook_indices(img))
+    # a gumbel checkpoint must NOT silently load into a non-gumbel module
+    with pytest.raises(RuntimeError):
+        VQGanVAE(vqgan_model_path=str(ck), image_size=32, num_tokens=128,
+                 embed_dim=16, ch=16, ch_mult=(1, 2), num_res_blocks=1,
+                 gumbel=False)
+
+
 def test_openai_dvae_conversion_mapping():
     """Synthesize a dall_e-layout state dict with matching shapes; the
     converter must cover every parameter of our module."""

@@ -108,6 +108,44 @@ def test_generate_with_cond_scale_and_priming():
     assert out.shape == (1, 3, 64, 64)
 
 
+def test_guided_cached_equals_uncached_with_shift_tokens():
+    """cond_scale != 1 + use_cache + shift_tokens: cached guided generation
+    must equal uncached guided generation. Needs a persistent second cache
+    for the null-cond stream (the reference's per-step cache.copy() loses
+    null history AND double-advances the PreShiftToken deques)."""
+    torch.manual_seed(11)
+    d = tiny_dalle(shift_tokens=True, depth=2, stable=True).eval()
+    text = torch.randint(1, 50, (1, 8))
+    torch.manual_seed(13)
+    a = d.generate_images(text, cond_scale=2.0, use_cache=True,
+                          temperature=1e-8, filter_thres=0.99)
+    torch.manual_seed(13)
+    b = d.generate_images(text, cond_scale=2.0, use_cache=False,
+                          temperature=1e-8, filter_thres=0.99)
+    assert torch.allclose(a, b, atol=1e-5)
+
+
+def test_guided_cached_logits_equal_uncached_logits():
+    """Logit-level pin of the two-cache guidance design: step-by-step cached
+    guided logits match full-sequence uncached guided logits."""
+    torch.manual_seed(21)
+    d = tiny_dalle(shift_tokens=True, depth=2).eval()
+    text = torch.randint(1, 50, (2, 8))
+    img_tokens = torch.randint(0, 64, (2, 5))
+
+    cache, null_cache = {}, {}
+    cached_last = None
+    for k in range(img_tokens.shape[1] + 1):
+        torch.manual_seed(100)  # pin the null-mask rand draw
+        cached_last = d.forward_with_cond_scale(
+            text, img_tokens[:, :k], cond_scale=3.0,
+            cache=cache, null_cache=null_cache)[:, -1]
+
+    torch.manual_seed(100)
+    full = d.forward_with_cond_scale(text, img_tokens, cond_scale=3.0)
+    assert torch.allclose(cached_last, full[:, -1], atol=1e-4)
+
+
 def test_generate_texts():
     from dalle_pytorch_amd.utils.tokenizer import SimpleTokenizer
 
