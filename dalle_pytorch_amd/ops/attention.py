@@ -27,7 +27,8 @@ from dalle_pytorch_amd.ops.dispatch import hip_module, using_eager_fallback
 _SUPPORTED_HEAD_DIMS = (64,)
 
 
-def _eager_attention(q, k, v, scale, causal, key_mask, static_mask):
+def _eager_attention(q, k, v, scale, causal, key_mask, static_mask,
+                     return_lse=False):
     dots = torch.matmul(q * scale, k.transpose(-1, -2))
     big_neg = -torch.finfo(dots.dtype).max
     if key_mask is not None:
@@ -39,11 +40,14 @@ def _eager_attention(q, k, v, scale, causal, key_mask, static_mask):
     if static_mask is not None:
         dots = dots.masked_fill(~static_mask, big_neg)
     attn = dots.softmax(dim=-1)
-    return torch.matmul(attn, v)
+    out = torch.matmul(attn, v)
+    if return_lse:
+        return out, torch.logsumexp(dots.float(), dim=-1)
+    return out
 
 
 def _flash_bwd_composite(q, k, v, out, lse, dout, scale, causal, key_mask,
-                         static_mask, q_chunk=128):
+                         static_mask, q_chunk=128, grad_lse=None):
     """Flash-style backward at the torch level: recompute P per q-chunk from
     the saved logsumexp, never materializing the full n x n matrix. All
     GEMMs run in bf16 on rocBLAS MFMA paths (fp32 accumulation inside);
@@ -80,6 +84,8 @@ def _flash_bwd_composite(q, k, v, out, lse, dout, scale, causal, key_mask,
         dv += torch.matmul(pb.transpose(-1, -2), doc).float()
         dp = torch.matmul(doc, v.transpose(-1, -2)).float()
         Dc = (doc.float() * oc.float()).sum(dim=-1, keepdim=True)
+        if grad_lse is not None:
+            Dc = Dc - grad_lse[:, :, c0:c1].unsqueeze(-1)
         ds = (p * (dp - Dc) * scale).to(q.dtype)
         dq[:, :, c0:c1] = torch.matmul(ds, k)
         dk += torch.matmul(ds.transpose(-1, -2), qc).float()
@@ -106,26 +112,28 @@ class _FlashAttention(torch.autograd.Function):
         ctx.key_mask, ctx.static_mask = key_mask, static_mask
         ctx.tile_map, ctx.tile_map_t = tile_map, tile_map_t
         ctx.fold_heads = fold_heads
-        return out
+        return out, lse
 
     @staticmethod
-    def backward(ctx, dout):
+    def backward(ctx, dout, dlse):
         q, k, v, out, lse = ctx.saved_tensors
         ext = hip_module()
+        if dlse is not None:
+            dlse = dlse.contiguous().float()
         if hasattr(ext, 'fa_bwd'):
             # with fold_heads, out/dout stay in their contiguous [b,n,h,d]
             # layout — the bwd kernels read them with bnhd strides directly
             dq, dk, dv = ext.fa_bwd(
                 q, k, v, out, lse, dout.contiguous(),
                 ctx.scale, ctx.causal, ctx.key_mask, ctx.static_mask,
-                ctx.tile_map, ctx.tile_map_t, ctx.fold_heads)
+                ctx.tile_map, ctx.tile_map_t, ctx.fold_heads, dlse)
         else:
             if ctx.fold_heads:
                 out = out.permute(0, 2, 1, 3)
                 dout = dout.permute(0, 2, 1, 3)
             dq, dk, dv = _flash_bwd_composite(
                 q, k, v, out, lse, dout.contiguous(), ctx.scale, ctx.causal,
-                ctx.key_mask, ctx.static_mask)
+                ctx.key_mask, ctx.static_mask, grad_lse=dlse)
         return dq, dk, dv, None, None, None, None, None, None, None
 
 
@@ -151,7 +159,8 @@ def _fold(out):
 
 
 def attention_core(q, k, v, scale, causal=True, key_mask=None, static_mask=None,
-                   static_tiles=None, static_tiles_t=None, fold_heads=False):
+                   static_tiles=None, static_tiles_t=None, fold_heads=False,
+                   return_lse=False):
     """Scaled-dot-product attention with the reference's masking semantics.
 
     q: [b, h, nq, d] (unscaled), k/v: [b, h, nk, d],
@@ -161,31 +170,35 @@ def attention_core(q, k, v, scale, causal=True, key_mask=None, static_mask=None,
     static_tiles: optional uint8 [ceil(nq/64), ceil(nk/32)] block map of
     static_mask — fully-zero tiles are skipped by the kernel (this is how
     axial/conv/block-sparse patterns become truly sparse),
-    fold_heads: return [b, nq, h*d] (GPU kernel writes it directly).
-    Returns [b, h, nq, d] or [b, nq, h*d].
+    fold_heads: return [b, nq, h*d] (GPU kernel writes it directly),
+    return_lse: also return the per-row logsumexp [b, h, nq] (fp32),
+    DIFFERENTIABLE — partial attentions combined by lse-merge backprop
+    correctly through both outputs (the axial decomposition path).
+    Returns [b, h, nq, d] or [b, nq, h*d] (+ lse).
     """
-    if using_eager_fallback(q):
-        out = _eager_attention(q, k, v, scale, causal, key_mask, static_mask)
-        return _fold(out) if fold_heads else out
-    if not _hip_supported(q, k, causal, key_mask):
-        if q.shape[-2] >= 32:  # decode fallback is intentional and silent
+    if using_eager_fallback(q) or not _hip_supported(q, k, causal, key_mask):
+        if q.is_cuda and q.shape[-2] >= 32:
             key = (q.shape[-1], str(q.dtype))
             if key not in _warned_shapes:
                 _warned_shapes.add(key)
                 warnings.warn(f'attention_core: shape/dtype {key} not covered '
                               'by the HIP kernel yet; using eager path on GPU')
-        out = _eager_attention(q, k, v, scale, causal, key_mask, static_mask)
-        return _fold(out) if fold_heads else out
+        res = _eager_attention(q, k, v, scale, causal, key_mask, static_mask,
+                               return_lse=return_lse)
+        out, lse = res if return_lse else (res, None)
+        out = _fold(out) if fold_heads else out
+        return (out, lse) if return_lse else out
     if static_mask is not None:
         static_mask = static_mask.contiguous()
     if key_mask is not None:
         key_mask = key_mask.contiguous()
-    out = _FlashAttention.apply(q, k, v, scale, causal, key_mask, static_mask,
-                                static_tiles, static_tiles_t, fold_heads)
+    out, lse = _FlashAttention.apply(q, k, v, scale, causal, key_mask,
+                                     static_mask, static_tiles, static_tiles_t,
+                                     fold_heads)
     if fold_heads:
         b, n, h, d = out.shape
-        return out.view(b, n, h * d)
-    return out
+        out = out.view(b, n, h * d)
+    return (out, lse) if return_lse else out
 
 
 def build_tile_map(static_mask, causal=False):

@@ -1707,7 +1707,8 @@ std::vector<torch::Tensor> fa_bwd(torch::Tensor q, torch::Tensor k, torch::Tenso
                                   std::optional<torch::Tensor> static_mask,
                                   std::optional<torch::Tensor> tile_map,
                                   std::optional<torch::Tensor> tile_map_t,
-                                  bool out_bnhd) {
+                                  bool out_bnhd,
+                                  std::optional<torch::Tensor> grad_lse) {
   CHK(q.is_cuda() && q.dtype() == torch::kBFloat16);
   CHK(q.is_contiguous() && k.is_contiguous() && v.is_contiguous());
   CHK(out.is_contiguous() && dout.is_contiguous());
@@ -1720,6 +1721,13 @@ std::vector<torch::Tensor> fa_bwd(torch::Tensor q, torch::Tensor k, torch::Tenso
                        reinterpret_cast<const short*>(dout.data_ptr()),
                        reinterpret_cast<const short*>(out.data_ptr()),
                        Dv.data_ptr<float>(), b, h, nq, out_bnhd ? 1 : 0);
+  }
+  if (grad_lse.has_value()) {
+    // dL/dS_j = P_j * (dP_j - (D - g_lse)): an upstream logsumexp gradient
+    // (the axial lse-merge path) folds into the per-row delta term
+    CHK(grad_lse->dtype() == torch::kFloat32);
+    CHK(grad_lse->sizes() == Dv.sizes());
+    Dv.sub_(*grad_lse);
   }
 
   auto dq = torch::empty_like(q);

@@ -116,6 +116,36 @@ def test_axial_attention_matches_static_mask_dense():
         assert torch.allclose(sparse(x), dense(x), atol=1e-5), f'axis={axis}'
 
 
+def test_axial_lse_merge_gradients_match_masked_path(monkeypatch):
+    """The decomposed (lse-merge) axial forward must produce the same
+    gradients as the masked-dense formulation — exercises the logsumexp
+    gradient path through attention_core's backward (Dv - grad_lse)."""
+    torch.manual_seed(2)
+    S, text_len = 4, 3
+    seq_len = text_len + S * S - 1
+    for axis in (0, 1):
+        sparse = SparseAxialCausalAttention(
+            dim=32, seq_len=seq_len, image_size=S, axis=axis, heads=2,
+            dim_head=16)
+        x = torch.randn(2, seq_len, 32, requires_grad=True)
+        mask = torch.ones(2, text_len, dtype=torch.bool)
+        mask[1, -1] = False
+
+        monkeypatch.setenv('DALLE_AMD_AXIAL_MASKED', '1')
+        ref = sparse(x, mask=mask)
+        gref = torch.autograd.grad(ref.square().sum(), (x,) + tuple(
+            sparse.parameters()), retain_graph=False)
+
+        monkeypatch.setenv('DALLE_AMD_AXIAL_MASKED', '0')
+        out = sparse(x, mask=mask)
+        gnew = torch.autograd.grad(out.square().sum(), (x,) + tuple(
+            sparse.parameters()))
+
+        assert torch.allclose(out, ref, atol=1e-5), f'axis={axis}'
+        for a, b_ in zip(gnew, gref):
+            assert torch.allclose(a, b_, atol=1e-4), f'axis={axis}'
+
+
 def test_conv_attention_matches_dense_mask():
     """conv_like == dense attention under the unfolded-neighborhood mask."""
     torch.manual_seed(2)
