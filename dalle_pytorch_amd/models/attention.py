@@ -30,7 +30,7 @@ import torch.nn.functional as F
 from torch import nn
 
 from dalle_pytorch_amd.models.positional import apply_rotary_to_qkv
-from dalle_pytorch_amd.ops import attention_core
+from dalle_pytorch_amd.ops import attention_core, axial_attention
 from dalle_pytorch_amd.ops.attention import build_tile_map
 from dalle_pytorch_amd.ops.rope import rope_split, rope_split_supported, trig_tables
 
@@ -285,66 +285,25 @@ class SparseAxialCausalAttention(_StaticMaskSparseAttention):
     # tests/test_attention.py::test_axial_attention_lse_decomposition).
 
     def forward(self, x, mask=None, rotary_pos_emb=None):
-        """Axial attention as three DENSE attentions merged by logsumexp —
-        no pattern masks, no partial MFMA tiles:
-
-        (a) text queries: plain causal over the text prefix,
-        (b) image queries over ALL text keys (non-causal), keeping lse,
-        (c) image queries over their own grid row/col (causal), with the
-            grid axis folded into the head-batch dimension (a free reshape
-            for axis 0, one transposed copy for axis 1), keeping lse,
-        then out_img = lse-weighted average of (b) and (c). Exactness is
-        pinned against the masked-dense oracle; gradients flow through the
-        merge via the differentiable lse of attention_core.
-        """
+        """Axial attention without pattern masks: on GPU, ONE fused kernel
+        call in axial mode (liveness evaluated arithmetically, virtual
+        column-major coordinates for axis 1 — no transposes, no mask/tile-map
+        traffic, no partial tiles beyond the grid lines themselves).
+        Elsewhere, the lse-merge decomposition into three dense attentions
+        (ops.attention.axial_attention). Set DALLE_AMD_AXIAL_MASKED=1 to
+        force the round-1 masked-dense path for A/B comparison."""
         t = self.text_len
         b, n, _ = x.shape
-        if n <= t or os.environ.get('DALLE_AMD_AXIAL_MASKED', '0') == '1':
+        if (n <= t or not self.causal
+                or os.environ.get('DALLE_AMD_AXIAL_MASKED', '0') == '1'):
             return super().forward(x, mask=mask, rotary_pos_emb=rotary_pos_emb)
 
-        h, S = self.heads, self.image_size
-        n_img = n - t
-        q, k, v = _qkv_heads(x, self.to_qkv, h, self.dim_head,
+        q, k, v = _qkv_heads(x, self.to_qkv, self.heads, self.dim_head,
                              rotary_pos_emb, 0)
-        km_t = self._key_mask(mask, b, t, t, x.device)
-        kt, vt = k[:, :, :t], v[:, :, :t]
-
-        out_text = attention_core(q[:, :, :t], kt, vt, self.scale,
-                                  causal=self.causal, key_mask=km_t,
-                                  fold_heads=True)
-        o1, l1 = attention_core(q[:, :, t:], kt, vt, self.scale, causal=False,
-                                key_mask=km_t, return_lse=True)
-
-        pad = S * S - n_img
-
-        def grid(z):
-            zi = z[:, :, t:]
-            if pad:
-                zi = F.pad(zi, (0, 0, 0, pad))
-            g = zi.reshape(b, h, S, S, -1)
-            if self.axis == 1:
-                g = g.transpose(2, 3)
-            return g.reshape(b, h * S, S, -1).contiguous()
-
-        o2g, l2g = attention_core(grid(q), grid(k), grid(v), self.scale,
-                                  causal=True, return_lse=True)
-
-        def ungrid(z):
-            g = z.reshape(b, h, S, S, *z.shape[3:])
-            if self.axis == 1:
-                g = g.transpose(2, 3)
-            return g.reshape(b, h, S * S, *z.shape[3:])[:, :, :n_img]
-
-        o2 = ungrid(o2g)
-        l2 = ungrid(l2g.unsqueeze(-1)).squeeze(-1)
-
-        mx = torch.maximum(l1, l2)
-        w1 = (l1 - mx).exp().unsqueeze(-1)
-        w2 = (l2 - mx).exp().unsqueeze(-1)
-        out_img = (o1.float() * w1 + o2.float() * w2) / (w1 + w2)
-        out_img = out_img.to(x.dtype).permute(0, 2, 1, 3).reshape(b, n_img, -1)
-
-        return self.to_out(torch.cat((out_text, out_img), dim=1))
+        km = self._key_mask(mask, b, n, t, x.device)
+        out = axial_attention(q, k, v, self.scale, t, self.image_size,
+                              self.axis, key_mask=km)
+        return self.to_out(out)
 
 
 class SparseConvCausalAttention(_StaticMaskSparseAttention):

@@ -8,7 +8,7 @@ runs, which is also what the numerics tests compare the kernels against.
 """
 
 from dalle_pytorch_amd.ops.dispatch import hip_available, hip_module, using_eager_fallback
-from dalle_pytorch_amd.ops.attention import attention_core
+from dalle_pytorch_amd.ops.attention import attention_core, axial_attention
 from dalle_pytorch_amd.ops.fused import geglu
 
-__all__ = ['attention_core', 'geglu', 'hip_available', 'hip_module', 'using_eager_fallback']
+__all__ = ['attention_core', 'axial_attention', 'geglu', 'hip_available', 'hip_module', 'using_eager_fallback']

@@ -102,16 +102,19 @@ class _FlashAttention(torch.autograd.Function):
 
     @staticmethod
     def forward(ctx, q, k, v, scale, causal, key_mask, static_mask,
-                tile_map, tile_map_t, fold_heads):
+                tile_map, tile_map_t, fold_heads, axial=None):
         ext = hip_module()
         q, k, v = (t.contiguous() for t in (q, k, v))
+        ax_t, ax_s, ax_axis = axial if axial is not None else (0, 0, -1)
         out, lse = ext.fa_fwd(q, k, v, scale, causal,
-                              key_mask, static_mask, tile_map, fold_heads)
+                              key_mask, static_mask, tile_map, fold_heads,
+                              ax_t, ax_s, ax_axis)
         ctx.save_for_backward(q, k, v, out, lse)
         ctx.scale, ctx.causal = scale, causal
         ctx.key_mask, ctx.static_mask = key_mask, static_mask
         ctx.tile_map, ctx.tile_map_t = tile_map, tile_map_t
         ctx.fold_heads = fold_heads
+        ctx.axial = (ax_t, ax_s, ax_axis)
         return out, lse
 
     @staticmethod
@@ -126,15 +129,19 @@ class _FlashAttention(torch.autograd.Function):
             dq, dk, dv = ext.fa_bwd(
                 q, k, v, out, lse, dout.contiguous(),
                 ctx.scale, ctx.causal, ctx.key_mask, ctx.static_mask,
-                ctx.tile_map, ctx.tile_map_t, ctx.fold_heads, dlse)
+                ctx.tile_map, ctx.tile_map_t, ctx.fold_heads, dlse,
+                *ctx.axial)
         else:
+            if ctx.axial[2] >= 0:
+                raise RuntimeError('axial-mode backward needs the fa_bwd '
+                                   'kernel (extension built without it)')
             if ctx.fold_heads:
                 out = out.permute(0, 2, 1, 3)
                 dout = dout.permute(0, 2, 1, 3)
             dq, dk, dv = _flash_bwd_composite(
                 q, k, v, out, lse, dout.contiguous(), ctx.scale, ctx.causal,
                 ctx.key_mask, ctx.static_mask, grad_lse=dlse)
-        return dq, dk, dv, None, None, None, None, None, None, None
+        return (dq, dk, dv) + (None,) * 8
 
 
 def _hip_supported(q, k, causal, key_mask):
@@ -199,6 +206,67 @@ def attention_core(q, k, v, scale, causal=True, key_mask=None, static_mask=None,
         b, n, h, d = out.shape
         out = out.view(b, n, h * d)
     return (out, lse) if return_lse else out
+
+
+def axial_attention(q, k, v, scale, text_len, image_size, axis, key_mask=None):
+    """Axial (row/col) causal attention over [text prefix + image grid]
+    (reference attention.py:225-335 semantics). Returns folded [b, n, h*d].
+
+    GPU + extension: ONE fused kernel call in axial mode — the pattern is
+    evaluated arithmetically in virtual (column-major for axis 1) coords; no
+    mask tensors, no tile maps, no transposes.
+    Fallback (CPU / unsupported shapes): lse-merge decomposition into three
+    dense attentions (pinned by tests/test_attention.py).
+    """
+    b, h, n, d = q.shape
+    t, S = text_len, image_size
+    fused = (not using_eager_fallback(q) and _hip_supported(q, k, True, None)
+             and S & (S - 1) == 0 and n > t)
+    if fused:
+        if key_mask is not None:
+            key_mask = key_mask.contiguous()
+        out, _ = _FlashAttention.apply(q, k, v, scale, True, key_mask, None,
+                                       None, None, True, (t, S, 0 if axis == 0 else 1))
+        return out.view(b, n, h * d)
+
+    # ---- decomposition fallback: text-causal + img->text dense + grid-local
+    km_t = key_mask[:, :t].contiguous() if key_mask is not None else None
+    kt, vt = k[:, :, :t], v[:, :, :t]
+    out_text = attention_core(q[:, :, :t], kt, vt, scale, causal=True,
+                              key_mask=km_t, fold_heads=True)
+    if n <= t:
+        return out_text
+    n_img = n - t
+    o1, l1 = attention_core(q[:, :, t:], kt, vt, scale, causal=False,
+                            key_mask=km_t, return_lse=True)
+    pad = S * S - n_img
+
+    def grid(z):
+        zi = z[:, :, t:]
+        if pad:
+            zi = torch.nn.functional.pad(zi, (0, 0, 0, pad))
+        g = zi.reshape(b, h, S, S, -1)
+        if axis == 1:
+            g = g.transpose(2, 3)
+        return g.reshape(b, h * S, S, -1).contiguous()
+
+    o2g, l2g = attention_core(grid(q), grid(k), grid(v), scale, causal=True,
+                              return_lse=True)
+
+    def ungrid(z):
+        g = z.reshape(b, h, S, S, *z.shape[3:])
+        if axis == 1:
+            g = g.transpose(2, 3)
+        return g.reshape(b, h, S * S, *z.shape[3:])[:, :, :n_img]
+
+    o2 = ungrid(o2g)
+    l2 = ungrid(l2g.unsqueeze(-1)).squeeze(-1)
+    mx = torch.maximum(l1, l2)
+    w1 = (l1 - mx).exp().unsqueeze(-1)
+    w2 = (l2 - mx).exp().unsqueeze(-1)
+    out_img = (o1.float() * w1 + o2.float() * w2) / (w1 + w2)
+    out_img = out_img.to(q.dtype).permute(0, 2, 1, 3).reshape(b, n_img, -1)
+    return torch.cat((out_text, out_img), dim=1)
 
 
 def build_tile_map(static_mask, causal=False):

@@ -82,6 +82,30 @@ DEVFN int swz_key(int d, int key) {
 }
 
 
+// ---------------------------------------------------------------------------
+// Axial attention mode (reference attention.py:225-335): instead of a
+// static mask in HBM + tile maps, the axial pattern is evaluated
+// ARITHMETICALLY in a virtual coordinate space. Virtual rows 0..n-1 are the
+// text prefix (identity) followed by the image grid in row-major order for
+// axis 0 or COLUMN-major order for axis 1 — so the "own grid line" of every
+// query is a contiguous virtual key range for both axes, and axis 1 needs
+// no tensor transposes (keys are fetched through the row map; each key row
+// is a contiguous 128 B line either way). Liveness: text keys are visible
+// to everything causal; image keys only within the same grid line, causal.
+// ---------------------------------------------------------------------------
+
+DEVFN int ax_phys(int vrow, int t, int logS, int axis) {
+  if (axis != 1 || vrow < t) return vrow;
+  const int i = vrow - t;
+  return t + ((i & ((1 << logS) - 1)) << logS) + (i >> logS);
+}
+
+DEVFN bool ax_ok(int vq, int vk, int t, int logS) {
+  if (vk < t) return vq < t ? vk <= vq : true;
+  if (vq < vk || vq < t) return false;
+  return ((vq - t) >> logS) == ((vk - t) >> logS);
+}
+
 // XCD-aware block mapping (guide T1, bijective m204 form): the runtime
 // round-robins flat block ids across the 8 XCDs, so consecutive ids land on
 // different L2s. Remapping gives each XCD a CONTIGUOUS range of the
@@ -108,7 +132,8 @@ void fa_fwd_d64_kernel(
     const bool* __restrict__ static_mask, // [nq, nk] or null
     const unsigned char* __restrict__ tile_map,  // [ceil(nq/64), ceil(nk/32)] or null
     int b, int h, int nq, int nk,
-    float scale, int causal, int out_bnhd) {
+    float scale, int causal, int out_bnhd,
+    int ax_t, int ax_logS, int ax_axis) {        // axial mode if ax_axis >= 0
 
   // KV tiles are 64 keys (2 map granules); staging is software-pipelined:
   // the next live tile's global loads are issued before this tile's MFMA
@@ -133,19 +158,21 @@ void fa_fwd_d64_kernel(
   const int qrow = q0 + wave * 16 + lq;   // this lane's global query row
   const int diag = nk - nq;               // causal offset (ref triu(j-i+1))
 
+  const bool axial = ax_axis >= 0;
   const short* qp = q + (long)bh * nq * FA_D;
   const short* kp = k + (long)bh * nk * FA_D;
   const short* vp = v + (long)bh * nk * FA_D;
 
   // Q fragments (B-operand of the swapped QK^T):
   // lane holds Q[lq][8*grp + e + 32*c], c = 0,1
+  const int qphys = axial ? ax_phys(qrow, ax_t, ax_logS, ax_axis) : qrow;
   bf16x8 qfrag[2];
   {
     const bool qok = qrow < nq;
     #pragma unroll
     for (int c = 0; c < 2; ++c) {
       qfrag[c] = qok
-          ? *reinterpret_cast<const bf16x8*>(qp + (long)qrow * FA_D + 8 * grp + 32 * c)
+          ? *reinterpret_cast<const bf16x8*>(qp + (long)qphys * FA_D + 8 * grp + 32 * c)
           : bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
     }
   }
@@ -166,8 +193,17 @@ void fa_fwd_d64_kernel(
 
   // block-sparse skip: the host precomputes, per (64q, 32k) granule,
   // whether any static-mask entry is set; a 64-key tile is live if either
-  // of its granules is (axial/conv/block-sparse patterns)
+  // of its granules is (axial/conv/block-sparse patterns). In axial mode
+  // liveness is pure arithmetic: text tiles always, image tiles only when
+  // they overlap the q-block's own grid-line span.
   auto tile_live = [&](int t) -> bool {
+    if (axial) {
+      const int kbase = t * KV;
+      if (kbase < ax_t) return true;
+      if (q0 + FA_QBLK <= ax_t) return false;
+      const int l0 = (max(q0, ax_t) - ax_t) >> ax_logS;
+      return kbase + KV > ax_t + (l0 << ax_logS);
+    }
     if (!tmap_row) return true;
     const int g0 = 2 * t;
     bool live = tmap_row[g0] != 0;
@@ -202,8 +238,9 @@ void fa_fwd_d64_kernel(
     for (int half = 0; half < 2; ++half) {
       const int kg = kbase + srow0 + 32 * half;
       if (kg < nk) {
-        kreg[half] = *reinterpret_cast<const int4v*>(kp + (long)kg * FA_D + sc8);
-        vreg[half] = *reinterpret_cast<const int4v*>(vp + (long)kg * FA_D + sc8);
+        const int kph = axial ? ax_phys(kg, ax_t, ax_logS, ax_axis) : kg;
+        kreg[half] = *reinterpret_cast<const int4v*>(kp + (long)kph * FA_D + sc8);
+        vreg[half] = *reinterpret_cast<const int4v*>(vp + (long)kph * FA_D + sc8);
       } else {
         kreg[half] = int4v{0, 0, 0, 0};
         vreg[half] = int4v{0, 0, 0, 0};
@@ -272,14 +309,26 @@ void fa_fwd_d64_kernel(
     // s16[i] = S[qrow][kbase + (i>>2)*16 + grp*4 + (i&3)].
 
     // ---- scale + masks (uniform fast path for fully-unmasked interior)
-    const bool interior =
-        (kbase + KV <= nk) && (q0 + FA_QBLK <= nq) &&
-        (!causal || (kbase + KV - 1 <= q0 + diag)) &&
-        key_mask == nullptr &&
-        (static_mask == nullptr || (tmap_row != nullptr && tile_full(kt)));
+    const bool interior = key_mask == nullptr &&
+        (q0 + FA_QBLK <= nq) && (kbase + KV <= nk) &&
+        (axial
+             ? ((q0 >= ax_t && kbase + KV <= ax_t) ||        // img q x text k
+                (q0 + FA_QBLK <= ax_t && kbase + KV - 1 <= q0))  // text causal
+             : ((!causal || (kbase + KV - 1 <= q0 + diag)) &&
+                (static_mask == nullptr ||
+                 (tmap_row != nullptr && tile_full(kt)))));
     if (interior) {
       #pragma unroll
       for (int i = 0; i < 16; ++i) s16[i] *= scale;
+    } else if (axial) {
+      #pragma unroll
+      for (int i = 0; i < 16; ++i) {
+        const int kg = kbase + (i >> 2) * 16 + grp * 4 + (i & 3);
+        bool ok = (kg < nk) & (qrow < nq) && ax_ok(qrow, kg, ax_t, ax_logS);
+        if (key_mask != nullptr && ok)
+          ok &= key_mask[(long)batch * nk + ax_phys(kg, ax_t, ax_logS, ax_axis)];
+        s16[i] = ok ? s16[i] * scale : NEG_INF;
+      }
     } else {
       #pragma unroll
       for (int i = 0; i < 16; ++i) {
@@ -370,9 +419,10 @@ void fa_fwd_d64_kernel(
     const float l_r = __shfl(l_run, grp * 4 + r);
     const float inv = l_r > 0.f ? 1.f / l_r : 0.f;
     if (qr < nq) {
+      const int qrp = axial ? ax_phys(qr, ax_t, ax_logS, ax_axis) : qr;
       const long base = out_bnhd
-          ? (((long)batch_i * nq + qr) * h + head_i) * FA_D
-          : ((long)bh * nq + qr) * FA_D;
+          ? (((long)batch_i * nq + qrp) * h + head_i) * FA_D
+          : ((long)bh * nq + qrp) * FA_D;
       #pragma unroll
       for (int nt = 0; nt < 4; ++nt) {
         out[base + 16 * nt + lq] = f2bf(acc[nt][r] * inv);
@@ -380,7 +430,7 @@ void fa_fwd_d64_kernel(
     }
   }
   if (grp == 0 && qrow < nq) {
-    lse[(long)bh * nq + qrow] = (l_run > 0.f) ? m_run + __logf(l_run) : NEG_INF;
+    lse[(long)bh * nq + qphys] = (l_run > 0.f) ? m_run + __logf(l_run) : NEG_INF;
   }
 }
 
@@ -410,7 +460,8 @@ void fa_bwd_dq_kernel(
     const bool* __restrict__ key_mask,
     const bool* __restrict__ static_mask,
     const unsigned char* __restrict__ tile_map,   // [nq/64, nk/32]
-    int b, int h, int nq, int nk, float scale, int causal, int do_bnhd) {
+    int b, int h, int nq, int nk, float scale, int causal, int do_bnhd,
+    int ax_t, int ax_logS, int ax_axis) {
 
   constexpr int KV = 2 * FA_KBLK;          // 64 keys per LDS tile
   __shared__ short Kt[KV][KPAD];           // K row-major
@@ -438,20 +489,22 @@ void fa_bwd_dq_kernel(
       ? dout + ((long)batch * nq * h + (bh - batch * h)) * FA_D
       : dout + (long)bh * nq * FA_D;
 
+  const bool axial = ax_axis >= 0;
+  const int qphys = axial ? ax_phys(qrow, ax_t, ax_logS, ax_axis) : qrow;
   bf16x8 qfrag[2], dofrag[2];
   float lse_q = 0.f, D_q = 0.f;
   {
     const bool qok = qrow < nq;
     #pragma unroll
     for (int c = 0; c < 2; ++c) {
-      qfrag[c] = qok ? *reinterpret_cast<const bf16x8*>(qp + (long)qrow * FA_D + 8 * grp + 32 * c)
+      qfrag[c] = qok ? *reinterpret_cast<const bf16x8*>(qp + (long)qphys * FA_D + 8 * grp + 32 * c)
                      : bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
-      dofrag[c] = qok ? *reinterpret_cast<const bf16x8*>(dop + (long)qrow * do_stride + 8 * grp + 32 * c)
+      dofrag[c] = qok ? *reinterpret_cast<const bf16x8*>(dop + (long)qphys * do_stride + 8 * grp + 32 * c)
                       : bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
     }
     if (qok) {
-      lse_q = lse[(long)bh * nq + qrow];
-      D_q = Dv[(long)bh * nq + qrow];
+      lse_q = lse[(long)bh * nq + qphys];
+      D_q = Dv[(long)bh * nq + qphys];
     }
   }
 
@@ -468,6 +521,13 @@ void fa_bwd_dq_kernel(
       tile_map ? tile_map + (long)qtile * ntk : nullptr;
 
   auto tile_live = [&](int t) -> bool {
+    if (axial) {
+      const int kbase = t * KV;
+      if (kbase < ax_t) return true;
+      if (q0 + FA_QBLK <= ax_t) return false;
+      const int l0 = (max(q0, ax_t) - ax_t) >> ax_logS;
+      return kbase + KV > ax_t + (l0 << ax_logS);
+    }
     if (!tmap_row) return true;
     const int g0 = 2 * t;
     bool live = tmap_row[g0] != 0;
@@ -495,8 +555,9 @@ void fa_bwd_dq_kernel(
     for (int half = 0; half < 2; ++half) {
       const int kg = t * KV + srow0 + 32 * half;
       if (kg < nk) {
-        kreg[half] = *reinterpret_cast<const int4v*>(kp + (long)kg * FA_D + sc8);
-        vreg[half] = *reinterpret_cast<const int4v*>(vp + (long)kg * FA_D + sc8);
+        const int kph = axial ? ax_phys(kg, ax_t, ax_logS, ax_axis) : kg;
+        kreg[half] = *reinterpret_cast<const int4v*>(kp + (long)kph * FA_D + sc8);
+        vreg[half] = *reinterpret_cast<const int4v*>(vp + (long)kph * FA_D + sc8);
       } else {
         kreg[half] = int4v{0, 0, 0, 0};
         vreg[half] = int4v{0, 0, 0, 0};
@@ -563,16 +624,29 @@ void fa_bwd_dq_kernel(
     }
     __builtin_amdgcn_s_setprio(0);
 
-    const bool interior =
+    const bool interior = key_mask == nullptr &&
         (kbase + KV <= nk) && (q0 + FA_QBLK <= nq) &&
-        (!causal || (kbase + KV - 1 <= q0 + diag)) &&
-        key_mask == nullptr &&
-        (static_mask == nullptr || (tmap_row != nullptr && tile_full(kt)));
+        (axial
+             ? ((q0 >= ax_t && kbase + KV <= ax_t) ||
+                (q0 + FA_QBLK <= ax_t && kbase + KV - 1 <= q0))
+             : ((!causal || (kbase + KV - 1 <= q0 + diag)) &&
+                (static_mask == nullptr ||
+                 (tmap_row != nullptr && tile_full(kt)))));
     float ds16[16];
     if (interior) {
       #pragma unroll
       for (int i = 0; i < 16; ++i) {
         const float p = __expf(s16[i] * scale - lse_q);
+        ds16[i] = p * (dp16[i] - D_q) * scale;
+      }
+    } else if (axial) {
+      #pragma unroll
+      for (int i = 0; i < 16; ++i) {
+        const int kg = kbase + (i >> 2) * 16 + grp * 4 + (i & 3);
+        bool ok = (kg < nk) & (qrow < nq) && ax_ok(qrow, kg, ax_t, ax_logS);
+        if (key_mask != nullptr && ok)
+          ok &= key_mask[(long)batch * nk + ax_phys(kg, ax_t, ax_logS, ax_axis)];
+        const float p = ok ? __expf(s16[i] * scale - lse_q) : 0.f;
         ds16[i] = p * (dp16[i] - D_q) * scale;
       }
     } else {
@@ -616,9 +690,10 @@ void fa_bwd_dq_kernel(
   for (int r = 0; r < 4; ++r) {
     const int qr = q0 + wave * 16 + grp * 4 + r;
     if (qr < nq) {
+      const int qrp = axial ? ax_phys(qr, ax_t, ax_logS, ax_axis) : qr;
       #pragma unroll
       for (int nt = 0; nt < 4; ++nt) {
-        dq[((long)bh * nq + qr) * FA_D + 16 * nt + lq] = f2bf(acc[nt][r]);
+        dq[((long)bh * nq + qrp) * FA_D + 16 * nt + lq] = f2bf(acc[nt][r]);
       }
     }
   }
@@ -637,7 +712,8 @@ void fa_bwd_dkv_kernel(
     const bool* __restrict__ key_mask,
     const bool* __restrict__ static_mask,
     const unsigned char* __restrict__ tile_map_t,  // [nk/64, nq/32]
-    int b, int h, int nq, int nk, float scale, int causal, int do_bnhd) {
+    int b, int h, int nq, int nk, float scale, int causal, int do_bnhd,
+    int ax_t, int ax_logS, int ax_axis) {
 
   constexpr int KV = 2 * FA_KBLK;          // 64 q rows per LDS tile
   __shared__ short Qr[KV][KPAD];           // Q rows (B-operand of s^T)
@@ -667,14 +743,16 @@ void fa_bwd_dkv_kernel(
       ? dout + ((long)batch * nq * h + (bh - batch * h)) * FA_D
       : dout + (long)bh * nq * FA_D;
 
+  const bool axial = ax_axis >= 0;
+  const int kphys = axial ? ax_phys(krow, ax_t, ax_logS, ax_axis) : krow;
   bf16x8 kfrag[2], vfrag[2];
   {
     const bool kok = krow < nk;
     #pragma unroll
     for (int c = 0; c < 2; ++c) {
-      kfrag[c] = kok ? *reinterpret_cast<const bf16x8*>(kp + (long)krow * FA_D + 8 * grp + 32 * c)
+      kfrag[c] = kok ? *reinterpret_cast<const bf16x8*>(kp + (long)kphys * FA_D + 8 * grp + 32 * c)
                      : bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
-      vfrag[c] = kok ? *reinterpret_cast<const bf16x8*>(vp + (long)krow * FA_D + 8 * grp + 32 * c)
+      vfrag[c] = kok ? *reinterpret_cast<const bf16x8*>(vp + (long)kphys * FA_D + 8 * grp + 32 * c)
                      : bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
     }
   }
@@ -695,6 +773,12 @@ void fa_bwd_dkv_kernel(
       tile_map_t ? tile_map_t + (long)ktile * nqg : nullptr;
 
   auto tile_live = [&](int t) -> bool {
+    if (axial) {
+      if (k0 < ax_t) return true;           // text keys: every causal q tile
+      // image keys [k0, k0+64): queries live only inside the keys' lines
+      const int lk1 = (min(k0 + FA_QBLK, nk) - 1 - ax_t) >> ax_logS;
+      return t * KV < ax_t + ((lk1 + 1) << ax_logS);
+    }
     if (!tmap_row) return true;
     const int g0 = 2 * t;
     bool live = tmap_row[g0] != 0;
@@ -722,8 +806,9 @@ void fa_bwd_dkv_kernel(
     for (int half = 0; half < 2; ++half) {
       const int qg = t * KV + srow0 + 32 * half;
       if (qg < nq) {
-        qreg[half] = *reinterpret_cast<const int4v*>(qp + (long)qg * FA_D + sc8);
-        doreg[half] = *reinterpret_cast<const int4v*>(dop + (long)qg * do_stride + sc8);
+        const int qph = axial ? ax_phys(qg, ax_t, ax_logS, ax_axis) : qg;
+        qreg[half] = *reinterpret_cast<const int4v*>(qp + (long)qph * FA_D + sc8);
+        doreg[half] = *reinterpret_cast<const int4v*>(dop + (long)qph * do_stride + sc8);
       } else {
         qreg[half] = int4v{0, 0, 0, 0};
         doreg[half] = int4v{0, 0, 0, 0};
@@ -795,22 +880,50 @@ void fa_bwd_dkv_kernel(
     }
     __builtin_amdgcn_s_setprio(0);
 
-    const bool interior =
+    const bool interior = key_mask == nullptr &&
         (qbase + KV <= nq) && (k0 + FA_QBLK <= nk) &&
-        (!causal || (k0 + FA_QBLK - 1 <= qbase + diag)) &&
-        key_mask == nullptr &&
-        (static_mask == nullptr || (tmap_row != nullptr && tile_full(qt)));
+        (axial
+             // text-key blocks under the causal bound; image-key blocks are
+             // never interior (t=257 misaligns lines against 64-key blocks)
+             ? (k0 + FA_QBLK <= ax_t && k0 + FA_QBLK - 1 <= qbase)
+             : ((!causal || (k0 + FA_QBLK - 1 <= qbase + diag)) &&
+                (static_mask == nullptr ||
+                 (tmap_row != nullptr && tile_full(qt)))));
     if (interior) {
       #pragma unroll
       for (int mt = 0; mt < 4; ++mt) {
         const int qg = qbase + mt * 16 + lq;
-        const float l = lse[(long)bh * nq + qg];
-        const float Dq = Dv[(long)bh * nq + qg];
+        const int qph = axial ? ax_phys(qg, ax_t, ax_logS, ax_axis) : qg;
+        const float l = lse[(long)bh * nq + qph];
+        const float Dq = Dv[(long)bh * nq + qph];
         #pragma unroll
         for (int r = 0; r < 4; ++r) {
           const float p = __expf(st4[mt][r] * scale - l);
           const float ds = p * (dpt4[mt][r] - Dq) * scale;
           const int cc = (mt * 16 + lq) ^ (grp << 3);   // swz by row>>2
+          Pt[wave][grp * 4 + r][cc] = f2bf(p);
+          DSt[wave][grp * 4 + r][cc] = f2bf(ds);
+        }
+      }
+    } else if (axial) {
+      #pragma unroll
+      for (int mt = 0; mt < 4; ++mt) {
+        const int qg = qbase + mt * 16 + lq;
+        const int qph = (qg < nq) ? ax_phys(qg, ax_t, ax_logS, ax_axis) : 0;
+        const float l = (qg < nq) ? lse[(long)bh * nq + qph] : 0.f;
+        const float Dq = (qg < nq) ? Dv[(long)bh * nq + qph] : 0.f;
+        #pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const int key = k0 + wave * 16 + grp * 4 + r;
+          bool ok = (key < nk) & (qg < nq) && ax_ok(qg, key, ax_t, ax_logS);
+          if (key_mask != nullptr && ok)
+            ok &= key_mask[(long)batch * nk + ax_phys(key, ax_t, ax_logS, ax_axis)];
+          float p = 0.f, ds = 0.f;
+          if (ok) {
+            p = __expf(st4[mt][r] * scale - l);
+            ds = p * (dpt4[mt][r] - Dq) * scale;
+          }
+          const int cc = (mt * 16 + lq) ^ (grp << 3);
           Pt[wave][grp * 4 + r][cc] = f2bf(p);
           DSt[wave][grp * 4 + r][cc] = f2bf(ds);
         }
@@ -868,10 +981,11 @@ void fa_bwd_dkv_kernel(
   for (int r = 0; r < 4; ++r) {
     const int kr = k0 + wave * 16 + grp * 4 + r;
     if (kr < nk) {
+      const int krp = axial ? ax_phys(kr, ax_t, ax_logS, ax_axis) : kr;
       #pragma unroll
       for (int nt = 0; nt < 4; ++nt) {
-        dk[((long)bh * nk + kr) * FA_D + 16 * nt + lq] = f2bf(acc_dk[nt][r]);
-        dv[((long)bh * nk + kr) * FA_D + 16 * nt + lq] = f2bf(acc_dv[nt][r]);
+        dk[((long)bh * nk + krp) * FA_D + 16 * nt + lq] = f2bf(acc_dk[nt][r]);
+        dv[((long)bh * nk + krp) * FA_D + 16 * nt + lq] = f2bf(acc_dv[nt][r]);
       }
     }
   }
@@ -1651,17 +1765,29 @@ static hipStream_t cur_stream() {
   return at::hip::getCurrentHIPStream().stream();
 }
 
+static int ax_log2(int64_t S) {
+  int logS = 0;
+  while ((1 << logS) < S) ++logS;
+  TORCH_CHECK((1 << logS) == S, "axial image_size must be a power of 2");
+  return logS;
+}
+
 std::vector<torch::Tensor> fa_fwd(torch::Tensor q, torch::Tensor k, torch::Tensor v,
                                   double scale, bool causal,
                                   std::optional<torch::Tensor> key_mask,
                                   std::optional<torch::Tensor> static_mask,
                                   std::optional<torch::Tensor> tile_map,
-                                  bool out_bnhd) {
+                                  bool out_bnhd,
+                                  int64_t ax_t, int64_t ax_S, int64_t ax_axis) {
   CHK(q.is_cuda() && k.is_cuda() && v.is_cuda());
   CHK(q.dtype() == torch::kBFloat16);
   CHK(q.size(-1) == FA_D);
   CHK(q.is_contiguous() && k.is_contiguous() && v.is_contiguous());
   const int b = q.size(0), h = q.size(1), nq = q.size(2), nk = k.size(2);
+  const int ax_logS = ax_axis >= 0 ? ax_log2(ax_S) : 0;
+  if (ax_axis >= 0) {
+    CHK(causal && nq == nk && !static_mask.has_value() && !tile_map.has_value());
+  }
 
   auto out = out_bnhd ? torch::empty({b, nq, h, FA_D}, q.options())
                       : torch::empty_like(q);
@@ -1695,7 +1821,8 @@ std::vector<torch::Tensor> fa_fwd(torch::Tensor q, torch::Tensor k, torch::Tenso
                      reinterpret_cast<short*>(out.data_ptr()),
                      lse.data_ptr<float>(), km, sm, tm,
                      b, h, nq, nk, (float)scale, causal ? 1 : 0,
-                     out_bnhd ? 1 : 0);
+                     out_bnhd ? 1 : 0,
+                     (int)ax_t, ax_logS, (int)ax_axis);
   return {out, lse};
 }
 
@@ -1708,11 +1835,13 @@ std::vector<torch::Tensor> fa_bwd(torch::Tensor q, torch::Tensor k, torch::Tenso
                                   std::optional<torch::Tensor> tile_map,
                                   std::optional<torch::Tensor> tile_map_t,
                                   bool out_bnhd,
-                                  std::optional<torch::Tensor> grad_lse) {
+                                  std::optional<torch::Tensor> grad_lse,
+                                  int64_t ax_t, int64_t ax_S, int64_t ax_axis) {
   CHK(q.is_cuda() && q.dtype() == torch::kBFloat16);
   CHK(q.is_contiguous() && k.is_contiguous() && v.is_contiguous());
   CHK(out.is_contiguous() && dout.is_contiguous());
   const int b = q.size(0), h = q.size(1), nq = q.size(2), nk = k.size(2);
+  const int ax_logS = ax_axis >= 0 ? ax_log2(ax_S) : 0;
 
   auto Dv = torch::empty({b, h, nq}, q.options().dtype(torch::kFloat32));
   {
@@ -1752,7 +1881,7 @@ std::vector<torch::Tensor> fa_bwd(torch::Tensor q, torch::Tensor k, torch::Tenso
                      lse.data_ptr<float>(), Dv.data_ptr<float>(),
                      reinterpret_cast<short*>(dq.data_ptr()),
                      km, sm, tm, b, h, nq, nk, (float)scale, causal ? 1 : 0,
-                     out_bnhd ? 1 : 0);
+                     out_bnhd ? 1 : 0, (int)ax_t, ax_logS, (int)ax_axis);
   dim3 grid_k(((nk + FA_QBLK - 1) / FA_QBLK) * b * h);
   hipLaunchKernelGGL(fa_bwd_dkv_kernel, grid_k, dim3(256), 0, cur_stream(),
                      reinterpret_cast<const short*>(q.data_ptr()),
@@ -1763,7 +1892,7 @@ std::vector<torch::Tensor> fa_bwd(torch::Tensor q, torch::Tensor k, torch::Tenso
                      reinterpret_cast<short*>(dk.data_ptr()),
                      reinterpret_cast<short*>(dv.data_ptr()),
                      km, sm, tmt, b, h, nq, nk, (float)scale, causal ? 1 : 0,
-                     out_bnhd ? 1 : 0);
+                     out_bnhd ? 1 : 0, (int)ax_t, ax_logS, (int)ax_axis);
   return {dq, dk, dv};
 }
 
