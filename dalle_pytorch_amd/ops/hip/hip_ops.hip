@@ -2180,8 +2180,18 @@ std::vector<torch::Tensor> ln_bwd(torch::Tensor x, torch::Tensor dy,
 }
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
-  m.def("fa_fwd", &fa_fwd, "flash attention forward (gfx950, d=64)");
-  m.def("fa_bwd", &fa_bwd, "flash attention backward (gfx950, d=64)");
+  m.def("fa_fwd", &fa_fwd, "flash attention forward (gfx950, d=64)",
+        py::arg("q"), py::arg("k"), py::arg("v"), py::arg("scale"),
+        py::arg("causal"), py::arg("key_mask"), py::arg("static_mask"),
+        py::arg("tile_map"), py::arg("out_bnhd"),
+        py::arg("ax_t") = 0, py::arg("ax_S") = 0, py::arg("ax_axis") = -1);
+  m.def("fa_bwd", &fa_bwd, "flash attention backward (gfx950, d=64)",
+        py::arg("q"), py::arg("k"), py::arg("v"), py::arg("out"),
+        py::arg("lse"), py::arg("dout"), py::arg("scale"), py::arg("causal"),
+        py::arg("key_mask"), py::arg("static_mask"), py::arg("tile_map"),
+        py::arg("tile_map_t"), py::arg("out_bnhd"),
+        py::arg("grad_lse") = std::nullopt,
+        py::arg("ax_t") = 0, py::arg("ax_S") = 0, py::arg("ax_axis") = -1);
   m.def("rope_split_fwd", &rope_split_fwd,
         "fused qkv split + rotary (q,k,v all rotated)");
   m.def("rope_split_bwd", &rope_split_bwd, "rope_split backward");
