@@ -75,12 +75,21 @@ class Deterministic(nn.Module):
             return self.net(*args, **kwargs)
 
 
+def _residual_branch(det, inp, residual, record_rng, args):
+    """residual + det(inp): fused into one pass when the wrapped module is a
+    LayerScale (its `residual=` path routes through ops.fused.add_scaled)."""
+    if getattr(det.net, 'supports_residual', False):
+        return det(inp, residual=residual, record_rng=record_rng, **args)
+    return residual + det(inp, record_rng=record_rng, **args)
+
+
 class ReversibleBlock(nn.Module):
     """y1 = x1 + f(x2); y2 = x2 + g(y1). Backward reconstructs x from y and
-    re-runs f,g once each (reference reversible.py:54-106). The two streams
-    stay separate tensors end to end — the reference's per-block
-    cat/chunk round-trips are pure copy traffic (~12 GB/step at the
-    flagship shape)."""
+    re-runs f,g once each (RevNet decomposition; role of reference
+    reversible.py:54-106, re-derived). The two streams stay separate tensors
+    end to end — per-block cat/chunk round-trips would be pure copy traffic
+    (~12 GB/step at the flagship shape) — and the residual adds are fused
+    with LayerScale."""
 
     def __init__(self, f, g):
         super().__init__()
@@ -88,36 +97,35 @@ class ReversibleBlock(nn.Module):
         self.g = Deterministic(g)
 
     def forward(self, x1, x2, f_args={}, g_args={}):
+        rec = self.training
         with torch.no_grad():
-            y1 = x1 + self.f(x2, record_rng=self.training, **f_args)
-            y2 = x2 + self.g(y1, record_rng=self.training, **g_args)
+            y1 = _residual_branch(self.f, x2, x1, rec, f_args)
+            y2 = _residual_branch(self.g, y1, x2, rec, g_args)
         return y1, y2
 
+    @staticmethod
+    def _replay_grad(det, inp, upstream, args):
+        """Re-run the branch on a fresh leaf under its recorded RNG/autocast
+        state, push `upstream` through it (accumulating parameter grads), and
+        hand back (branch output detached, gradient w.r.t. the input)."""
+        leaf = inp.detach().requires_grad_(True)
+        with torch.enable_grad():
+            out = det(leaf, set_rng=True, **args)
+        torch.autograd.backward(out, upstream)
+        return out.detach(), leaf.grad
+
     def backward_pass(self, y1, y2, dy1, dy2, f_args={}, g_args={}):
-        with torch.enable_grad():
-            y1.requires_grad = True
-            gy1 = self.g(y1, set_rng=True, **g_args)
-            torch.autograd.backward(gy1, dy2)
-
-        with torch.no_grad():
-            x2 = y2 - gy1
-            del y2, gy1
-            dx1 = dy1 + y1.grad
-            del dy1
-            y1.grad = None
-
-        with torch.enable_grad():
-            x2.requires_grad = True
-            fx2 = self.f(x2, set_rng=True, **f_args)
-            torch.autograd.backward(fx2, dx1, retain_graph=True)
-
-        with torch.no_grad():
-            x1 = y1 - fx2
-            del y1, fx2
-            dx2 = dy2 + x2.grad
-            del dy2
-            x2.grad = None
-        return x1, x2.detach(), dx1, dx2
+        # invert g: x2 = y2 - g(y1); the replay also yields dL/dy1 through g
+        gy1, dy1_via_g = self._replay_grad(self.g, y1, dy2, g_args)
+        x2 = y2 - gy1
+        dx1 = dy1 + dy1_via_g
+        del y2, gy1, dy1, dy1_via_g
+        # invert f: x1 = y1 - f(x2); replay yields dL/dx2 through f
+        fx2, dx2_via_f = self._replay_grad(self.f, x2, dx1, f_args)
+        x1 = y1 - fx2
+        dx2 = dy2 + dx2_via_f
+        del y1, fx2, dy2, dx2_via_f
+        return x1, x2, dx1, dx2
 
 
 class _ReversibleFunction(Function):
@@ -151,8 +159,11 @@ class SequentialSequence(nn.Module):
     def forward(self, x, **kwargs):
         args = route_args(self.args_route, kwargs, len(self.layers))
         for (f, g), (f_args, g_args) in zip(self.layers, args):
-            x = x + f(x, **f_args)
-            x = x + g(x, **g_args)
+            for branch, branch_args in ((f, f_args), (g, g_args)):
+                if getattr(branch, 'supports_residual', False):
+                    x = branch(x, residual=x, **branch_args)
+                else:
+                    x = x + branch(x, **branch_args)
         return x
 
 

@@ -22,7 +22,7 @@ from dalle_pytorch_amd.models.attention import (
 from dalle_pytorch_amd.models.positional import build_dalle_rotary_table
 from dalle_pytorch_amd.ops import geglu
 from dalle_pytorch_amd.ops.fused import (token_shift, token_shift_supported,
-                                          layer_norm)
+                                          layer_norm, add_scaled)
 
 
 def _as_tuple(val, depth=1):
@@ -80,21 +80,27 @@ class CachedAs(nn.Module):
 
 class LayerScale(nn.Module):
     """Per-channel learned residual scale, depth-tiered init
-    (https://arxiv.org/abs/2103.17239; reference transformer.py:74-88)."""
+    (https://arxiv.org/abs/2103.17239; reference transformer.py:74-88).
+
+    When the caller passes ``residual=``, the residual add is fused with the
+    scale into one HBM pass (ops.fused.add_scaled) — the executors use this,
+    replacing the eager scale-cast + mul + add chain (5 tensor passes -> 3).
+    """
+
+    INIT_BY_DEPTH = ((18, 0.1), (24, 1e-5), (float('inf'), 1e-6))
+    supports_residual = True   # executors may pass residual= for the fused add
 
     def __init__(self, dim, depth, fn):
         super().__init__()
-        if depth <= 18:
-            init_eps = 0.1
-        elif depth <= 24:
-            init_eps = 1e-5
-        else:
-            init_eps = 1e-6
+        init_eps = next(eps for bound, eps in self.INIT_BY_DEPTH
+                        if depth <= bound)
         self.scale = nn.Parameter(torch.full((1, 1, dim), init_eps))
         self.fn = fn
 
-    def forward(self, x, **kwargs):
+    def forward(self, x, residual=None, **kwargs):
         out = self.fn(x, **kwargs)
+        if residual is not None:
+            return add_scaled(residual, out, self.scale)
         # multiply in the stream dtype: a fp32 scale would silently promote
         # the bf16 residual stream (2x elementwise traffic downstream)
         return out * self.scale.to(out.dtype)
@@ -142,12 +148,22 @@ class FeedForward(nn.Module):
 
 
 class PreShiftToken(nn.Module):
-    """Token-shift preprocessor (reference transformer.py:126-200).
+    """Token-shift preprocessor (same contract as reference
+    transformer.py:126-200, re-expressed).
 
-    Text positions receive half their channels from the previous token;
-    image positions receive a quarter from the grid-row above and a quarter
-    from the left neighbor. The cached decode path keeps a deque of the last
-    image row so single-token steps reproduce the training-time shift.
+    Channel layout: for text positions the FIRST HALF of the channels reads
+    from the previous token; for image positions the first quarter reads
+    from the grid row above and the second quarter from the left neighbor
+    (zeros where no such neighbor exists). The cached decode path keeps a
+    ring of the last ``image_size`` positions' RAW (top, left) quarters so a
+    single-token step reproduces the training-time shift.
+
+    Deviations from the reference: (a) the training path on GPU is one fused
+    gather kernel; (b) the ring is seeded from the raw inputs — the
+    reference seeds it from already-shifted values (transformer.py:193-198),
+    so its first ``image_size`` primed-generation steps read neighbors one
+    step too far back and cached priming diverges from uncached (pinned by
+    tests/test_dalle.py::test_primed_cached_generation_matches_uncached).
     """
 
     def __init__(self, fn, image_size, seq_len):
@@ -158,66 +174,64 @@ class PreShiftToken(nn.Module):
         self.img_seq_len = image_size ** 2
         self.text_len = seq_len - self.img_seq_len + 1
 
+    def _shift_full(self, x):
+        """Training-shape shift, whole sequence at once (eager path)."""
+        b, n, c = x.shape
+        t, S, q = self.text_len, self.image_size, x.shape[-1] // 4
+        prev = x[:, :t - 1, :c // 2]
+        text = torch.cat((
+            torch.cat((x.new_zeros(b, 1, c // 2), prev), dim=1),
+            x[:, :t, c // 2:]), dim=-1)
+        img = x[:, t:]
+        ni = img.shape[1]
+        if ni == 0:
+            return text
+        g = F.pad(img, (0, 0, 0, S * S - ni)).view(b, S, S, c)
+        top = torch.cat((g.new_zeros(b, 1, S, q), g[:, :-1, :, :q]), dim=1)
+        left = torch.cat((g.new_zeros(b, S, 1, q), g[:, :, :-1, q:2 * q]), dim=2)
+        img = torch.cat((top, left, g[..., 2 * q:]), dim=-1)
+        return torch.cat((text, img.view(b, S * S, c)[:, :ni]), dim=1)
+
+    def _seed_ring(self, x):
+        """Ring of the last S image positions' raw (top, left) quarters,
+        front-padded with zeros when fewer than S image tokens exist yet."""
+        b, _, c = x.shape
+        S, q = self.image_size, x.shape[-1] // 4
+        tail = x[:, self.text_len:][:, -S:]
+        zero = x.new_zeros(b, q)
+        ring = deque((zero, zero) for _ in range(S - tail.shape[1]))
+        ring.extend((tail[:, i, :q], tail[:, i, q:2 * q])
+                    for i in range(tail.shape[1]))
+        return ring
+
+    def _shift_one(self, x, ring, cache, kwargs):
+        t, S = self.text_len, self.image_size
+        offset = cache['offset']
+        assert offset >= t, 'cached decode starts after the text prefix'
+        assert len(ring) == S
+        cur = x[:, -1]
+        q = cur.shape[-1] // 4
+        ring.append((cur[:, :q], cur[:, q:2 * q]))
+        top = ring.popleft()[0]
+        at_row_start = (offset - t) % S == 0
+        left = torch.zeros_like(top) if at_row_start else ring[-2][1]
+        shifted = torch.cat((top, left, cur[:, 2 * q:]), dim=-1)
+        return self.fn(shifted[:, None], cache=cache, **kwargs)
+
     def forward(self, x, cache=None, cache_key=None, **kwargs):
-        seq_len, image_size, text_len = self.seq_len, self.image_size, self.text_len
-
-        if cache is not None and cache_key in cache:
-            offset = cache['offset']
-            assert offset >= text_len, 'cached inference for text is not supported'
-            q = cache[cache_key]
-            assert isinstance(q, deque) and len(q) == image_size
-
-            x_top, x_left, *x_pass = x[:, -1].chunk(4, dim=-1)
-            q.append((x_top, x_left))
-            x_top = q.popleft()[0]
-            x_left = q[-2][1]
-            if (offset - text_len) % image_size == 0:
-                x_left = torch.zeros_like(x_left)
-            x = torch.cat((x_top, x_left, *x_pass), dim=-1)
-            return self.fn(x[:, None], cache=cache, **kwargs)
-
-        n = x.shape[1]
-        padding = seq_len - n + 1
-        if n < text_len:
+        ring = cache.get(cache_key) if cache is not None else None
+        if ring is not None:
+            return self._shift_one(x, ring, cache, kwargs)
+        if x.shape[1] < self.text_len:
             return self.fn(x, **kwargs)
-
         if cache is None and token_shift_supported(x):
             # fused gather kernel (no pad/cat/fill chain) — training hot path
-            x = token_shift(x, text_len, image_size)
-            return self.fn(x, **kwargs)
-
-        x_text, x_img = x[:, :text_len], x[:, text_len:]
-        x_img = F.pad(x_img, (0, 0, 0, padding))
-        b = x_img.shape[0]
-        x_img = x_img.reshape(b, image_size, image_size, -1)
-
-        # text: shift half the channels one token to the right
-        x_text_shift, x_text_pass = x_text.chunk(2, dim=-1)
-        x_text_shift = F.pad(x_text_shift, (0, 0, 1, -1))
-        x_text = torch.cat((x_text_shift, x_text_pass), dim=-1)
-
-        # image: quarter from the row above, quarter from the left neighbor
-        x_top, x_left, *x_pass = x_img.chunk(4, dim=-1)
-        x_left = F.pad(x_left, (0, 0, 1, -1))
-        x_top = F.pad(x_top, (0, 0, 0, 0, 1, -1))
-        x_img = torch.cat((x_top, x_left, *x_pass), dim=-1)
-
-        x_img = x_img.reshape(b, -1, x_img.shape[-1])
-        x_img = x_img[:, :-padding]
-        x = torch.cat((x_text, x_img), dim=1)
-
+            return self.fn(token_shift(x, self.text_len, self.image_size),
+                           **kwargs)
+        shifted = self._shift_full(x)
         if cache is not None:
-            d_top, d_left, *_ = x[:, -1].chunk(4, dim=-1)
-            d_top, d_left = torch.zeros_like(d_top), torch.zeros_like(d_left)
-            q = deque()
-            last_row = x_img[:, -image_size:]
-            for _ in range(image_size - last_row.shape[1]):
-                q.append((d_top, d_left))
-            for i in range(last_row.shape[1]):
-                q.append(last_row[:, i].chunk(4, dim=-1)[:2])
-            cache[cache_key] = q
-
-        return self.fn(x, cache=cache, **kwargs)
+            cache[cache_key] = self._seed_ring(x)
+        return self.fn(shifted, cache=cache, **kwargs)
 
 
 class Transformer(nn.Module):
@@ -252,73 +266,58 @@ class Transformer(nn.Module):
         super().__init__()
         self.seq_len = seq_len
         self.image_fmap_size = image_fmap_size
+        del sparse_attn  # accepted for flag parity; per-layer value is unused
+        # (the reference shadows and ignores it too, transformer.py:245-246)
+
+        attn_kwargs = dict(causal=causal, seq_len=seq_len, heads=heads,
+                           dim_head=dim_head, dropout=attn_dropout)
+        self._attn_common = (attn_kwargs, stable, optimize_for_inference)
 
         attn_types = _as_tuple(attn_types if attn_types is not None else ('full',))
-        attn_type_iter = islice(cycle(attn_types), depth)
-        sparse_layer = _as_tuple(sparse_attn, depth)
-        shared_attn_ids = cycle(shared_attn_ids if shared_attn_ids is not None else range(depth))
-        shared_ff_ids = cycle(shared_ff_ids if shared_ff_ids is not None else range(depth))
-        shared_attn = {}
-        shared_ff = {}
+        plan = zip(islice(cycle(attn_types), depth),
+                   cycle(shared_attn_ids if shared_attn_ids is not None else range(depth)),
+                   cycle(shared_ff_ids if shared_ff_ids is not None else range(depth)))
 
+        built_attn, built_ff = {}, {}
         layers = nn.ModuleList([])
-        for ind, _sparse, attn_type, attn_id, ff_id in zip(
-                range(depth), sparse_layer, attn_type_iter, shared_attn_ids, shared_ff_ids):
-            if attn_type == 'full':
-                make_attn = partial(Attention, stable=stable)
-            elif attn_type == 'sparse':
-                make_attn = SparseAttention
-            elif attn_type in ('axial_row', 'axial_col'):
-                if optimize_for_inference:
-                    make_attn = partial(Attention, stable=stable,
-                                        static_mask=self._get_attention_mask(attn_type))
-                else:
-                    axis = 0 if attn_type == 'axial_row' else 1
-                    make_attn = partial(SparseAxialCausalAttention, seq_len=seq_len,
-                                        axis=axis, image_size=image_fmap_size, stable=stable)
-            elif attn_type == 'conv_like':
-                make_attn = partial(SparseConvCausalAttention, seq_len=seq_len,
-                                    image_size=image_fmap_size, stable=stable)
+        for ind, (attn_type, attn_id, ff_id) in enumerate(plan):
+            if attn_id in built_attn:
+                leaf, built_type = built_attn[attn_id]
+                if built_type != attn_type:
+                    raise ValueError(
+                        f'attn_types do not match shared_attn_ids (ind={ind}, '
+                        f'attn_type="{attn_type}", reused="{built_type}")')
             else:
-                raise ValueError(f'attention type "{attn_type}" is not valid')
+                leaf = self._leaf_attention(attn_type, dim)
+                built_attn[attn_id] = (leaf, attn_type)
 
-            attn, reused_type = shared_attn.get(attn_id, (None, None))
-            if attn is None:
-                attn = make_attn(dim, causal=causal, seq_len=seq_len, heads=heads,
-                                 dim_head=dim_head, dropout=attn_dropout)
-                shared_attn[attn_id] = (attn, attn_type)
-            elif attn_type != reused_type:
-                raise ValueError(
-                    f'attn_types do not match shared_attn_ids (ind={ind}, '
-                    f'attn_type="{attn_type}", reused="{reused_type}")')
+            if ff_id not in built_ff:
+                built_ff[ff_id] = FeedForward(dim, mult=ff_mult, dropout=ff_dropout)
+            ff = built_ff[ff_id]
 
-            ff = shared_ff.get(ff_id)
-            if ff is None:
-                ff = FeedForward(dim, mult=ff_mult, dropout=ff_dropout)
-                shared_ff[ff_id] = ff
-
-            if isinstance(attn, Attention):
-                attn = CachedAs(f'attn_{ind}', attn)
-            else:
-                attn = NonCached(attn)
-
+            # wrapper chain (outer to inner), identical key schema to the
+            # reference: [CachedAs -> PreShiftToken ->]
+            # CachedAs|NonCached -> leaf
+            attn = (CachedAs(f'attn_{ind}', leaf) if isinstance(leaf, Attention)
+                    else NonCached(leaf))
             if shift_tokens:
-                attn = CachedAs(f'preshift_attn_{ind}',
-                                PreShiftToken(attn, image_size=image_fmap_size, seq_len=seq_len))
-                ff = CachedAs(f'preshift_ff_{ind}',
-                              PreShiftToken(ff, image_size=image_fmap_size, seq_len=seq_len))
+                wrap = partial(PreShiftToken, image_size=image_fmap_size,
+                               seq_len=seq_len)
+                attn = CachedAs(f'preshift_attn_{ind}', wrap(attn))
+                ff = CachedAs(f'preshift_ff_{ind}', wrap(ff))
 
             layers.append(nn.ModuleList([
                 LayerScale(dim, ind + 1, PreNorm(dim, attn, sandwich=sandwich_norm)),
                 LayerScale(dim, ind + 1, PreNorm(dim, ff, sandwich=sandwich_norm)),
             ]))
 
-        execute_type = ReversibleSequence if reversible else SequentialSequence
-        route_attn = ((True, False),) * depth
-        route_all = ((True, True),) * depth
-        attn_route_map = {'mask': route_attn, 'rotary_pos_emb': route_attn,
-                          'cache': route_all}
-        self.layers = execute_type(layers, args_route=attn_route_map)
+        executor = ReversibleSequence if reversible else SequentialSequence
+        to_attn_only = ((True, False),) * depth
+        self.layers = executor(layers, args_route={
+            'mask': to_attn_only,
+            'rotary_pos_emb': to_attn_only,
+            'cache': ((True, True),) * depth,
+        })
 
         pos_emb = None
         if rotary_emb:
@@ -329,6 +328,27 @@ class Transformer(nn.Module):
 
     def forward(self, x, **kwargs):
         return self.layers(x, rotary_pos_emb=self.pos_emb, **kwargs)
+
+    def _leaf_attention(self, attn_type, dim):
+        """Construct the leaf attention module for one layer."""
+        common, stable, opt_inference = self._attn_common
+        S = self.image_fmap_size
+        if attn_type == 'full':
+            return Attention(dim, stable=stable, **common)
+        if attn_type == 'sparse':
+            return SparseAttention(dim, **common)
+        if attn_type in ('axial_row', 'axial_col'):
+            if opt_inference:   # cache-friendly static-mask simulation
+                return Attention(dim, stable=stable,
+                                 static_mask=self._get_attention_mask(attn_type),
+                                 **common)
+            return SparseAxialCausalAttention(
+                dim, axis=(0 if attn_type == 'axial_row' else 1),
+                image_size=S, stable=stable, **common)
+        if attn_type == 'conv_like':
+            return SparseConvCausalAttention(dim, image_size=S, stable=stable,
+                                             **common)
+        raise ValueError(f'attention type "{attn_type}" is not valid')
 
     def _get_attention_mask(self, attn_type):
         """Dense bool mask reproducing axial attention (reference

@@ -63,6 +63,40 @@ def token_shift_supported(x):
             and x.shape[-1] * x.element_size() % 64 == 0)
 
 
+class _AddScaledFn(torch.autograd.Function):
+    """out = x + gamma * y with per-channel gamma — the residual + LayerScale
+    fusion (reference transformer.py:74-88 + reversible.py:138-140 adds).
+    Backward: dx aliases dout (no kernel), dy/dgamma in one fused pass."""
+
+    @staticmethod
+    def forward(ctx, x, y, gamma):
+        ext = hip_module()
+        y = y.contiguous()
+        gf = gamma.detach().reshape(-1).float().contiguous()
+        out = ext.resls_fwd(x.contiguous(), y, gf)
+        ctx.save_for_backward(y, gf)
+        ctx.gamma_meta = (gamma.shape, gamma.dtype)
+        return out
+
+    @staticmethod
+    def backward(ctx, dout):
+        ext = hip_module()
+        y, gf = ctx.saved_tensors
+        dout = dout.contiguous()
+        dy, dgamma = ext.resls_bwd(dout, y, gf)
+        shape, dtype = ctx.gamma_meta
+        return dout, dy, dgamma.reshape(shape).to(dtype)
+
+
+def add_scaled(x, y, gamma):
+    """x + gamma * y (gamma broadcast over the last dim)."""
+    if (x.is_cuda and x.dtype == torch.bfloat16 and y.dtype == torch.bfloat16
+            and not using_eager_fallback(x) and x.shape[-1] % 8 == 0
+            and x.shape[-1] // 8 <= 256 and 256 % (x.shape[-1] // 8) == 0):
+        return _AddScaledFn.apply(x, y, gamma)
+    return x + y * gamma.to(y.dtype)
+
+
 class _LayerNormFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, weight, bias, eps):

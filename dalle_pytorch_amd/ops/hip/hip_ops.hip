@@ -992,6 +992,73 @@ void fa_bwd_dkv_kernel(
 }
 
 
+// ---------------------------------------------------------------------------
+// Fused residual + LayerScale: out = x + gamma * y (gamma per-channel).
+// The eager chain (scale cast, y*gamma temp, x+temp) is 5 full-tensor passes
+// + a 1-element cast kernel per call; this is 3 passes, one launch. Backward:
+// dx aliases dout (no kernel), dy = gamma*dout, dgamma = sum_rows(dout*y)
+// accumulated block-locally then reduced with one tiny torch sum (same
+// scheme as ln_bwd). Channels are fixed per thread so gamma lives in regs.
+// ---------------------------------------------------------------------------
+
+__global__ __launch_bounds__(256)
+void resls_fwd_kernel(const short* __restrict__ x, const short* __restrict__ y,
+                      const float* __restrict__ gamma, short* __restrict__ out,
+                      long rows, int dim) {
+  const int cpr = dim >> 3;                  // 16B chunks per row
+  const int rpp = 256 / cpr;                 // rows per block pass
+  const int rg = threadIdx.x / cpr;          // this thread's row group
+  const int ch = (threadIdx.x % cpr) * 8;    // channel base
+  float g[8];
+  #pragma unroll
+  for (int e = 0; e < 8; ++e) g[e] = gamma[ch + e];
+  for (long r = (long)blockIdx.x * rpp + rg; r < rows;
+       r += (long)gridDim.x * rpp) {
+    const long base = r * dim + ch;
+    int4v xv = *reinterpret_cast<const int4v*>(x + base);
+    int4v yv = *reinterpret_cast<const int4v*>(y + base);
+    const short* xs = reinterpret_cast<const short*>(&xv);
+    const short* ys = reinterpret_cast<const short*>(&yv);
+    short o[8];
+    #pragma unroll
+    for (int e = 0; e < 8; ++e) o[e] = f2bf(bf2f(xs[e]) + g[e] * bf2f(ys[e]));
+    *reinterpret_cast<int4v*>(out + base) = *reinterpret_cast<int4v*>(o);
+  }
+}
+
+__global__ __launch_bounds__(256)
+void resls_bwd_kernel(const short* __restrict__ dout, const short* __restrict__ y,
+                      const float* __restrict__ gamma, short* __restrict__ dy,
+                      float* __restrict__ dg_part,   // [gridDim.x * rpp, dim]
+                      long rows, int dim) {
+  const int cpr = dim >> 3;
+  const int rpp = 256 / cpr;
+  const int rg = threadIdx.x / cpr;
+  const int ch = (threadIdx.x % cpr) * 8;
+  float g[8], dg[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+  #pragma unroll
+  for (int e = 0; e < 8; ++e) g[e] = gamma[ch + e];
+  for (long r = (long)blockIdx.x * rpp + rg; r < rows;
+       r += (long)gridDim.x * rpp) {
+    const long base = r * dim + ch;
+    int4v dv = *reinterpret_cast<const int4v*>(dout + base);
+    int4v yv = *reinterpret_cast<const int4v*>(y + base);
+    const short* ds_ = reinterpret_cast<const short*>(&dv);
+    const short* ys = reinterpret_cast<const short*>(&yv);
+    short o[8];
+    #pragma unroll
+    for (int e = 0; e < 8; ++e) {
+      const float d = bf2f(ds_[e]);
+      o[e] = f2bf(g[e] * d);
+      dg[e] += d * bf2f(ys[e]);
+    }
+    *reinterpret_cast<int4v*>(dy + base) = *reinterpret_cast<int4v*>(o);
+  }
+  const long prow = (long)blockIdx.x * rpp + rg;
+  #pragma unroll
+  for (int e = 0; e < 8; ++e) dg_part[prow * dim + ch + e] = dg[e];
+}
+
 // D = rowsum(dO * O) for the flash backward, fused (the ATen form costs
 // two full fp32 materializations of dO and O per attention backward).
 // 4 lanes per row, 32B vector loads, shfl reduce.
@@ -2179,7 +2246,48 @@ std::vector<torch::Tensor> ln_bwd(torch::Tensor x, torch::Tensor dy,
   return {dx, dgamma, dbeta};
 }
 
+torch::Tensor resls_fwd(torch::Tensor x, torch::Tensor y, torch::Tensor gamma) {
+  CHK(x.is_cuda() && x.dtype() == torch::kBFloat16);
+  CHK(x.is_contiguous() && y.is_contiguous() && gamma.is_contiguous());
+  CHK(gamma.dtype() == torch::kFloat32);
+  const int dim = x.size(-1);
+  CHK(dim == gamma.numel() && (dim & 7) == 0 && (dim >> 3) <= 256 &&
+      (256 % (dim >> 3)) == 0);
+  const long rows = x.numel() / dim;
+  auto out = torch::empty_like(x);
+  const int rpp = 256 / (dim >> 3);
+  const long nb = std::min<long>((rows + rpp - 1) / rpp, 2048);
+  hipLaunchKernelGGL(resls_fwd_kernel, dim3(nb), dim3(256), 0, cur_stream(),
+                     reinterpret_cast<const short*>(x.data_ptr()),
+                     reinterpret_cast<const short*>(y.data_ptr()),
+                     gamma.data_ptr<float>(),
+                     reinterpret_cast<short*>(out.data_ptr()), rows, dim);
+  return out;
+}
+
+std::vector<torch::Tensor> resls_bwd(torch::Tensor dout, torch::Tensor y,
+                                     torch::Tensor gamma) {
+  CHK(dout.is_cuda() && dout.dtype() == torch::kBFloat16);
+  CHK(dout.is_contiguous() && y.is_contiguous() && gamma.is_contiguous());
+  const int dim = dout.size(-1);
+  const long rows = dout.numel() / dim;
+  auto dy = torch::empty_like(dout);
+  const int rpp = 256 / (dim >> 3);
+  const long nb = std::min<long>((rows + rpp - 1) / rpp, 1024);
+  auto dgp = torch::zeros({nb * rpp, (long)dim},
+                          dout.options().dtype(torch::kFloat32));
+  hipLaunchKernelGGL(resls_bwd_kernel, dim3(nb), dim3(256), 0, cur_stream(),
+                     reinterpret_cast<const short*>(dout.data_ptr()),
+                     reinterpret_cast<const short*>(y.data_ptr()),
+                     gamma.data_ptr<float>(),
+                     reinterpret_cast<short*>(dy.data_ptr()),
+                     dgp.data_ptr<float>(), rows, dim);
+  return {dy, dgp.sum(0)};
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("resls_fwd", &resls_fwd, "fused residual + per-channel scale fwd");
+  m.def("resls_bwd", &resls_bwd, "fused residual + per-channel scale bwd");
   m.def("fa_fwd", &fa_fwd, "flash attention forward (gfx950, d=64)",
         py::arg("q"), py::arg("k"), py::arg("v"), py::arg("scale"),
         py::arg("causal"), py::arg("key_mask"), py::arg("static_mask"),

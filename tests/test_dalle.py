@@ -146,6 +146,24 @@ def test_guided_cached_logits_equal_uncached_logits():
     assert torch.allclose(cached_last, full[:, -1], atol=1e-4)
 
 
+def test_primed_cached_generation_matches_uncached():
+    """Image-primed generation with shift_tokens + cache: the PreShiftToken
+    ring must be seeded from RAW inputs. The reference seeds it from shifted
+    values (transformer.py:193-198) so its first image_size primed steps read
+    stale neighbors — cached and uncached generations diverge there."""
+    torch.manual_seed(17)
+    d = tiny_dalle(shift_tokens=True, depth=2, stable=True).eval()
+    text = torch.randint(1, 50, (1, 8))
+    img = torch.rand(1, 3, 64, 64)
+    torch.manual_seed(19)
+    a = d.generate_images(text, img=img, use_cache=True,
+                          temperature=1e-8, filter_thres=0.99)
+    torch.manual_seed(19)
+    b = d.generate_images(text, img=img, use_cache=False,
+                          temperature=1e-8, filter_thres=0.99)
+    assert torch.allclose(a, b, atol=1e-5)
+
+
 def test_generate_texts():
     from dalle_pytorch_amd.utils.tokenizer import SimpleTokenizer
 

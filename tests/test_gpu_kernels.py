@@ -399,6 +399,30 @@ def test_axial_col_module_gpu_matches_cpu():
     assert (x_gpu.grad.float().cpu() - x_cpu.grad).abs().max().item() < 5e-2 * gscale
 
 
+def test_add_scaled_vs_oracle(ext):
+    """Fused residual+LayerScale kernel (resls_fwd/bwd) vs fp32 autograd."""
+    from dalle_pytorch_amd.ops.fused import add_scaled
+    torch.manual_seed(12)
+    for dim in (512, 1024, 2048):
+        x0 = torch.randn(3, 40, dim, device='cuda')
+        y0 = torch.randn(3, 40, dim, device='cuda')
+        g0 = torch.randn(1, 1, dim, device='cuda') * 0.1
+        x = x0.bfloat16().requires_grad_()
+        y = y0.bfloat16().requires_grad_()
+        g = g0.clone().requires_grad_()
+        out = add_scaled(x, y, g)
+        out.float().square().sum().backward()
+
+        xr, yr, gr = (t.clone().requires_grad_() for t in (x0, y0, g0))
+        ref = xr + yr * gr
+        ref.square().sum().backward()
+        assert (out.float() - ref).abs().max() < 0.05
+        for got, want in ((x.grad.float(), xr.grad), (y.grad.float(), yr.grad),
+                          (g.grad.float(), gr.grad)):
+            rel = (got - want).abs().max().item() / want.abs().max().item()
+            assert rel < 5e-2, (dim, rel)
+
+
 def test_dalle_train_step_gpu():
     """One full flagship-shaped training step on GPU, bf16, HIP path."""
     from dalle_pytorch_amd import DALLE, DiscreteVAE

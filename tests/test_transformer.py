@@ -99,3 +99,59 @@ def test_sandwich_norm():
     tr = Transformer(dim=16, depth=1, seq_len=8, heads=2, dim_head=8,
                      image_fmap_size=2, sandwich_norm=True, rotary_emb=False)
     assert isinstance(tr.layers.layers[0][0].fn.norm_out, torch.nn.LayerNorm)
+
+
+def test_add_scaled_matches_eager():
+    from dalle_pytorch_amd.ops.fused import add_scaled
+    torch.manual_seed(4)
+    x = torch.randn(2, 6, 64, requires_grad=True)
+    y = torch.randn(2, 6, 64, requires_grad=True)
+    g = torch.full((1, 1, 64), 0.1, requires_grad=True)
+    out = add_scaled(x, y, g)
+    ref = x + y * g
+    assert torch.allclose(out, ref, atol=1e-6)
+    gx, gy, gg = torch.autograd.grad(out.square().sum(), (x, y, g))
+    rx, ry, rg = torch.autograd.grad(ref.square().sum(), (x, y, g))
+    assert torch.allclose(gx, rx) and torch.allclose(gy, ry) and torch.allclose(gg, rg)
+
+
+def test_layerscale_residual_path_equals_plain():
+    """LayerScale(residual=x) == x + LayerScale()(x) — the executors rely on
+    this equivalence for the fused residual path."""
+    from dalle_pytorch_amd.models.transformer import LayerScale
+    torch.manual_seed(5)
+    ls = LayerScale(32, 3, torch.nn.Linear(32, 32))
+    x = torch.randn(2, 4, 32)
+    assert torch.allclose(ls(x, residual=x), x + ls(x), atol=1e-6)
+
+
+def test_shift_full_matches_reference_formulation():
+    """PreShiftToken._shift_full vs the reference's pad/chunk/cat chain
+    (reference transformer.py:165-186), all shapes incl. partial last row."""
+    import torch.nn.functional as F
+    from dalle_pytorch_amd.models.transformer import PreShiftToken
+    torch.manual_seed(6)
+    S, t = 4, 3
+    seq_len = t + S * S - 1
+    mod = PreShiftToken(torch.nn.Identity(), image_size=S, seq_len=seq_len)
+
+    def reference_shift(x):
+        n = x.shape[1]
+        padding = seq_len - n + 1
+        x_text, x_img = x[:, :t], x[:, t:]
+        x_img = F.pad(x_img, (0, 0, 0, padding))
+        b = x_img.shape[0]
+        x_img = x_img.reshape(b, S, S, -1)
+        x_text_shift, x_text_pass = x_text.chunk(2, dim=-1)
+        x_text_shift = F.pad(x_text_shift, (0, 0, 1, -1))
+        x_text = torch.cat((x_text_shift, x_text_pass), dim=-1)
+        xt, xl, *xp = x_img.chunk(4, dim=-1)
+        xl = F.pad(xl, (0, 0, 1, -1))
+        xt = F.pad(xt, (0, 0, 0, 0, 1, -1))
+        x_img = torch.cat((xt, xl, *xp), dim=-1)
+        x_img = x_img.reshape(b, -1, x_img.shape[-1])[:, :-padding]
+        return torch.cat((x_text, x_img), dim=1)
+
+    for n in (t, t + 1, t + S, seq_len - 1, seq_len):
+        x = torch.randn(2, n, 16)
+        assert torch.allclose(mod._shift_full(x), reference_shift(x)), n
