@@ -44,7 +44,7 @@ def _check_broadcast(rank, world):
 def _check_grad_allreduce(rank, world):
     torch.manual_seed(0)
     model = _model(seed=0)
-    engine = DataParallelEngine(model, bucket_bytes=1 << 10)
+    engine = DataParallelEngine(model, bucket_bytes=1 << 10, wire_dtype=None)
     torch.manual_seed(100 + rank)
     x = torch.randn(4, 8)
     model(x).sum().backward()
@@ -67,7 +67,7 @@ def _check_grad_allreduce(rank, world):
 
 def _check_no_sync_accumulation(rank, world):
     model = _model(seed=0)
-    engine = DataParallelEngine(model, bucket_bytes=1 << 10)
+    engine = DataParallelEngine(model, bucket_bytes=1 << 10, wire_dtype=None)
     torch.manual_seed(200 + rank)
     x1, x2 = torch.randn(4, 8), torch.randn(4, 8)
     with engine.no_sync():
@@ -96,7 +96,7 @@ def _check_average_scalar(rank, world):
 def _check_distributed_clip(rank, world):
     """Clip after all-reduce == torch clip on the averaged-grad oracle."""
     model = _model(seed=0)
-    engine = DataParallelEngine(model, bucket_bytes=1 << 10)
+    engine = DataParallelEngine(model, bucket_bytes=1 << 10, wire_dtype=None)
     torch.manual_seed(300 + rank)
     x = torch.randn(4, 8)
     (model(x) ** 2).sum().backward()
@@ -120,12 +120,85 @@ def _check_distributed_clip(rank, world):
         assert torch.allclose(p.grad, q.grad, rtol=1e-5, atol=1e-8)
 
 
+def _check_bf16_wire_allreduce(rank, world):
+    """Default bf16-wire buckets: halved payload; grads
+    match the fp32 oracle within one bf16 rounding step."""
+    model = _model(seed=0)
+    engine = DataParallelEngine(model, bucket_bytes=1 << 10)   # default wire
+    assert engine.wire_dtype == torch.bfloat16
+    torch.manual_seed(100 + rank)
+    x = torch.randn(4, 8)
+    model(x).sum().backward()
+    engine.finish_gradient_sync()
+
+    acc = [torch.zeros_like(p) for p in model.parameters()]
+    for r in range(world):
+        m = _model(seed=0)
+        torch.manual_seed(100 + r)
+        xr = torch.randn(4, 8)
+        m(xr).sum().backward()
+        for a, p in zip(acc, m.parameters()):
+            a += p.grad / world
+    for p, a in zip(model.parameters(), acc):
+        scale = a.abs().max().clamp(min=1e-6)
+        assert ((p.grad - a).abs().max() / scale) < 2e-2
+
+
+def _check_convergence_equivalence(rank, world):
+    """DP loss curve == single-process full-batch loss curve:
+    each rank trains on its half of a fixed global batch; with fp32 wire the
+    curves match to float tolerance, with bf16 wire to bf16 tolerance."""
+    torch.manual_seed(42)
+    data = torch.randn(8, 8)
+    target = torch.randn(8, 1)
+
+    def run(world_slice=None, engine_kw=None, steps=6):
+        model = _model(seed=0)
+        engine = DataParallelEngine(model, bucket_bytes=1 << 10,
+                                    **(engine_kw or {}))
+        opt = torch.optim.Adam(model.parameters(), lr=1e-2)
+        losses = []
+        for _ in range(steps):
+            x, y = (data, target) if world_slice is None else (
+                data.chunk(world, 0)[rank], target.chunk(world, 0)[rank])
+            loss = ((model(x) - y) ** 2).mean()
+            opt.zero_grad(set_to_none=False)
+            engine.zero_grad()
+            loss.backward()
+            engine.finish_gradient_sync()
+            opt.step()
+            losses.append(engine.average_all(loss.detach()).item())
+        return losses
+
+    dp = run(world_slice=True, engine_kw=dict(wire_dtype=None))
+    dp_bf16 = run(world_slice=True)
+
+    # single-process oracle computed identically on every rank
+    model = _model(seed=0)
+    opt = torch.optim.Adam(model.parameters(), lr=1e-2)
+    ref = []
+    for _ in range(6):
+        loss = ((model(data) - target) ** 2).mean()
+        opt.zero_grad()
+        loss.backward()
+        opt.step()
+        ref.append(loss.item())
+
+    for a, b in zip(dp, ref):
+        assert abs(a - b) < 1e-5, (dp, ref)
+    for a, b in zip(dp_bf16, ref):
+        assert abs(a - b) < 5e-2, (dp_bf16, ref)
+    assert dp[-1] < dp[0]   # it actually trains
+
+
 @pytest.mark.parametrize('fn,port', [
     ('_check_broadcast', 29611),
     ('_check_grad_allreduce', 29612),
     ('_check_no_sync_accumulation', 29613),
     ('_check_average_scalar', 29614),
     ('_check_distributed_clip', 29615),
+    ('_check_bf16_wire_allreduce', 29616),
+    ('_check_convergence_equivalence', 29617),
 ])
 def test_distributed_gloo(fn, port):
     _spawn(fn, world=2, port=port)
@@ -133,7 +206,7 @@ def test_distributed_gloo(fn, port):
 
 def test_single_process_noop():
     model = _model(seed=0)
-    engine = DataParallelEngine(model, bucket_bytes=1 << 10)
+    engine = DataParallelEngine(model, bucket_bytes=1 << 10, wire_dtype=None)
     x = torch.randn(4, 8)
     model(x).sum().backward()
     engine.finish_gradient_sync()
