@@ -399,6 +399,56 @@ def test_axial_col_module_gpu_matches_cpu():
     assert (x_gpu.grad.float().cpu() - x_cpu.grad).abs().max().item() < 5e-2 * gscale
 
 
+def _poison_allocator(mb=512):
+    """Fill a chunk of the caching allocator with NaNs so freshly 'empty'
+    kernel outputs start poisoned — un-written output elements then surface
+    as NaNs instead of silently reading stale zeros."""
+    t = torch.full((mb << 18,), float('nan'), device='cuda')  # mb MiB of fp32
+    del t
+    torch.cuda.synchronize()
+
+
+def test_fa_fwd_poisoned_output_fully_written(ext):
+    torch.manual_seed(13)
+    b, h, n = 2, 3, 193   # ragged: partial q tiles and k tiles
+    q = torch.randn(b, h, n, 64, device='cuda').bfloat16()
+    k = torch.randn(b, h, n, 64, device='cuda').bfloat16()
+    v = torch.randn(b, h, n, 64, device='cuda').bfloat16()
+    _poison_allocator()
+    out, lse = ext.fa_fwd(q, k, v, 0.125, True, None, None, None, False)
+    assert torch.isfinite(out.float()).all(), 'unwritten/NaN forward output'
+    assert torch.isfinite(lse).all()
+    _poison_allocator()
+    dout = torch.randn_like(out)
+    dq, dk, dv = ext.fa_bwd(q, k, v, out, lse, dout, 0.125, True,
+                            None, None, None, None, False, None)
+    for t_, name in ((dq, 'dq'), (dk, 'dk'), (dv, 'dv')):
+        assert torch.isfinite(t_.float()).all(), f'unwritten/NaN {name}'
+
+
+def test_fa_fully_masked_rows_no_nan(ext):
+    """A batch element whose key_mask masks EVERY key: forward must produce
+    zero output and -inf lse for it, backward must produce zero (not NaN)
+    gradients everywhere it touches."""
+    from dalle_pytorch_amd.ops import attention_core
+    torch.manual_seed(14)
+    b, h, n = 2, 2, 96
+    km = torch.ones(b, n, dtype=torch.bool, device='cuda')
+    km[1] = False   # batch 1: nothing attendable
+    q = torch.randn(b, h, n, 64, device='cuda').bfloat16().requires_grad_()
+    k = torch.randn(b, h, n, 64, device='cuda').bfloat16().requires_grad_()
+    v = torch.randn(b, h, n, 64, device='cuda').bfloat16().requires_grad_()
+    out, lse = attention_core(q, k, v, 0.125, causal=True, key_mask=km,
+                              return_lse=True)
+    assert (out[1].float() == 0).all()
+    assert torch.isinf(lse[1]).all() and (lse[1] < 0).all()
+    assert torch.isfinite(out[0].float()).all()
+    out.float().square().sum().backward()
+    for g in (q.grad, k.grad, v.grad):
+        assert torch.isfinite(g.float()).all(), 'NaN grads from masked rows'
+        assert (g[1].float() == 0).all(), 'masked batch leaked gradient'
+
+
 def test_add_scaled_vs_oracle(ext):
     """Fused residual+LayerScale kernel (resls_fwd/bwd) vs fp32 autograd."""
     from dalle_pytorch_amd.ops.fused import add_scaled
