@@ -133,10 +133,22 @@ class FastDecoder:
         # head weights, pre-cast once (the decode loop never re-casts)
         cast = lambda t: None if t is None else \
             t.detach().to(self.device, self.dtype)
+        castf = lambda t: None if t is None else \
+            t.detach().to(self.device, torch.float32)
         head_ln, head_lin = dalle.to_logits[0], dalle.to_logits[1]
+        ntt, nit = dalle.num_text_tokens, dalle.num_image_tokens
+        w_full = cast(head_lin.weight)
+        b_full = cast(head_lin.bias)
         self.head_w = {
             'ln_w': cast(head_ln.weight), 'ln_b': cast(head_ln.bias),
-            'w': cast(head_lin.weight), 'b': cast(head_lin.bias),
+            'w': w_full, 'b': b_full,
+            # decode steps are always image positions: the logits mask there
+            # allows exactly the image-vocab rows, so generation projects
+            # against this slice only (~7x less head GEMM + sampling work)
+            'w_img': w_full[ntt:ntt + nit].contiguous(),
+            'b_img': None if b_full is None else b_full[ntt:ntt + nit].contiguous(),
+            'b_img32': castf(None if head_lin.bias is None
+                             else head_lin.bias[ntt:ntt + nit]),
         }
 
         from dalle_pytorch_amd.ops.dispatch import hip_available
@@ -197,13 +209,29 @@ class FastDecoder:
             w['qkv'] = cast(st.leaf.to_qkv.weight)
             w['out_w'] = cast(st.leaf.to_out[0].weight)
             w['out_b'] = cast(st.leaf.to_out[0].bias)
+            w['out_b32'] = castf(st.leaf.to_out[0].bias)
         else:
             net = st.leaf.net
             w['ff1_w'] = cast(net[0].weight)
             w['ff1_b'] = cast(net[0].bias)
+            w['ff1_b32'] = castf(net[0].bias)
             w['ff2_w'] = cast(net[3].weight)
             w['ff2_b'] = cast(net[3].bias)
+            w['ff2_b32'] = castf(net[3].bias)
         return w
+
+    def _lin(self, x, w, bias32, bias):
+        """Decode-step linear: the skinny-M weights-streaming kernel when the
+        fused path is on (hipBLASLt runs M<=128 GEMMs at ~0.5 TB/s; the
+        kernel streams the weight at HBM rate), F.linear otherwise."""
+        rows = x.numel() // x.shape[-1]
+        if (self._fused_decode and rows <= 128 and x.shape[-1] % 32 == 0
+                and w.shape[0] % 4 == 0):
+            from dalle_pytorch_amd.ops.dispatch import hip_module
+            out = hip_module().skinny_gemm(
+                x.reshape(rows, x.shape[-1]), w, bias32)
+            return out.view(*x.shape[:-1], w.shape[0])
+        return F.linear(x, w, bias)
 
     # ----------------------------------------------------------- branches
 
@@ -224,7 +252,8 @@ class FastDecoder:
         offset 0; decode has n == 1)."""
         leaf = st.leaf
         h, d = leaf.heads, leaf.dim_head
-        qkv = F.linear(x, st.w['qkv'])
+        qkv = (self._lin(x, st.w['qkv'], None, None) if n == 1
+               else F.linear(x, st.w['qkv']))
         if n == 1 and self._fused_decode:
             from dalle_pytorch_amd.ops.dispatch import hip_module
             out = hip_module().fa_decode(
@@ -233,7 +262,7 @@ class FastDecoder:
                 self.sin if self.rotary else None,
                 offset_t, st.pattern, leaf.scale,
                 st.live, st.live_cnt).view(self.b, 1, h * d)
-            return F.linear(out, st.w['out_w'], st.w['out_b'])
+            return self._lin(out, st.w['out_w'], st.w['out_b32'], st.w['out_b'])
         q, k, v = (t.reshape(self.b, n, h, d).permute(0, 2, 1, 3)
                    for t in qkv.chunk(3, dim=-1))
         if self.rotary:
@@ -332,6 +361,11 @@ class FastDecoder:
                 self._shift_decode(st, y, offset_t)
         if st.is_attn:
             y = self._attn(st, y, offset_t, n)
+        elif n == 1:
+            from dalle_pytorch_amd.ops import geglu
+            y = self._lin(y, st.w['ff1_w'], st.w['ff1_b32'], st.w['ff1_b'])
+            y = geglu(y)
+            y = self._lin(y, st.w['ff2_w'], st.w['ff2_b32'], st.w['ff2_b'])
         else:
             from dalle_pytorch_amd.ops import geglu
             y = F.linear(y, st.w['ff1_w'], st.w['ff1_b'])
@@ -374,6 +408,18 @@ class FastDecoder:
         lm = d.logits_mask[0].to(self.device).index_select(0, position_mask_rows)
         return logits.masked_fill(lm.unsqueeze(0), -torch.finfo(logits.dtype).max)
 
+    def _head_img(self, x):
+        """Image-vocab-only head for decode steps: an image position's logits
+        mask allows exactly the image-vocab rows, so projecting against that
+        slice is equivalent and ~7x cheaper (plus top-k/gumbel shrink with it)."""
+        d = self.dalle
+        if d.stable:
+            x = x / x.amax(dim=-1, keepdim=True)
+        x = F.layer_norm(x, (x.shape[-1],), self.head_w['ln_w'],
+                         self.head_w['ln_b'], d.to_logits[0].eps)
+        return self._lin(x, self.head_w['w_img'], self.head_w['b_img32'],
+                         self.head_w['b_img'])
+
     # ----------------------------------------------------------- prefill
 
     @torch.no_grad()
@@ -405,7 +451,8 @@ class FastDecoder:
 
     def step(self, token):
         """One decode step: image-token ids [b] at position offset ->
-        masked logits [b, total_tokens]; advances the offset."""
+        logits [b, total_tokens] (or [b, num_image_tokens] in the generate
+        loop's image-head mode); advances the offset."""
         d = self.dalle
         off = self.offset_t
         emb = d.image_emb(token).unsqueeze(1)
@@ -415,33 +462,63 @@ class FastDecoder:
                 .reshape(1, -1, emb.shape[-1])
             emb = emb + full.index_select(1, g)
         x = self._run_stack(emb.to(self.dtype), off, 1)
-        logits = self._head(x, off)[:, 0]
+        if getattr(self, '_img_head', False):
+            logits = self._head_img(x)[:, 0]
+        else:
+            logits = self._head(x, off)[:, 0]
         self.offset_t += 1
         return logits
 
     # --------------------------------------------------------- generate
 
     @torch.no_grad()
-    def generate(self, text, filter_thres=0.9, temperature=1.0):
+    def generate(self, text, filter_thres=0.9, temperature=1.0, cond_scale=1.0):
+        """Batch image generation; with ``cond_scale != 1`` classifier-free
+        guidance runs the conditioned and null-conditioned streams as ONE
+        doubled batch (one static graph, one kernel sequence — the reference
+        pays a second full forward per step, dalle_pytorch.py:564-574).
+        Requires a decoder built with batch_size == 2 * text batch then."""
         d = self.dalle
-        assert text.shape[0] == self.b, \
-            f'decoder built for batch {self.b}, got {text.shape[0]}'
+        guided = cond_scale != 1
+        nb = text.shape[0]
+        expect = 2 * nb if guided else nb
+        assert expect == self.b, \
+            f'decoder built for batch {self.b}, got {nb} (guided={guided})'
         was_training = d.training
         d.eval()
         self._graph = None   # offsets differ per call; recapture
-        logits = self.prefill(text[:, :d.text_seq_len])
-        step_fn = self._graph_step if self.use_graph else self.step
-        out_tokens = []
-        for i in range(d.image_seq_len):
-            filtered = top_k(logits.float(), thres=filter_thres)
-            sample = gumbel_sample(filtered, temperature=temperature)
-            token = (sample - d.num_text_tokens).clamp(min=0)
-            out_tokens.append(token)
-            if i + 1 < d.image_seq_len:
-                logits = step_fn(token)
-        img_seq = torch.stack(out_tokens, dim=1)
-        images = d.vae.decode(img_seq)
-        d.train(was_training)
+        self._img_head = True
+        text = text[:, :d.text_seq_len]
+        if guided:
+            # null-conditioned half: zeroed text (forward's null_cond path
+            # maps zeros to the unique padding ids)
+            text = torch.cat((text, torch.zeros_like(text)), dim=0)
+        try:
+            ntt, nit = d.num_text_tokens, d.num_image_tokens
+            logits = self.prefill(text)[:, ntt:ntt + nit]
+            step_fn = self._graph_step if self.use_graph else self.step
+            # reference top-k semantics: k is a fraction of the FULL vocab
+            k = max(int((1 - filter_thres) * d.total_tokens), 1)
+            k = min(k, nit)
+            out_tokens = []
+            for i in range(d.image_seq_len):
+                li = logits.float()
+                if guided:
+                    cond, null = li[:nb], li[nb:]
+                    li = null + (cond - null) * cond_scale
+                vals, idx = li.topk(k, dim=-1)
+                filtered = torch.full_like(li, -torch.finfo(li.dtype).max)
+                filtered.scatter_(1, idx, vals)
+                token = gumbel_sample(filtered, temperature=temperature)
+                out_tokens.append(token)
+                if i + 1 < d.image_seq_len:
+                    feed = torch.cat((token, token), dim=0) if guided else token
+                    logits = step_fn(feed)
+            img_seq = torch.stack(out_tokens, dim=1)
+            images = d.vae.decode(img_seq)
+        finally:
+            self._img_head = False
+            d.train(was_training)
         return images
 
     def _snapshot(self):

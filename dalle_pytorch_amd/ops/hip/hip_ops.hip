@@ -993,6 +993,94 @@ void fa_bwd_dkv_kernel(
 
 
 // ---------------------------------------------------------------------------
+// Skinny-M GEMM: out[M, N] = x[M, K] @ W[N, K]^T (+ bias), M <= 128.
+//
+// The decode step's projections are M = batch (64..128) against multi-MB
+// weight matrices — pure weight-bandwidth problems that hipBLASLt runs at
+// ~520 GB/s (measured round 1, ~26% of generation). Here each 4-wave block
+// owns 64 rows of W and streams them once from HBM as MFMA A-fragments
+// (row-major 16 B loads); x is tiny (<=256 KB) and stays L2-resident, read
+// directly as B-fragments. K is split across blocks so every GEMM shape
+// fills the 256-CU chip; fp32 partials are reduced (+bias, bf16 cast) by a
+// trailing elementwise kernel.
+// ---------------------------------------------------------------------------
+
+__global__ __launch_bounds__(256)
+void skinny_gemm_kernel(
+    const short* __restrict__ x,    // [M, K] bf16
+    const short* __restrict__ w,    // [N, K] bf16 (torch Linear layout)
+    float* __restrict__ outf,       // [M, N] fp32, pre-zeroed
+    int M, int N, int K, int kslice) {
+  const int nt = blockIdx.x;        // n-tile (64 rows of W)
+  const int z = blockIdx.y;         // k-split index
+  const int wave = threadIdx.x >> 6, lane = threadIdx.x & 63;
+  const int lq = lane & 15, grp = lane >> 4;
+  const int n0 = nt * 64 + wave * 16;          // this wave's 16 W rows
+  const int k0 = z * kslice;
+  const int k1 = min(K, k0 + kslice);
+  const int mt_n = (M + 15) >> 4;
+  const bool single = gridDim.y == 1;
+
+  const short* wrow = w + (long)(n0 + lq) * K;  // A-frag: row = lane&15
+  f32x4 acc[8];
+  #pragma unroll
+  for (int i = 0; i < 8; ++i) acc[i] = f32x4{0, 0, 0, 0};
+
+  for (int k = k0; k < k1; k += 32) {
+    const bf16x8 af = (n0 + lq < N)
+        ? *reinterpret_cast<const bf16x8*>(wrow + k + 8 * grp)
+        : bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
+    for (int mt = 0; mt < mt_n; ++mt) {
+      const int m = mt * 16 + lq;
+      const bf16x8 xf = (m < M)
+          ? *reinterpret_cast<const bf16x8*>(x + (long)m * K + k + 8 * grp)
+          : bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
+      acc[mt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, xf, acc[mt], 0, 0, 0);
+    }
+  }
+
+  // C[row = n (grp*4+r)][col = m (lane&15)]; K-split partial sums combine
+  // through fp32 atomics that stay in L2 (the [M, N] buffer is MBs at most)
+  for (int mt = 0; mt < mt_n; ++mt) {
+    const int m = mt * 16 + lq;
+    if (m >= M) continue;
+    const int n = nt * 64 + wave * 16 + grp * 4;
+    if (single) {
+      if (n + 3 < N) {
+        *reinterpret_cast<f32x4*>(outf + (long)m * N + n) = acc[mt];
+      } else {
+        for (int r = 0; r < 4 && n + r < N; ++r)
+          outf[(long)m * N + n + r] = acc[mt][r];
+      }
+    } else {
+      for (int r = 0; r < 4 && n + r < N; ++r)
+        atomicAdd(outf + (long)m * N + n + r, acc[mt][r]);
+    }
+  }
+}
+
+__global__ __launch_bounds__(256)
+void skinny_cast_kernel(
+    const float* __restrict__ outf,   // [M, N] fp32
+    const float* __restrict__ bias,   // [N] or null
+    short* __restrict__ out,          // [M, N] bf16
+    long MN, int N) {
+  const long i0 = ((long)blockIdx.x * 256 + threadIdx.x) * 4;
+  if (i0 >= MN) return;
+  if (i0 + 3 >= MN) {
+    for (long i = i0; i < MN; ++i)
+      out[i] = f2bf(outf[i] + (bias ? bias[i % N] : 0.f));
+    return;
+  }
+  f32x4 v = *reinterpret_cast<const f32x4*>(outf + i0);
+  short o[4];
+  #pragma unroll
+  for (int e = 0; e < 4; ++e)
+    o[e] = f2bf(v[e] + (bias ? bias[(i0 + e) % N] : 0.f));
+  *reinterpret_cast<int2*>(out + i0) = *reinterpret_cast<int2*>(o);
+}
+
+// ---------------------------------------------------------------------------
 // Fused residual + LayerScale: out = x + gamma * y (gamma per-channel).
 // The eager chain (scale cast, y*gamma temp, x+temp) is 5 full-tensor passes
 // + a 1-element cast kernel per call; this is 3 passes, one launch. Backward:
@@ -1487,12 +1575,13 @@ void fa_decode_part_kernel(
 // (~288 of 1281 under the flagship axial patterns). Blocks iterate listed
 // keys only — every K/V row loaded is a live one.
 //
-// ONE wave per block, 64 listed keys per block: at ~288 live keys the
-// 4-wave form spent its time in barriers and cross-wave staging for tiny
-// per-phase work (59 us/dispatch measured vs ~5 us of live traffic).
-// Wave-synchronous execution needs no barriers — softmax stats reduce by
-// shuffle, the only LDS is the q/k/v broadcast and the 64-entry p row.
-__global__ __launch_bounds__(64)
+// FOUR heads per 256-thread block, one wave per head: at ~288 live keys a
+// one-wave-per-(head, part) launch is dispatch-rate bound (5120 blocks ~
+// 47.8 us/dispatch vs ~5 us of live traffic, profiled round 1) — packing 4
+// heads per block cuts the block count 4x. Waves stay independent (private
+// LDS slices, per-wave shuffle reductions); the single __syncthreads covers
+// the q/k/v broadcast.
+__global__ __launch_bounds__(256)
 void fa_decode_part_list_kernel(
     const short* __restrict__ qkv,    // [b, 3*h*64]
     short* __restrict__ kc,           // [b, h, N, 64]
@@ -1505,15 +1594,17 @@ void fa_decode_part_list_kernel(
     float* __restrict__ scratch,      // [b, h, KS, 66]
     int b, int h, int N, int rot, float scale, int KS, int chunk, int Lmax) {
 
-  __shared__ float qs[64], ksn[64], vsn[64];
-  __shared__ float Pl[64];
-  __shared__ int   Ki[64];
+  __shared__ float qs[4][64], ksn[4][64], vsn[4][64];
+  __shared__ float Pl[4][64];
+  __shared__ int   Ki[4][64];
 
-  const int head = blockIdx.x, bi = blockIdx.y, z = blockIdx.z;
-  const int lane = threadIdx.x;
+  const int wave = threadIdx.x >> 6, lane = threadIdx.x & 63;
+  const int head = blockIdx.x * 4 + wave;
+  const int bi = blockIdx.y, z = blockIdx.z;
   const long off = *offset;
+  const bool head_ok = head < h;
 
-  {
+  if (head_ok) {
     const int d = lane;
     const long base = (long)bi * 3 * h * 64 + (long)head * 64 + d;
     float qv = bf2f(qkv[base]);
@@ -1534,15 +1625,16 @@ void fa_decode_part_list_kernel(
       kc[cbase] = f2bf(kv);
       vc[cbase] = f2bf(vv);
     }
-    qs[d] = qv * scale;
-    ksn[d] = kv;
-    vsn[d] = vv;
+    qs[wave][d] = qv * scale;
+    ksn[wave][d] = kv;
+    vsn[wave][d] = vv;
   }
-  __syncthreads();   // single wave: compiles to a waitcnt, no barrier cost
+  __syncthreads();
+  if (!head_ok) return;
 
   const int cnt = live_cnt[off];
   const int j0 = z * 64;
-  const int jn = min(64, cnt - j0);             // live entries in this block
+  const int jn = min(64, cnt - j0);             // live entries in this part
   const int* lrow = live + off * (long)Lmax;
   const short* krow0 = kc + ((long)bi * h + head) * N * 64;
 
@@ -1554,7 +1646,7 @@ void fa_decode_part_list_kernel(
     float p = 0.f;
     if (key == (int)off) {
       #pragma unroll
-      for (int d = 0; d < 64; ++d) p += qs[d] * ksn[d];
+      for (int d = 0; d < 64; ++d) p += qs[wave][d] * ksn[wave][d];
     } else {
       const short* krow = krow0 + (long)key * 64;
       #pragma unroll
@@ -1562,7 +1654,7 @@ void fa_decode_part_list_kernel(
         int4v kk = *reinterpret_cast<const int4v*>(krow + c * 8);
         const short* ks = reinterpret_cast<const short*>(&kk);
         #pragma unroll
-        for (int e = 0; e < 8; ++e) p += qs[c * 8 + e] * bf2f(ks[e]);
+        for (int e = 0; e < 8; ++e) p += qs[wave][c * 8 + e] * bf2f(ks[e]);
       }
     }
     dot = p;
@@ -1577,18 +1669,18 @@ void fa_decode_part_list_kernel(
   #pragma unroll
   for (int s = 32; s > 0; s >>= 1) lsum += __shfl_xor(lsum, s);
 
-  Pl[lane] = p;
-  Ki[lane] = key;
-  __syncthreads();
+  Pl[wave][lane] = p;
+  Ki[wave][lane] = key;
+  // same-wave produce/consume: no cross-wave barrier needed
 
-  // P*V: lane = d, serial over this block's listed keys
+  // P*V: lane = d, serial over this part's listed keys
   const short* vrow0 = vc + ((long)bi * h + head) * N * 64;
   float acc = 0.f;
   for (int i = 0; i < jn; ++i) {
-    const float pi = Pl[i];
+    const float pi = Pl[wave][i];
     if (pi != 0.f) {
-      const int ki = Ki[i];
-      const float vv = (ki == (int)off) ? vsn[lane]
+      const int ki = Ki[wave][i];
+      const float vv = (ki == (int)off) ? vsn[wave][lane]
                                         : bf2f(vrow0[(long)ki * 64 + lane]);
       acc += pi * vv;
     }
@@ -1598,13 +1690,14 @@ void fa_decode_part_list_kernel(
   if (lane == 0) { sl[0] = m; sl[1] = lsum; }
 }
 
-__global__ __launch_bounds__(64)
+__global__ __launch_bounds__(256)
 void fa_decode_combine_kernel(
     const float* __restrict__ scratch,  // [b, h, KS, 66]
     short* __restrict__ out,            // [b, h*64]
     int b, int h, int KS) {
-  const int head = blockIdx.x, bi = blockIdx.y;
-  const int lane = threadIdx.x;
+  const int head = blockIdx.x * 4 + (threadIdx.x >> 6), bi = blockIdx.y;
+  const int lane = threadIdx.x & 63;
+  if (head >= h) return;
   const float* s0 = scratch + ((long)bi * h + head) * KS * 66;
   float m = NEG_INF;
   for (int z = 0; z < KS; ++z) m = fmaxf(m, s0[z * 66]);
@@ -2111,7 +2204,8 @@ torch::Tensor fa_decode(torch::Tensor qkv, torch::Tensor kc, torch::Tensor vc,
   if (live.has_value()) {
     CHK(live->dtype() == torch::kInt32 && live->is_contiguous());
     CHK(live_cnt.has_value() && live_cnt->dtype() == torch::kInt32);
-    hipLaunchKernelGGL(fa_decode_part_list_kernel, grid, dim3(64), 0,
+    dim3 grid4((h + 3) / 4, b, KS);
+    hipLaunchKernelGGL(fa_decode_part_list_kernel, grid4, dim3(256), 0,
                        cur_stream(),
                        reinterpret_cast<const short*>(qkv.data_ptr()),
                        reinterpret_cast<short*>(kc.data_ptr()),
@@ -2129,8 +2223,8 @@ torch::Tensor fa_decode(torch::Tensor qkv, torch::Tensor kc, torch::Tensor vc,
                      scratch.data_ptr<float>(),
                      b, h, N, rot, (float)scale, KS, chunk);
   }
-  hipLaunchKernelGGL(fa_decode_combine_kernel, dim3(h, b), dim3(64), 0,
-                     cur_stream(), scratch.data_ptr<float>(),
+  hipLaunchKernelGGL(fa_decode_combine_kernel, dim3((h + 3) / 4, b), dim3(256),
+                     0, cur_stream(), scratch.data_ptr<float>(),
                      reinterpret_cast<short*>(out.data_ptr()),
                      b, h, KS);
   return out;
@@ -2246,6 +2340,39 @@ std::vector<torch::Tensor> ln_bwd(torch::Tensor x, torch::Tensor dy,
   return {dx, dgamma, dbeta};
 }
 
+torch::Tensor skinny_gemm(torch::Tensor x, torch::Tensor w,
+                          std::optional<torch::Tensor> bias) {
+  CHK(x.is_cuda() && x.dtype() == torch::kBFloat16 && x.is_contiguous());
+  CHK(w.dtype() == torch::kBFloat16 && w.is_contiguous());
+  const int M = x.size(0), K = x.size(1), N = w.size(0);
+  CHK(w.size(1) == K && M <= 128 && (K & 31) == 0);
+  const float* bp = nullptr;
+  torch::Tensor bf32;
+  if (bias.has_value()) {
+    bf32 = bias->detach().to(torch::kFloat32).contiguous();
+    bp = bf32.data_ptr<float>();
+  }
+  const int ntiles = (N + 63) / 64;
+  int ksplit = std::min<int>({16, std::max(1, 512 / ntiles), (K + 63) / 64});
+  int kslice = ((K + ksplit - 1) / ksplit + 31) & ~31;
+  ksplit = (K + kslice - 1) / kslice;
+  auto outf = ksplit > 1
+      ? torch::zeros({(long)M, (long)N}, x.options().dtype(torch::kFloat32))
+      : torch::empty({(long)M, (long)N}, x.options().dtype(torch::kFloat32));
+  hipLaunchKernelGGL(skinny_gemm_kernel, dim3(ntiles, ksplit), dim3(256), 0,
+                     cur_stream(),
+                     reinterpret_cast<const short*>(x.data_ptr()),
+                     reinterpret_cast<const short*>(w.data_ptr()),
+                     outf.data_ptr<float>(), M, N, K, kslice);
+  auto out = torch::empty({(long)M, (long)N}, x.options());
+  const long MN = (long)M * N;
+  hipLaunchKernelGGL(skinny_cast_kernel,
+                     dim3((MN + 1023) / 1024), dim3(256), 0,
+                     cur_stream(), outf.data_ptr<float>(), bp,
+                     reinterpret_cast<short*>(out.data_ptr()), MN, N);
+  return out;
+}
+
 torch::Tensor resls_fwd(torch::Tensor x, torch::Tensor y, torch::Tensor gamma) {
   CHK(x.is_cuda() && x.dtype() == torch::kBFloat16);
   CHK(x.is_contiguous() && y.is_contiguous() && gamma.is_contiguous());
@@ -2288,6 +2415,9 @@ std::vector<torch::Tensor> resls_bwd(torch::Tensor dout, torch::Tensor y,
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("resls_fwd", &resls_fwd, "fused residual + per-channel scale fwd");
   m.def("resls_bwd", &resls_bwd, "fused residual + per-channel scale bwd");
+  m.def("skinny_gemm", &skinny_gemm,
+        "skinny-M weights-streaming GEMM (decode projections)",
+        py::arg("x"), py::arg("w"), py::arg("bias") = std::nullopt);
   m.def("fa_fwd", &fa_fwd, "flash attention forward (gfx950, d=64)",
         py::arg("q"), py::arg("k"), py::arg("v"), py::arg("scale"),
         py::arg("causal"), py::arg("key_mask"), py::arg("static_mask"),

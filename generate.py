@@ -95,14 +95,19 @@ def main(argv=None):
                                        truncate_text=True).to(device)
             text_tokens = text_tokens.repeat(args.num_images, 1)
 
-        use_fast = (not args.no_fast and not args.gentxt
-                    and args.cond_scale == 1.0)
+        # the fast decoder covers guided generation (cond_scale != 1) with a
+        # doubled-batch stream and gentxt (text completion happens above,
+        # image decode is the same); only explicit --no_fast falls back
+        guided = args.cond_scale != 1.0
+        use_fast = not args.no_fast
         decoder = None
         if use_fast:
             try:
                 from dalle_pytorch_amd.engine import FastDecoder
-                decoder = FastDecoder(dalle, batch_size=args.batch_size,
-                                      use_graph=device.type == 'cuda')
+                decoder = FastDecoder(
+                    dalle,
+                    batch_size=args.batch_size * (2 if guided else 1),
+                    use_graph=device.type == 'cuda')
             except (ValueError, AssertionError) as e:
                 print(f'fast decoder unavailable ({e}); using cached decode')
 
@@ -112,7 +117,7 @@ def main(argv=None):
             if decoder is not None and chunk.shape[0] == args.batch_size:
                 images.append(decoder.generate(
                     chunk, filter_thres=args.top_k,
-                    temperature=args.temperature))
+                    temperature=args.temperature, cond_scale=args.cond_scale))
             else:
                 images.append(dalle.generate_images(
                     chunk, filter_thres=args.top_k, temperature=args.temperature,

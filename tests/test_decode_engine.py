@@ -44,6 +44,26 @@ def test_engine_matches_model_generation(kw):
         (got - ref).abs().max().item()
 
 
+def test_engine_guided_matches_model_guided():
+    """cond_scale != 1 on the fast path: the doubled-batch (cond + null)
+    stream must reproduce the model's guided generation (which is itself
+    pinned cached == uncached by test_dalle.py)."""
+    torch.manual_seed(6)
+    d = tiny_dalle(attn_types=('full', 'axial_row')).eval()
+    text = torch.randint(1, 50, (2, 8))
+
+    torch.manual_seed(23)
+    ref = d.generate_images(text, use_cache=True, temperature=1e-8,
+                            filter_thres=0.99, cond_scale=2.0)
+    dec = FastDecoder(d, batch_size=4)   # 2*b for the guided stream pair
+    torch.manual_seed(23)
+    got = dec.generate(text, temperature=1e-8, filter_thres=0.99,
+                       cond_scale=2.0)
+    assert got.shape == ref.shape
+    assert torch.allclose(got, ref, atol=1e-4), \
+        (got - ref).abs().max().item()
+
+
 def test_engine_step_logits_match_model_cache():
     """Per-step logits comparison (tighter than end-to-end argmax parity)."""
     torch.manual_seed(4)
