@@ -166,6 +166,11 @@ class FastDecoder:
         self._g_logits = None
         if self._fused_decode:
             self._build_live_tables()
+        self._fused_prelude = self._fused_decode and all(
+            st.info['norm_out'] is None
+            and st.info['scale'] is not None
+            and st.info['norm'].normalized_shape[0] in (512, 1024, 1536, 2048)
+            for st in self.states)
 
     def _build_live_tables(self):
         """Per attention layer, per offset row: the indices of the keys the
@@ -201,7 +206,9 @@ class FastDecoder:
         # (a bf16 copy would be re-cast on every graph replay)
         w = {'ln_w': castf(st.info['norm'].weight),
              'ln_b': castf(st.info['norm'].bias),
-             'scale': cast(st.info['scale'])}
+             'scale': cast(st.info['scale']),
+             'scale32': None if st.info['scale'] is None else
+             castf(st.info['scale']).reshape(-1).contiguous()}
         if st.info['norm_out'] is not None:
             w['lno_w'] = castf(st.info['norm_out'].weight)
             w['lno_b'] = castf(st.info['norm_out'].bias)
@@ -221,12 +228,16 @@ class FastDecoder:
         return w
 
     def _lin(self, x, w, bias32, bias):
-        """Decode-step linear: the skinny-M weights-streaming kernel when the
-        fused path is on (hipBLASLt runs M<=128 GEMMs at ~0.5 TB/s; the
-        kernel streams the weight at HBM rate), F.linear otherwise."""
+        """Decode-step linear. The skinny-M weights-streaming kernel is kept
+        behind DALLE_AMD_SKINNY=1: measured on-box, the decode step is bound
+        by a ~5 us per-kernel execution floor (gen_kernel_stats profile), so
+        the 3-kernel skinny path (zero + gemm + cast) loses to one hipBLASLt
+        dispatch even though neither is near the weight-bandwidth floor."""
+        import os
         rows = x.numel() // x.shape[-1]
         if (self._fused_decode and rows <= 128 and x.shape[-1] % 32 == 0
-                and w.shape[0] % 4 == 0):
+                and w.shape[0] % 4 == 0
+                and os.environ.get('DALLE_AMD_SKINNY', '0') == '1'):
             from dalle_pytorch_amd.ops.dispatch import hip_module
             out = hip_module().skinny_gemm(
                 x.reshape(rows, x.shape[-1]), w, bias32)
@@ -382,7 +393,63 @@ class FastDecoder:
             return torch.addcmul(x, y, st.w['scale'])
         return x + y
 
+    def _body(self, st, z, offset_t):
+        """Branch compute after the (fused) LN+shift prelude: attention or FF."""
+        if st.is_attn:
+            return self._attn(st, z, offset_t, 1)
+        from dalle_pytorch_amd.ops import geglu
+        y = self._lin(z, st.w['ff1_w'], st.w['ff1_b32'], st.w['ff1_b'])
+        y = geglu(y)
+        return self._lin(y, st.w['ff2_w'], st.w['ff2_b32'], st.w['ff2_b'])
+
+    def _prelude(self, ext, stream, pend, st, off):
+        """One fused kernel: apply the previous branch's residual to the
+        stream, LayerNorm it for this branch, token-shift if wrapped."""
+        y, scale32 = pend
+        shift = st.info['shift']
+        xn, z = ext.dec_prelude(
+            stream.view(self.b, -1),
+            None if y is None else y.view(self.b, -1),
+            scale32, st.w['ln_w'], st.w['ln_b'],
+            st.ring if shift is not None else None,
+            self.offset_t, st.info['norm'].eps,
+            shift.image_size if shift is not None else 1,
+            shift.text_len if shift is not None else 0)
+        return xn.view(self.b, 1, -1), z.view(self.b, 1, -1)
+
+    def _run_stack_decode_fused(self, x, off):
+        """Decode step with fused preludes (the step is bound by a ~5 us
+        per-kernel floor: residual+LN+shift as one kernel per branch)."""
+        from dalle_pytorch_amd.ops.dispatch import hip_module
+        ext = hip_module()
+        it = iter(self.states)
+        if not self.reversible:
+            pend = (None, None)
+            last_scale = None
+            for st in self.states:
+                x, z = self._prelude(ext, x, pend, st, off)
+                pend = (self._body(st, z, off), st.w['scale32'])
+                last_scale = st.w['scale']
+            y = pend[0]
+            return torch.addcmul(x, y, last_scale) if last_scale is not None \
+                else x + y
+        x1, x2 = x, x.clone()
+        pend2 = (None, None)
+        g_st = None
+        for f_st in it:
+            g_st = next(it)
+            x2, zf = self._prelude(ext, x2, pend2, f_st, off)
+            f_out = self._body(f_st, zf, off)
+            x1, zg = self._prelude(ext, x1, (f_out, f_st.w['scale32']), g_st, off)
+            g_out = self._body(g_st, zg, off)
+            pend2 = (g_out, g_st.w['scale32'])
+        x2 = torch.addcmul(x2, pend2[0], g_st.w['scale']) \
+            if g_st.w['scale'] is not None else x2 + pend2[0]
+        return (x1 + x2) / 2
+
     def _run_stack(self, x, offset_t, n):
+        if n == 1 and getattr(self, '_fused_prelude', False):
+            return self._run_stack_decode_fused(x, offset_t)
         if not self.reversible:
             it = iter(self.states)
             for attn_st in it:

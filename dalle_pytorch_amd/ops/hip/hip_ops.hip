@@ -1011,6 +1011,9 @@ void skinny_gemm_kernel(
     const short* __restrict__ w,    // [N, K] bf16 (torch Linear layout)
     float* __restrict__ outf,       // [M, N] fp32, pre-zeroed
     int M, int N, int K, int kslice) {
+  // x slice staged once per block (M*kslice*2 <= 128 KB, host-enforced);
+  // +8 row pad despreads the B-frag banks
+  extern __shared__ short Xs[];
   const int nt = blockIdx.x;        // n-tile (64 rows of W)
   const int z = blockIdx.y;         // k-split index
   const int wave = threadIdx.x >> 6, lane = threadIdx.x & 63;
@@ -1018,24 +1021,39 @@ void skinny_gemm_kernel(
   const int n0 = nt * 64 + wave * 16;          // this wave's 16 W rows
   const int k0 = z * kslice;
   const int k1 = min(K, k0 + kslice);
+  const int kn = k1 - k0;
+  const int xpitch = kslice + 8;
   const int mt_n = (M + 15) >> 4;
   const bool single = gridDim.y == 1;
 
+  // cooperative x stage: 16B chunks, coalesced
+  for (int c = threadIdx.x; c < M * (kn >> 3); c += 256) {
+    const int m = c / (kn >> 3);
+    const int kk = (c - m * (kn >> 3)) * 8;
+    *reinterpret_cast<int4v*>(&Xs[m * xpitch + kk]) =
+        *reinterpret_cast<const int4v*>(x + (long)m * K + k0 + kk);
+  }
+  __syncthreads();
+
   const short* wrow = w + (long)(n0 + lq) * K;  // A-frag: row = lane&15
+  const bool wok = n0 + lq < N;
   f32x4 acc[8];
   #pragma unroll
   for (int i = 0; i < 8; ++i) acc[i] = f32x4{0, 0, 0, 0};
 
-  for (int k = k0; k < k1; k += 32) {
-    const bf16x8 af = (n0 + lq < N)
-        ? *reinterpret_cast<const bf16x8*>(wrow + k + 8 * grp)
-        : bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
+  // double-buffered A-frag stream over k; B-frags come from LDS
+  bf16x8 af = (wok && kn > 0)
+      ? *reinterpret_cast<const bf16x8*>(wrow + k0 + 8 * grp)
+      : bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
+  for (int k = 0; k < kn; k += 32) {
+    const bf16x8 cur = af;
+    if (wok && k + 32 < kn)
+      af = *reinterpret_cast<const bf16x8*>(wrow + k0 + k + 32 + 8 * grp);
+    #pragma unroll 4
     for (int mt = 0; mt < mt_n; ++mt) {
-      const int m = mt * 16 + lq;
-      const bf16x8 xf = (m < M)
-          ? *reinterpret_cast<const bf16x8*>(x + (long)m * K + k + 8 * grp)
-          : bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
-      acc[mt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, xf, acc[mt], 0, 0, 0);
+      const bf16x8 xf = *reinterpret_cast<const bf16x8*>(
+          &Xs[(mt * 16 + lq) * xpitch + k + 8 * grp]);
+      acc[mt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(cur, xf, acc[mt], 0, 0, 0);
     }
   }
 
@@ -1145,6 +1163,118 @@ void resls_bwd_kernel(const short* __restrict__ dout, const short* __restrict__ 
   const long prow = (long)blockIdx.x * rpp + rg;
   #pragma unroll
   for (int e = 0; e < 8; ++e) dg_part[prow * dim + ch + e] = dg[e];
+}
+
+// ---------------------------------------------------------------------------
+// Decode prelude: residual-apply + LayerNorm + optional token-shift for one
+// decode token, fused (the decode step is bound by a ~5 us per-kernel
+// execution floor, so 3 tiny kernels -> 1 is a direct win x24 per step):
+//   x_new = x (+ scale * y)            [the previous branch's residual]
+//   z     = LN(x_new) * w + b
+//   z     = token_shift(z) via the S-slot ring when ring != null
+// One wave per batch row; dim = PER * 64.
+// ---------------------------------------------------------------------------
+
+template <int PER>
+__global__ __launch_bounds__(256)
+void dec_prelude_kernel(
+    const short* __restrict__ x,      // [rows, dim]
+    const short* __restrict__ y,      // [rows, dim] or null
+    const float* __restrict__ scale,  // [dim] or null (with y)
+    const float* __restrict__ w,      // LN gamma [dim]
+    const float* __restrict__ bia,    // LN beta [dim]
+    short* __restrict__ xn,           // [rows, dim] updated stream
+    short* __restrict__ z,            // [rows, dim] LN(+shift) output
+    short* __restrict__ ring,         // [rows, S, dim/2] or null
+    const long* __restrict__ offset,  // [1]
+    long rows, float eps, int S, int text_len) {
+  constexpr int per = PER;
+  constexpr int dim = PER * 64;
+  const long row = blockIdx.x * 4 + (threadIdx.x >> 6);
+  const int lane = threadIdx.x & 63;
+  if (row >= rows) return;
+  const short* xr = x + row * (long)dim;
+  const short* yr = y ? y + row * (long)dim : nullptr;
+
+  float sum = 0.f, sq = 0.f;
+  float vals[PER];
+  #pragma unroll
+  for (int i = 0; i < per; i += 8) {
+    int4v v = *reinterpret_cast<const int4v*>(xr + lane * per + i);
+    const short* vs = reinterpret_cast<const short*>(&v);
+    short upd8[8];
+    if (yr) {
+      int4v u = *reinterpret_cast<const int4v*>(yr + lane * per + i);
+      const short* us = reinterpret_cast<const short*>(&u);
+      #pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        const float f = bf2f(vs[e]) + scale[lane * per + i + e] * bf2f(us[e]);
+        vals[i + e] = f;
+        upd8[e] = f2bf(f);
+        sum += f;
+        sq += f * f;
+      }
+      *reinterpret_cast<int4v*>(xn + row * (long)dim + lane * per + i) =
+          *reinterpret_cast<const int4v*>(upd8);
+    } else {
+      #pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        const float f = bf2f(vs[e]);
+        vals[i + e] = f;
+        sum += f;
+        sq += f * f;
+      }
+    }
+  }
+  #pragma unroll
+  for (int sft = 32; sft > 0; sft >>= 1) {
+    sum += __shfl_xor(sum, sft);
+    sq += __shfl_xor(sq, sft);
+  }
+  const float mu = sum / dim;
+  const float rstd = __frsqrt_rn(sq / dim - mu * mu + eps);
+
+  constexpr int half = dim / 2, quarter = dim / 4;
+  long pos = 0, prev = 0;
+  bool row_start = false;
+  if (ring != nullptr) {
+    const long g0 = *offset - text_len;
+    const long g = g0 < 0 ? 0 : g0;
+    pos = g % S;
+    prev = ((g - 1) % S + S) % S;
+    row_start = (g % S) == 0;
+  }
+  short* rg = ring ? ring + (row * S) * (long)half : nullptr;
+  short* zr = z + row * (long)dim;
+  #pragma unroll
+  for (int i = 0; i < per; i += 8) {
+    short out8[8], ln8[8];
+    #pragma unroll
+    for (int e = 0; e < 8; ++e) {
+      const int d = lane * per + i + e;
+      const short lnv = f2bf((vals[i + e] - mu) * rstd * w[d] + bia[d]);
+      ln8[e] = lnv;
+      short o = lnv;
+      if (ring != nullptr) {
+        if (d < quarter) {
+          o = rg[pos * half + d];
+        } else if (d < half) {
+          o = row_start ? (short)0 : rg[prev * half + d];
+        }
+      }
+      out8[e] = o;
+    }
+    // write the ring slot AFTER its reads (same thread owns both)
+    if (ring != nullptr) {
+      #pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        const int d = lane * per + i + e;
+        if (d < half) rg[pos * half + d] = ln8[e];
+      }
+    }
+    *reinterpret_cast<int4v*>(zr + lane * per + i) =
+        *reinterpret_cast<const int4v*>(out8);
+  }
 }
 
 // D = rowsum(dO * O) for the flash backward, fused (the ATen form costs
@@ -1673,21 +1803,149 @@ void fa_decode_part_list_kernel(
   Ki[wave][lane] = key;
   // same-wave produce/consume: no cross-wave barrier needed
 
-  // P*V: lane = d, serial over this part's listed keys
+  // P*V: lane = d over this part's listed keys. 4 independent partial
+  // accumulators keep 4 V-row loads in flight — the branchy serial form
+  // was one 128 B line per ~HBM latency (49 us/dispatch profiled); every
+  // listed key is live by construction so no zero-p branch is needed.
   const short* vrow0 = vc + ((long)bi * h + head) * N * 64;
-  float acc = 0.f;
-  for (int i = 0; i < jn; ++i) {
-    const float pi = Pl[wave][i];
-    if (pi != 0.f) {
-      const int ki = Ki[wave][i];
+  float a4[4] = {0.f, 0.f, 0.f, 0.f};
+  int i = 0;
+  for (; i + 4 <= jn; i += 4) {
+    #pragma unroll
+    for (int u = 0; u < 4; ++u) {
+      const int ki = Ki[wave][i + u];
       const float vv = (ki == (int)off) ? vsn[wave][lane]
                                         : bf2f(vrow0[(long)ki * 64 + lane]);
-      acc += pi * vv;
+      a4[u] += Pl[wave][i + u] * vv;
     }
   }
+  for (; i < jn; ++i) {
+    const int ki = Ki[wave][i];
+    const float vv = (ki == (int)off) ? vsn[wave][lane]
+                                      : bf2f(vrow0[(long)ki * 64 + lane]);
+    a4[0] += Pl[wave][i] * vv;
+  }
+  const float acc = (a4[0] + a4[1]) + (a4[2] + a4[3]);
   float* sl = scratch + (((long)bi * h + head) * KS + z) * 66;
   sl[2 + lane] = acc;
   if (lane == 0) { sl[0] = m; sl[1] = lsum; }
+}
+
+// One-pass decode attention over the live-key list: with the flagship axial
+// patterns a head sees <=352 live keys, little enough that ONE wave handles
+// the whole head (lane-parallel dots, shuffle softmax, unrolled PV with 4
+// loads in flight) — no key-split scratch, no combine kernel, KS=1. Four
+// heads per 256-thread block; grid (ceil(h/4), b).
+#define DEC_LMAX 1344
+__global__ __launch_bounds__(256)
+void fa_decode_one_kernel(
+    const short* __restrict__ qkv,    // [b, 3*h*64]
+    short* __restrict__ kc,           // [b, h, N, 64]
+    short* __restrict__ vc,
+    const float* __restrict__ cosv,
+    const float* __restrict__ sinv,
+    const long* __restrict__ offset,  // [1]
+    const int* __restrict__ live,     // [N, Lmax]
+    const int* __restrict__ live_cnt, // [N]
+    short* __restrict__ out,          // [b, h*64]
+    int b, int h, int N, int rot, float scale, int Lmax) {
+
+  __shared__ float qs[4][64], ksn[4][64], vsn[4][64];
+  __shared__ float Pl[4][DEC_LMAX / 4];
+  __shared__ int   Ki[4][DEC_LMAX / 4];
+
+  const int wave = threadIdx.x >> 6, lane = threadIdx.x & 63;
+  const int head = blockIdx.x * 4 + wave;
+  const int bi = blockIdx.y;
+  const long off = *offset;
+  const bool head_ok = head < h;
+
+  if (head_ok) {
+    const int d = lane;
+    const long base = (long)bi * 3 * h * 64 + (long)head * 64 + d;
+    float qv = bf2f(qkv[base]);
+    float kv = bf2f(qkv[base + (long)h * 64]);
+    float vv = bf2f(qkv[base + 2L * h * 64]);
+    if (d < rot && cosv != nullptr) {
+      const int prt = d ^ 1;
+      const float cs = cosv[off * rot + d];
+      const float sn = sinv[off * rot + d];
+      const float sgn = (d & 1) ? 1.f : -1.f;
+      const long pbase = (long)bi * 3 * h * 64 + (long)head * 64 + prt;
+      qv = qv * cs + sgn * bf2f(qkv[pbase]) * sn;
+      kv = kv * cs + sgn * bf2f(qkv[pbase + (long)h * 64]) * sn;
+      vv = vv * cs + sgn * bf2f(qkv[pbase + 2L * h * 64]) * sn;
+    }
+    const long cbase = (((long)bi * h + head) * N + off) * 64 + d;
+    kc[cbase] = f2bf(kv);
+    vc[cbase] = f2bf(vv);
+    qs[wave][d] = qv * scale;
+    ksn[wave][d] = kv;
+    vsn[wave][d] = vv;
+  }
+  __syncthreads();
+  if (!head_ok) return;
+
+  const int jn = live_cnt[off];
+  const int* lrow = live + off * (long)Lmax;
+  const short* krow0 = kc + ((long)bi * h + head) * N * 64;
+
+  // dots: lane-parallel over the listed keys; per-lane max/sum partials
+  float m_loc = NEG_INF;
+  for (int j = lane; j < jn; j += 64) {
+    const int key = lrow[j];
+    float p = 0.f;
+    if (key == (int)off) {
+      #pragma unroll
+      for (int d = 0; d < 64; ++d) p += qs[wave][d] * ksn[wave][d];
+    } else {
+      const short* krow = krow0 + (long)key * 64;
+      #pragma unroll
+      for (int c = 0; c < 8; ++c) {
+        int4v kk = *reinterpret_cast<const int4v*>(krow + c * 8);
+        const short* ks = reinterpret_cast<const short*>(&kk);
+        #pragma unroll
+        for (int e = 0; e < 8; ++e) p += qs[wave][c * 8 + e] * bf2f(ks[e]);
+      }
+    }
+    Pl[wave][j] = p;
+    Ki[wave][j] = key;
+    m_loc = fmaxf(m_loc, p);
+  }
+  #pragma unroll
+  for (int s = 32; s > 0; s >>= 1) m_loc = fmaxf(m_loc, __shfl_xor(m_loc, s));
+
+  float l_loc = 0.f;
+  for (int j = lane; j < jn; j += 64) {
+    const float p = __expf(Pl[wave][j] - m_loc);
+    Pl[wave][j] = p;
+    l_loc += p;
+  }
+  #pragma unroll
+  for (int s = 32; s > 0; s >>= 1) l_loc += __shfl_xor(l_loc, s);
+  const float inv = l_loc > 0.f ? 1.f / l_loc : 0.f;
+
+  // PV: lane = d; 4 partial accumulators keep 4 V rows in flight
+  const short* vrow0 = vc + ((long)bi * h + head) * N * 64;
+  float a4[4] = {0.f, 0.f, 0.f, 0.f};
+  int i = 0;
+  for (; i + 4 <= jn; i += 4) {
+    #pragma unroll
+    for (int u = 0; u < 4; ++u) {
+      const int ki = Ki[wave][i + u];
+      const float vv = (ki == (int)off) ? vsn[wave][lane]
+                                        : bf2f(vrow0[(long)ki * 64 + lane]);
+      a4[u] += Pl[wave][i + u] * vv;
+    }
+  }
+  for (; i < jn; ++i) {
+    const int ki = Ki[wave][i];
+    const float vv = (ki == (int)off) ? vsn[wave][lane]
+                                      : bf2f(vrow0[(long)ki * 64 + lane]);
+    a4[0] += Pl[wave][i] * vv;
+  }
+  out[(long)bi * h * 64 + head * 64 + lane] =
+      f2bf(((a4[0] + a4[1]) + (a4[2] + a4[3])) * inv);
 }
 
 __global__ __launch_bounds__(256)
@@ -2187,6 +2445,22 @@ torch::Tensor fa_decode(torch::Tensor qkv, torch::Tensor kc, torch::Tensor vc,
   }
   auto out = torch::empty({b, (long)h * 64}, qkv.options());
   const int span = live.has_value() ? (int)live->size(1) : N;
+  if (live.has_value() && span <= DEC_LMAX / 4) {
+    // short live lists (sparse patterns): whole head in one wave, no
+    // key-split scratch, no combine kernel
+    CHK(live->dtype() == torch::kInt32 && live->is_contiguous());
+    CHK(live_cnt.has_value() && live_cnt->dtype() == torch::kInt32);
+    hipLaunchKernelGGL(fa_decode_one_kernel, dim3((h + 3) / 4, b), dim3(256),
+                       0, cur_stream(),
+                       reinterpret_cast<const short*>(qkv.data_ptr()),
+                       reinterpret_cast<short*>(kc.data_ptr()),
+                       reinterpret_cast<short*>(vc.data_ptr()),
+                       cp, sp, offset.data_ptr<long>(),
+                       live->data_ptr<int>(), live_cnt->data_ptr<int>(),
+                       reinterpret_cast<short*>(out.data_ptr()),
+                       b, h, N, rot, (float)scale, span);
+    return out;
+  }
   // live lists: one 64-thread wave per 64 listed keys (no barriers);
   // scan path: 4-wave blocks over ~192-slot chunks
   int KS, chunk;
@@ -2340,6 +2614,52 @@ std::vector<torch::Tensor> ln_bwd(torch::Tensor x, torch::Tensor dy,
   return {dx, dgamma, dbeta};
 }
 
+std::vector<torch::Tensor> dec_prelude(
+    torch::Tensor x, std::optional<torch::Tensor> y,
+    std::optional<torch::Tensor> scale,
+    torch::Tensor w, torch::Tensor b,
+    std::optional<torch::Tensor> ring, torch::Tensor offset,
+    double eps, int64_t S, int64_t text_len) {
+  CHK(x.is_cuda() && x.dtype() == torch::kBFloat16 && x.is_contiguous());
+  CHK(w.dtype() == torch::kFloat32 && b.dtype() == torch::kFloat32);
+  const int dim = x.size(-1);
+  const long rows = x.numel() / dim;
+  const short* yp = nullptr;
+  const float* sp_ = nullptr;
+  torch::Tensor xn = x;
+  if (y.has_value()) {
+    CHK(y->is_contiguous() && scale.has_value());
+    CHK(scale->dtype() == torch::kFloat32 && scale->is_contiguous());
+    yp = reinterpret_cast<const short*>(y->data_ptr());
+    sp_ = scale->data_ptr<float>();
+    xn = torch::empty_like(x);
+  }
+  short* rp = nullptr;
+  if (ring.has_value()) {
+    CHK(ring->dtype() == torch::kBFloat16 && ring->is_contiguous());
+    rp = reinterpret_cast<short*>(ring->data_ptr());
+  }
+  auto z = torch::empty_like(x);
+  dim3 grid((rows + 3) / 4);
+  auto launch = [&](auto kern) {
+    hipLaunchKernelGGL(kern, grid, dim3(256), 0, cur_stream(),
+                       reinterpret_cast<const short*>(x.data_ptr()), yp, sp_,
+                       w.data_ptr<float>(), b.data_ptr<float>(),
+                       reinterpret_cast<short*>(xn.data_ptr()),
+                       reinterpret_cast<short*>(z.data_ptr()),
+                       rp, offset.data_ptr<long>(),
+                       rows, (float)eps, (int)S, (int)text_len);
+  };
+  switch (dim) {
+    case 512: launch(dec_prelude_kernel<8>); break;
+    case 1024: launch(dec_prelude_kernel<16>); break;
+    case 1536: launch(dec_prelude_kernel<24>); break;
+    case 2048: launch(dec_prelude_kernel<32>); break;
+    default: TORCH_CHECK(false, "dec_prelude: unsupported dim ", dim);
+  }
+  return {xn, z};
+}
+
 torch::Tensor skinny_gemm(torch::Tensor x, torch::Tensor w,
                           std::optional<torch::Tensor> bias) {
   CHK(x.is_cuda() && x.dtype() == torch::kBFloat16 && x.is_contiguous());
@@ -2353,14 +2673,19 @@ torch::Tensor skinny_gemm(torch::Tensor x, torch::Tensor w,
     bp = bf32.data_ptr<float>();
   }
   const int ntiles = (N + 63) / 64;
+  const int m_pad = ((M + 15) / 16) * 16;
   int ksplit = std::min<int>({16, std::max(1, 512 / ntiles), (K + 63) / 64});
   int kslice = ((K + ksplit - 1) / ksplit + 31) & ~31;
+  // x slice must fit LDS: m_pad * (kslice + 8) * 2 bytes
+  while ((long)m_pad * (kslice + 8) * 2 > 131072 && kslice > 32)
+    kslice = ((kslice / 2) + 31) & ~31;
   ksplit = (K + kslice - 1) / kslice;
+  const int lds_bytes = m_pad * (kslice + 8) * 2;
   auto outf = ksplit > 1
       ? torch::zeros({(long)M, (long)N}, x.options().dtype(torch::kFloat32))
       : torch::empty({(long)M, (long)N}, x.options().dtype(torch::kFloat32));
-  hipLaunchKernelGGL(skinny_gemm_kernel, dim3(ntiles, ksplit), dim3(256), 0,
-                     cur_stream(),
+  hipLaunchKernelGGL(skinny_gemm_kernel, dim3(ntiles, ksplit), dim3(256),
+                     lds_bytes, cur_stream(),
                      reinterpret_cast<const short*>(x.data_ptr()),
                      reinterpret_cast<const short*>(w.data_ptr()),
                      outf.data_ptr<float>(), M, N, K, kslice);
@@ -2418,6 +2743,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("skinny_gemm", &skinny_gemm,
         "skinny-M weights-streaming GEMM (decode projections)",
         py::arg("x"), py::arg("w"), py::arg("bias") = std::nullopt);
+  m.def("dec_prelude", &dec_prelude,
+        "fused decode residual + LayerNorm + token-shift");
   m.def("fa_fwd", &fa_fwd, "flash attention forward (gfx950, d=64)",
         py::arg("q"), py::arg("k"), py::arg("v"), py::arg("scale"),
         py::arg("causal"), py::arg("key_mask"), py::arg("static_mask"),
