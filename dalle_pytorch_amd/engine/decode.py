@@ -169,7 +169,7 @@ class FastDecoder:
         self._fused_prelude = self._fused_decode and all(
             st.info['norm_out'] is None
             and st.info['scale'] is not None
-            and st.info['norm'].normalized_shape[0] in (512, 1024, 1536, 2048)
+            and st.info['norm'].normalized_shape[0] in (512, 1024, 2048)
             for st in self.states)
 
     def _build_live_tables(self):

@@ -1175,7 +1175,7 @@ void resls_bwd_kernel(const short* __restrict__ dout, const short* __restrict__ 
 // One wave per batch row; dim = PER * 64.
 // ---------------------------------------------------------------------------
 
-template <int PER>
+template <int CPT>   // channels per thread = dim / 256
 __global__ __launch_bounds__(256)
 void dec_prelude_kernel(
     const short* __restrict__ x,      // [rows, dim]
@@ -1188,42 +1188,32 @@ void dec_prelude_kernel(
     short* __restrict__ ring,         // [rows, S, dim/2] or null
     const long* __restrict__ offset,  // [1]
     long rows, float eps, int S, int text_len) {
-  constexpr int per = PER;
-  constexpr int dim = PER * 64;
-  const long row = blockIdx.x * 4 + (threadIdx.x >> 6);
-  const int lane = threadIdx.x & 63;
-  if (row >= rows) return;
+  constexpr int dim = CPT * 256;
+  __shared__ float stat[8];
+  const long row = blockIdx.x;
+  const int tid = threadIdx.x;
+  const int c0 = tid * CPT;
   const short* xr = x + row * (long)dim;
   const short* yr = y ? y + row * (long)dim : nullptr;
 
   float sum = 0.f, sq = 0.f;
-  float vals[PER];
-  #pragma unroll
-  for (int i = 0; i < per; i += 8) {
-    int4v v = *reinterpret_cast<const int4v*>(xr + lane * per + i);
-    const short* vs = reinterpret_cast<const short*>(&v);
-    short upd8[8];
+  float vals[CPT];
+  {
+    // CPT is 2/4/8: load as that many shorts in one access
+    #pragma unroll
+    for (int e = 0; e < CPT; ++e) {
+      float f = bf2f(xr[c0 + e]);
+      if (yr) f += scale[c0 + e] * bf2f(yr[c0 + e]);
+      vals[e] = f;
+      sum += f;
+      sq += f * f;
+    }
     if (yr) {
-      int4v u = *reinterpret_cast<const int4v*>(yr + lane * per + i);
-      const short* us = reinterpret_cast<const short*>(&u);
+      short upd[CPT];
       #pragma unroll
-      for (int e = 0; e < 8; ++e) {
-        const float f = bf2f(vs[e]) + scale[lane * per + i + e] * bf2f(us[e]);
-        vals[i + e] = f;
-        upd8[e] = f2bf(f);
-        sum += f;
-        sq += f * f;
-      }
-      *reinterpret_cast<int4v*>(xn + row * (long)dim + lane * per + i) =
-          *reinterpret_cast<const int4v*>(upd8);
-    } else {
+      for (int e = 0; e < CPT; ++e) upd[e] = f2bf(vals[e]);
       #pragma unroll
-      for (int e = 0; e < 8; ++e) {
-        const float f = bf2f(vs[e]);
-        vals[i + e] = f;
-        sum += f;
-        sq += f * f;
-      }
+      for (int e = 0; e < CPT; ++e) xn[row * (long)dim + c0 + e] = upd[e];
     }
   }
   #pragma unroll
@@ -1231,6 +1221,13 @@ void dec_prelude_kernel(
     sum += __shfl_xor(sum, sft);
     sq += __shfl_xor(sq, sft);
   }
+  if ((tid & 63) == 0) {
+    stat[tid >> 6] = sum;
+    stat[4 + (tid >> 6)] = sq;
+  }
+  __syncthreads();
+  sum = stat[0] + stat[1] + stat[2] + stat[3];
+  sq = stat[4] + stat[5] + stat[6] + stat[7];
   const float mu = sum / dim;
   const float rstd = __frsqrt_rn(sq / dim - mu * mu + eps);
 
@@ -1246,35 +1243,32 @@ void dec_prelude_kernel(
   }
   short* rg = ring ? ring + (row * S) * (long)half : nullptr;
   short* zr = z + row * (long)dim;
+  short out_c[CPT], ln_c[CPT];
   #pragma unroll
-  for (int i = 0; i < per; i += 8) {
-    short out8[8], ln8[8];
-    #pragma unroll
-    for (int e = 0; e < 8; ++e) {
-      const int d = lane * per + i + e;
-      const short lnv = f2bf((vals[i + e] - mu) * rstd * w[d] + bia[d]);
-      ln8[e] = lnv;
-      short o = lnv;
-      if (ring != nullptr) {
-        if (d < quarter) {
-          o = rg[pos * half + d];
-        } else if (d < half) {
-          o = row_start ? (short)0 : rg[prev * half + d];
-        }
-      }
-      out8[e] = o;
-    }
-    // write the ring slot AFTER its reads (same thread owns both)
+  for (int e = 0; e < CPT; ++e) {
+    const int d = c0 + e;
+    const short lnv = f2bf((vals[e] - mu) * rstd * w[d] + bia[d]);
+    ln_c[e] = lnv;
+    short o = lnv;
     if (ring != nullptr) {
-      #pragma unroll
-      for (int e = 0; e < 8; ++e) {
-        const int d = lane * per + i + e;
-        if (d < half) rg[pos * half + d] = ln8[e];
+      if (d < quarter) {
+        o = rg[pos * half + d];
+      } else if (d < half) {
+        o = row_start ? (short)0 : rg[prev * half + d];
       }
     }
-    *reinterpret_cast<int4v*>(zr + lane * per + i) =
-        *reinterpret_cast<const int4v*>(out8);
+    out_c[e] = o;
   }
+  if (ring != nullptr) {
+    // write the ring slot AFTER its reads (same thread owns both)
+    #pragma unroll
+    for (int e = 0; e < CPT; ++e) {
+      const int d = c0 + e;
+      if (d < half) rg[pos * half + d] = ln_c[e];
+    }
+  }
+  #pragma unroll
+  for (int e = 0; e < CPT; ++e) zr[c0 + e] = out_c[e];
 }
 
 // D = rowsum(dO * O) for the flash backward, fused (the ATen form costs
@@ -1890,9 +1884,9 @@ void fa_decode_one_kernel(
   const int* lrow = live + off * (long)Lmax;
   const short* krow0 = kc + ((long)bi * h + head) * N * 64;
 
-  // dots: lane-parallel over the listed keys; per-lane max/sum partials
+  // dots: lane-parallel over the listed keys, two rows per lane in flight
   float m_loc = NEG_INF;
-  for (int j = lane; j < jn; j += 64) {
+  auto dot_one = [&](int j) {
     const int key = lrow[j];
     float p = 0.f;
     if (key == (int)off) {
@@ -1900,10 +1894,13 @@ void fa_decode_one_kernel(
       for (int d = 0; d < 64; ++d) p += qs[wave][d] * ksn[wave][d];
     } else {
       const short* krow = krow0 + (long)key * 64;
+      int4v kk[8];
+      #pragma unroll
+      for (int c = 0; c < 8; ++c)
+        kk[c] = *reinterpret_cast<const int4v*>(krow + c * 8);
       #pragma unroll
       for (int c = 0; c < 8; ++c) {
-        int4v kk = *reinterpret_cast<const int4v*>(krow + c * 8);
-        const short* ks = reinterpret_cast<const short*>(&kk);
+        const short* ks = reinterpret_cast<const short*>(&kk[c]);
         #pragma unroll
         for (int e = 0; e < 8; ++e) p += qs[wave][c * 8 + e] * bf2f(ks[e]);
       }
@@ -1911,7 +1908,13 @@ void fa_decode_one_kernel(
     Pl[wave][j] = p;
     Ki[wave][j] = key;
     m_loc = fmaxf(m_loc, p);
+  };
+  int j = lane;
+  for (; j + 64 < jn; j += 128) {
+    dot_one(j);
+    dot_one(j + 64);
   }
+  if (j < jn) dot_one(j);
   #pragma unroll
   for (int s = 32; s > 0; s >>= 1) m_loc = fmaxf(m_loc, __shfl_xor(m_loc, s));
 
@@ -1925,27 +1928,45 @@ void fa_decode_one_kernel(
   for (int s = 32; s > 0; s >>= 1) l_loc += __shfl_xor(l_loc, s);
   const float inv = l_loc > 0.f ? 1.f / l_loc : 0.f;
 
-  // PV: lane = d; 4 partial accumulators keep 4 V rows in flight
+  // PV with a key-group split: lane = (kg 0..3) x (dd 0..15); each lane
+  // covers d = 4*dd..4*dd+3 via one 8-byte V load and walks keys kg, kg+4,
+  // ... — 4-way key parallelism x 4 loads in flight instead of one V row
+  // per HBM latency.
   const short* vrow0 = vc + ((long)bi * h + head) * N * 64;
-  float a4[4] = {0.f, 0.f, 0.f, 0.f};
-  int i = 0;
-  for (; i + 4 <= jn; i += 4) {
+  const int kg = lane >> 4, dd = lane & 15;
+  const int d0 = dd * 4;
+  float a4[4][4];
+  #pragma unroll
+  for (int u = 0; u < 4; ++u)
     #pragma unroll
-    for (int u = 0; u < 4; ++u) {
-      const int ki = Ki[wave][i + u];
-      const float vv = (ki == (int)off) ? vsn[wave][lane]
-                                        : bf2f(vrow0[(long)ki * 64 + lane]);
-      a4[u] += Pl[wave][i + u] * vv;
+    for (int e = 0; e < 4; ++e) a4[u][e] = 0.f;
+  auto pv_one = [&](int j, float* a) {
+    const int ki = Ki[wave][j];
+    const float pj = Pl[wave][j];
+    if (ki == (int)off) {
+      #pragma unroll
+      for (int e = 0; e < 4; ++e) a[e] += pj * vsn[wave][d0 + e];
+    } else {
+      const bf16x4 v4 = *reinterpret_cast<const bf16x4*>(
+          vrow0 + (long)ki * 64 + d0);
+      #pragma unroll
+      for (int e = 0; e < 4; ++e) a[e] += pj * bf2f(v4[e]);
     }
+  };
+  int i = kg;
+  for (; i + 12 < jn; i += 16) {
+    #pragma unroll
+    for (int u = 0; u < 4; ++u) pv_one(i + 4 * u, a4[u]);
   }
-  for (; i < jn; ++i) {
-    const int ki = Ki[wave][i];
-    const float vv = (ki == (int)off) ? vsn[wave][lane]
-                                      : bf2f(vrow0[(long)ki * 64 + lane]);
-    a4[0] += Pl[wave][i] * vv;
+  for (; i < jn; i += 4) pv_one(i, a4[0]);
+  #pragma unroll
+  for (int e = 0; e < 4; ++e) {
+    float s = (a4[0][e] + a4[1][e]) + (a4[2][e] + a4[3][e]);
+    s += __shfl_xor(s, 16);       // fold the 4 key groups
+    s += __shfl_xor(s, 32);
+    if (kg == 0)
+      out[(long)bi * h * 64 + head * 64 + d0 + e] = f2bf(s * inv);
   }
-  out[(long)bi * h * 64 + head * 64 + lane] =
-      f2bf(((a4[0] + a4[1]) + (a4[2] + a4[3])) * inv);
 }
 
 __global__ __launch_bounds__(256)
@@ -2640,7 +2661,7 @@ std::vector<torch::Tensor> dec_prelude(
     rp = reinterpret_cast<short*>(ring->data_ptr());
   }
   auto z = torch::empty_like(x);
-  dim3 grid((rows + 3) / 4);
+  dim3 grid(rows);
   auto launch = [&](auto kern) {
     hipLaunchKernelGGL(kern, grid, dim3(256), 0, cur_stream(),
                        reinterpret_cast<const short*>(x.data_ptr()), yp, sp_,
@@ -2651,10 +2672,9 @@ std::vector<torch::Tensor> dec_prelude(
                        rows, (float)eps, (int)S, (int)text_len);
   };
   switch (dim) {
-    case 512: launch(dec_prelude_kernel<8>); break;
-    case 1024: launch(dec_prelude_kernel<16>); break;
-    case 1536: launch(dec_prelude_kernel<24>); break;
-    case 2048: launch(dec_prelude_kernel<32>); break;
+    case 512: launch(dec_prelude_kernel<2>); break;
+    case 1024: launch(dec_prelude_kernel<4>); break;
+    case 2048: launch(dec_prelude_kernel<8>); break;
     default: TORCH_CHECK(false, "dec_prelude: unsupported dim ", dim);
   }
   return {xn, z};
