@@ -1954,9 +1954,9 @@ void fa_decode_one_kernel(
     }
   };
   int i = kg;
-  for (; i + 12 < jn; i += 16) {
+  for (; i + 28 < jn; i += 32) {
     #pragma unroll
-    for (int u = 0; u < 4; ++u) pv_one(i + 4 * u, a4[u]);
+    for (int u = 0; u < 8; ++u) pv_one(i + 4 * u, a4[u & 3]);
   }
   for (; i < jn; i += 4) pv_one(i, a4[0]);
   #pragma unroll
