@@ -435,6 +435,318 @@ void fa_fwd_d64_kernel(
 }
 
 // ---------------------------------------------------------------------------
+// 8-wave 32x32 MFMA attention forward ("fa8 ladder", guide §B attn /
+// NOTES_ROUND2 item 1, layouts machine-checked by scripts/plan_fa8.py +
+// tests/test_fa8_plan.py). Block = 8 waves x 32 q rows = 256 q rows;
+// KV staged in 64-key LDS tiles. Per wave everything stays in registers:
+// swapped QK^T (mfma(K, Q^T) -> S^T with q = lane&31 lane-local), 15-op
+// in-lane + 1 permlane32_swap row reduces, defer-max online softmax, P
+// packed to bf16 pairs and redistributed across the lane halves with 8
+// permlane32_swaps per tile (no P LDS round-trip), PV from the transposed
+// V tile. Covers dense causal/non-causal, key_mask, and axial mode; the
+// masked/tile-map patterns stay on the 4-wave 16x16 kernel.
+// ---------------------------------------------------------------------------
+
+using f32x16 = __attribute__((ext_vector_type(16))) float;
+
+constexpr int FA8_QBLK = 256;
+constexpr int FA8_KV = 64;
+
+// K tile [64 keys][64 d] bf16 at 128 B rows; byte ^= (row&3)<<5 floors both
+// the 16 B-chunk store and the b128 A-fragment read (plan_fa8 d64_swizzle)
+DEVFN short* k8_addr(short* base, int row, int byte) {
+  return reinterpret_cast<short*>(
+      reinterpret_cast<char*>(base) + ((row * 128 + byte) ^ ((row & 3) << 5)));
+}
+
+DEVFN unsigned pk_bf16(float a, float b) {
+  return (unsigned)(unsigned short)f2bf(a) |
+         ((unsigned)(unsigned short)f2bf(b) << 16);
+}
+
+// value of `w` in lane (lane ^ 32): one v_permlane32_swap_b32, result
+// element picked per half (semantics pinned by test_permlane_semantics)
+DEVFN unsigned partner_u32(unsigned w, bool hi_half) {
+  auto r = __builtin_amdgcn_permlane32_swap(w, w, false, false);
+  return hi_half ? r[0] : r[1];
+}
+
+DEVFN float partner_f32(float v, bool hi_half) {
+  union { float f; unsigned u; } c;
+  c.f = v;
+  c.u = partner_u32(c.u, hi_half);
+  return c.f;
+}
+
+__global__ __launch_bounds__(512, 2)
+void fa8_fwd_d64_kernel(
+    const short* __restrict__ q,    // [bh, nq, 64] bf16 bits
+    const short* __restrict__ k,    // [bh, nk, 64]
+    const short* __restrict__ v,    // [bh, nk, 64]
+    short* __restrict__ out,        // [bh, nq, 64] or [b, nq, h, 64]
+    float* __restrict__ lse,        // [bh, nq]
+    const bool* __restrict__ key_mask,   // [b, nk] or null
+    int b, int h, int nq, int nk,
+    float scale, int causal, int out_bnhd,
+    int ax_t, int ax_logS, int ax_axis) {
+
+  __shared__ short K8[64 * 64];
+  __shared__ short V8t[64][FA8_KV + 8];
+  __shared__ float bcast[8][32];
+
+  const int n_qt = (nq + FA8_QBLK - 1) / FA8_QBLK;
+  int qtile, bh;
+  xcd_chunked(blockIdx.x, n_qt * b * h, n_qt, &qtile, &bh);
+  const int batch = bh / h;
+  const int q0 = qtile * FA8_QBLK;
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6, lane = tid & 63;
+  const int lq = lane & 31;              // this lane's q row within the wave
+  const int half = lane >> 5;            // lane half (0/1)
+  const int qrow = q0 + wave * 32 + lq;
+  const int diag = nk - nq;
+  const bool axial = ax_axis >= 0;
+  const int qphys = axial ? ax_phys(qrow, ax_t, ax_logS, ax_axis) : qrow;
+
+  const short* qp = q + (long)bh * nq * FA_D;
+  const short* kp = k + (long)bh * nk * FA_D;
+  const short* vp = v + (long)bh * nk * FA_D;
+
+  // Q fragments (B operand): lane holds Q[q = lane&31][16c + 8*half + e]
+  bf16x8 qfrag[4];
+  {
+    const bool qok = qrow < nq;
+    #pragma unroll
+    for (int c = 0; c < 4; ++c) {
+      qfrag[c] = qok
+          ? *reinterpret_cast<const bf16x8*>(
+                qp + (long)qphys * FA_D + 16 * c + 8 * half)
+          : bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
+    }
+  }
+
+  float m_run = NEG_INF, l_run = 0.f;
+  f32x16 acc_o[2] = {};
+  const int wave_qmax = q0 + wave * 32 + 31 + diag;   // causal bound
+
+  int ntiles = (nk + FA8_KV - 1) / FA8_KV;
+  if (causal) {
+    const int lim = min(nk - 1, q0 + FA8_QBLK - 1 + diag);
+    ntiles = lim < 0 ? 0 : (lim / FA8_KV + 1);
+  }
+  auto tile_live_blk = [&](int t) -> bool {
+    if (!axial) return true;
+    const int kbase = t * FA8_KV;
+    if (kbase < ax_t) return true;
+    if (q0 + FA8_QBLK <= ax_t) return false;
+    const int l0 = (max(q0, ax_t) - ax_t) >> ax_logS;
+    return kbase + FA8_KV > ax_t + (l0 << ax_logS);
+  };
+  auto next_live = [&](int t) -> int {
+    while (t < ntiles && !tile_live_blk(t)) ++t;
+    return t;
+  };
+
+  // staging: 512 threads x 16 B = one 64x128 B tile per pass
+  const int srow = tid >> 3;
+  const int sc8 = (tid & 7) * 8;
+  int4v kreg, vreg;
+  auto prefetch = [&](int t) {
+    const int kg = t * FA8_KV + srow;
+    if (kg < nk) {
+      const int kph = axial ? ax_phys(kg, ax_t, ax_logS, ax_axis) : kg;
+      kreg = *reinterpret_cast<const int4v*>(kp + (long)kph * FA_D + sc8);
+      vreg = *reinterpret_cast<const int4v*>(vp + (long)kph * FA_D + sc8);
+    } else {
+      kreg = int4v{0, 0, 0, 0};
+      vreg = int4v{0, 0, 0, 0};
+    }
+  };
+
+  int kt = next_live(0);
+  if (kt < ntiles) prefetch(kt);
+
+  while (kt < ntiles) {
+    const int kbase = kt * FA8_KV;
+    const int kt_next = next_live(kt + 1);
+
+    __syncthreads();
+    *reinterpret_cast<int4v*>(k8_addr(K8, srow, sc8 * 2)) = kreg;
+    {
+      const short* vs = reinterpret_cast<const short*>(&vreg);
+      #pragma unroll
+      for (int e = 0; e < 8; ++e)
+        V8t[sc8 + e][swz_key(sc8 + e, srow)] = vs[e];
+    }
+    __syncthreads();
+    if (kt_next < ntiles) prefetch(kt_next);
+
+    const bool wave_dead = causal && kbase > wave_qmax;
+    if (!wave_dead) {
+      // ---- swapped QK^T: two 32-key subtiles, contraction K=64 in 4 mfmas
+      f32x16 acc_s[2] = {};
+      __builtin_amdgcn_s_setprio(1);
+      #pragma unroll
+      for (int s = 0; s < 2; ++s) {
+        #pragma unroll
+        for (int c = 0; c < 4; ++c) {
+          const int row = 32 * s + lq;
+          const bf16x8 af = *reinterpret_cast<const bf16x8*>(
+              k8_addr(K8, row, (16 * c + 8 * half) * 2));
+          acc_s[s] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              af, qfrag[c], acc_s[s], 0, 0, 0);
+        }
+      }
+      __builtin_amdgcn_s_setprio(0);
+
+      // ---- scale + masks. Element (s, r): key = kbase + 32 s + crow(r),
+      // q = qrow, with crow(r) = (r&3) + 8*(r>>2) + 4*half.
+      const bool interior = key_mask == nullptr &&
+          (kbase + FA8_KV <= nk) && (qrow < nq) &&
+          (axial
+               ? ((q0 >= ax_t && kbase + FA8_KV <= ax_t) ||
+                  (q0 + FA8_QBLK <= ax_t && kbase + FA8_KV - 1 <= qrow &&
+                   kbase + FA8_KV - 1 <= q0 + wave * 32 + diag))
+               : (!causal || kbase + FA8_KV - 1 <= q0 + wave * 32 + diag));
+      float s32[32];
+      if (interior) {
+        #pragma unroll
+        for (int s = 0; s < 2; ++s)
+          #pragma unroll
+          for (int r = 0; r < 16; ++r) s32[16 * s + r] = acc_s[s][r] * scale;
+      } else {
+        #pragma unroll
+        for (int s = 0; s < 2; ++s) {
+          #pragma unroll
+          for (int r = 0; r < 16; ++r) {
+            const int key = kbase + 32 * s + (r & 3) + 8 * (r >> 2) + 4 * half;
+            bool ok = (key < nk) & (qrow < nq);
+            if (axial) {
+              ok = ok && ax_ok(qrow, key, ax_t, ax_logS);
+            } else if (causal) {
+              ok &= key <= qrow + diag;
+            }
+            if (key_mask != nullptr && ok) {
+              const int kph = axial ? ax_phys(key, ax_t, ax_logS, ax_axis) : key;
+              ok &= key_mask[(long)batch * nk + kph];
+            }
+            s32[16 * s + r] = ok ? acc_s[s][r] * scale : NEG_INF;
+          }
+        }
+      }
+
+      // ---- online softmax: 31-value in-lane max + one permlane32_swap
+      float mt = NEG_INF;
+      #pragma unroll
+      for (int i = 0; i < 32; ++i) mt = fmaxf(mt, s32[i]);
+      mt = fmaxf(mt, partner_f32(mt, half));
+
+      const bool defer = __all(mt <= m_run + 8.f);
+      const float m_new = defer ? m_run : fmaxf(m_run, mt);
+
+      float p32[32];
+      float lsum = 0.f;
+      #pragma unroll
+      for (int i = 0; i < 32; ++i) {
+        p32[i] = (s32[i] == NEG_INF) ? 0.f : __expf(s32[i] - m_new);
+        lsum += p32[i];
+      }
+      lsum += partner_f32(lsum, half);
+
+      float alpha = 1.f;
+      if (!defer) {
+        alpha = (m_run == NEG_INF) ? 0.f : __expf(m_run - m_new);
+        if (m_new != NEG_INF) m_run = m_new;
+        // alpha is per q (= lane&31); the PV accumulator rows are crow(r)
+        // -> broadcast through this wave's 32-slot LDS row
+        if (lane < 32) bcast[wave][lq] = alpha;
+        #pragma unroll
+        for (int d0 = 0; d0 < 2; ++d0) {
+          #pragma unroll
+          for (int r = 0; r < 16; ++r) {
+            const int qr = (r & 3) + 8 * (r >> 2) + 4 * half;
+            acc_o[d0][r] *= bcast[wave][qr];
+          }
+        }
+      }
+      l_run = l_run * alpha + lsum;
+
+      // ---- P -> bf16 pairs, partner halves fetched with permlane32_swap
+      // (plan_fa8 p_value map). A-frag for key slice ks: rbase = 4*(2*(ks&1)
+      // + half) in subtile ks>>1; [own 4 | partner 4] ordered by half.
+      bf16x8 paf[4];
+      #pragma unroll
+      for (int ks = 0; ks < 4; ++ks) {
+        const float* ps = &p32[16 * (ks >> 1)];
+        const int rb = 4 * (2 * (ks & 1) + half);
+        const unsigned w0 = pk_bf16(ps[rb], ps[rb + 1]);
+        const unsigned w1 = pk_bf16(ps[rb + 2], ps[rb + 3]);
+        const unsigned pw0 = partner_u32(w0, half);
+        const unsigned pw1 = partner_u32(w1, half);
+        unsigned fr[4];
+        if (half == 0) {
+          fr[0] = w0; fr[1] = w1; fr[2] = pw0; fr[3] = pw1;
+        } else {
+          fr[0] = pw0; fr[1] = pw1; fr[2] = w0; fr[3] = w1;
+        }
+        paf[ks] = *reinterpret_cast<const bf16x8*>(fr);
+      }
+
+      // ---- PV over the transposed V tile
+      __builtin_amdgcn_s_setprio(1);
+      #pragma unroll
+      for (int d0 = 0; d0 < 2; ++d0) {
+        const int d = 32 * d0 + lq;
+        #pragma unroll
+        for (int ks = 0; ks < 4; ++ks) {
+          const bf16x8 vf = frag_from_lds(
+              &V8t[d][swz_key(d, 16 * ks + 8 * half)]);
+          acc_o[d0] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              paf[ks], vf, acc_o[d0], 0, 0, 0);
+        }
+      }
+      __builtin_amdgcn_s_setprio(0);
+    }
+
+    kt = kt_next;
+  }
+
+  // ---- epilogue: normalize rows by l_run[qrow] (LDS broadcast), store
+  if (lane < 32) bcast[wave][lq] = l_run > 0.f ? 1.f / l_run : 0.f;
+  const int batch_i = bh / h, head_i = bh - batch_i * h;
+  #pragma unroll
+  for (int d0 = 0; d0 < 2; ++d0) {
+    #pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      const int qr_loc = (r & 3) + 8 * (r >> 2) + 4 * half;
+      const int qr = q0 + wave * 32 + qr_loc;
+      if (qr < nq) {
+        const int qrp = axial ? ax_phys(qr, ax_t, ax_logS, ax_axis) : qr;
+        const long base = out_bnhd
+            ? (((long)batch_i * nq + qrp) * h + head_i) * FA_D
+            : ((long)bh * nq + qrp) * FA_D;
+        out[base + 32 * d0 + lq] = f2bf(acc_o[d0][r] * bcast[wave][qr_loc]);
+      }
+    }
+  }
+  if (lane < 32 && qrow < nq) {
+    lse[(long)bh * nq + qphys] = (l_run > 0.f) ? m_run + __logf(l_run) : NEG_INF;
+  }
+}
+
+// probe for v_permlane32_swap_b32 result-element semantics (consumed by
+// tests/test_gpu_kernels.py::test_permlane_semantics)
+__global__ void permlane_probe_kernel(const unsigned* __restrict__ a,
+                                      const unsigned* __restrict__ bsrc,
+                                      unsigned* __restrict__ r0,
+                                      unsigned* __restrict__ r1) {
+  const unsigned va = a[threadIdx.x], vb = bsrc[threadIdx.x];
+  auto r = __builtin_amdgcn_permlane32_swap(va, vb, false, false);
+  r0[threadIdx.x] = r[0];
+  r1[threadIdx.x] = r[1];
+}
+
+// ---------------------------------------------------------------------------
 // Flash attention backward, head_dim = 64 (two passes, standard flash
 // decomposition — no n x n matrix in HBM):
 //   D[q]   = rowsum(dO * O)                       (host-side fused reduce)
@@ -2252,6 +2564,25 @@ std::vector<torch::Tensor> fa_fwd(torch::Tensor q, torch::Tensor k, torch::Tenso
     tm = tile_map->data_ptr<uint8_t>();
   }
 
+  // 8-wave 32x32 ladder for dense/axial shapes (no static-mask support);
+  // DALLE_AMD_FA8=0 falls back to the 4-wave 16x16 kernel
+  static const bool fa8_on = []() {
+    const char* e = getenv("DALLE_AMD_FA8");
+    return e == nullptr || e[0] != '0';
+  }();
+  if (fa8_on && sm == nullptr && tm == nullptr && nq >= 64) {
+    dim3 grid8(((nq + FA8_QBLK - 1) / FA8_QBLK) * b * h);
+    hipLaunchKernelGGL(fa8_fwd_d64_kernel, grid8, dim3(512), 0, cur_stream(),
+                       reinterpret_cast<const short*>(q.data_ptr()),
+                       reinterpret_cast<const short*>(k.data_ptr()),
+                       reinterpret_cast<const short*>(v.data_ptr()),
+                       reinterpret_cast<short*>(out.data_ptr()),
+                       lse.data_ptr<float>(), km,
+                       b, h, nq, nk, (float)scale, causal ? 1 : 0,
+                       out_bnhd ? 1 : 0,
+                       (int)ax_t, ax_logS, (int)ax_axis);
+    return {out, lse};
+  }
   dim3 grid(((nq + FA_QBLK - 1) / FA_QBLK) * b * h);
   hipLaunchKernelGGL(fa_fwd_d64_kernel, grid, dim3(256), 0, cur_stream(),
                      reinterpret_cast<const short*>(q.data_ptr()),
@@ -2263,6 +2594,18 @@ std::vector<torch::Tensor> fa_fwd(torch::Tensor q, torch::Tensor k, torch::Tenso
                      out_bnhd ? 1 : 0,
                      (int)ax_t, ax_logS, (int)ax_axis);
   return {out, lse};
+}
+
+std::vector<torch::Tensor> permlane_probe(torch::Tensor a, torch::Tensor b_) {
+  CHK(a.is_cuda() && a.dtype() == torch::kInt32 && a.numel() == 64);
+  auto r0 = torch::empty_like(a);
+  auto r1 = torch::empty_like(a);
+  hipLaunchKernelGGL(permlane_probe_kernel, dim3(1), dim3(64), 0, cur_stream(),
+                     reinterpret_cast<const unsigned*>(a.data_ptr()),
+                     reinterpret_cast<const unsigned*>(b_.data_ptr()),
+                     reinterpret_cast<unsigned*>(r0.data_ptr()),
+                     reinterpret_cast<unsigned*>(r1.data_ptr()));
+  return {r0, r1};
 }
 
 std::vector<torch::Tensor> fa_bwd(torch::Tensor q, torch::Tensor k, torch::Tensor v,
@@ -2765,6 +3108,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("x"), py::arg("w"), py::arg("bias") = std::nullopt);
   m.def("dec_prelude", &dec_prelude,
         "fused decode residual + LayerNorm + token-shift");
+  m.def("permlane_probe", &permlane_probe,
+        "v_permlane32_swap_b32 semantics probe");
   m.def("fa_fwd", &fa_fwd, "flash attention forward (gfx950, d=64)",
         py::arg("q"), py::arg("k"), py::arg("v"), py::arg("scale"),
         py::arg("causal"), py::arg("key_mask"), py::arg("static_mask"),

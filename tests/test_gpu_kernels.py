@@ -449,6 +449,23 @@ def test_fa_fully_masked_rows_no_nan(ext):
         assert (g[1].float() == 0).all(), 'masked batch leaked gradient'
 
 
+def test_permlane_semantics(ext):
+    """Pin v_permlane32_swap_b32: with (a, b) the results must be
+    r0 = [a_lo | b_lo-from-partner...] — concretely, the fa8 kernel's
+    partner_u32(w) = (half ? r0 : r1) of swap(w, w) must yield w[lane^32]."""
+    a = torch.arange(64, dtype=torch.int32, device='cuda')
+    b = a + 1000
+    r0, r1 = ext.permlane_probe(a, b)
+    lanes = torch.arange(64)
+    partner = torch.where(lanes < 32, lanes + 32, lanes - 32)
+    # the kernel's assumption: swap(w, w) gives partner value in r1 for
+    # lanes < 32 and in r0 for lanes >= 32
+    s0, s1 = ext.permlane_probe(a, a)
+    got = torch.where(lanes.cuda() < 32, s1, s0)
+    assert torch.equal(got.cpu(), a.cpu()[partner]), \
+        (r0.cpu().tolist(), r1.cpu().tolist())
+
+
 def test_skinny_gemm_vs_oracle(ext):
     """Weights-streaming skinny-M GEMM (decode projections) vs rocBLAS at
     the real decode shapes, with and without bias, M 64 and 128."""
