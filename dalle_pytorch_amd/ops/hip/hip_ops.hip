@@ -460,8 +460,12 @@ DEVFN short* k8_addr(short* base, int row, int byte) {
 }
 
 DEVFN unsigned pk_bf16(float a, float b) {
-  return (unsigned)(unsigned short)f2bf(a) |
-         ((unsigned)(unsigned short)f2bf(b) << 16);
+  // native casts pair into one v_cvt_pk_bf16_f32 (guide T12/m240: the
+  // compiler's pick beats both hand asm and a manual RNE bit-trick here)
+  union { __bf16 h[2]; unsigned u; } c;
+  c.h[0] = (__bf16)a;
+  c.h[1] = (__bf16)b;
+  return c.u;
 }
 
 // value of `w` in lane (lane ^ 32): one v_permlane32_swap_b32, result
@@ -674,20 +678,27 @@ void fa8_fwd_d64_kernel(
       // ---- P -> bf16 pairs, partner halves fetched with permlane32_swap
       // (plan_fa8 p_value map). A-frag for key slice ks: rbase = 4*(2*(ks&1)
       // + half) in subtile ks>>1; [own 4 | partner 4] ordered by half.
+      // Every lane packs BOTH register groups of the slice — LOW (regs
+      // 8t..8t+3: the h=0 frag's domain) and HIGH (8t+4..8t+7: h=1's) —
+      // and ONE permlane32_swap(wL, wH) hands each half exactly the
+      // partner word it needs (r[1].lo = partner LOW for h=0, r[0].hi =
+      // partner HIGH for h=1): 8 cvt-packs + 2 swaps per 16-key slice.
       bf16x8 paf[4];
       #pragma unroll
       for (int ks = 0; ks < 4; ++ks) {
         const float* ps = &p32[16 * (ks >> 1)];
-        const int rb = 4 * (2 * (ks & 1) + half);
-        const unsigned w0 = pk_bf16(ps[rb], ps[rb + 1]);
-        const unsigned w1 = pk_bf16(ps[rb + 2], ps[rb + 3]);
-        const unsigned pw0 = partner_u32(w0, half);
-        const unsigned pw1 = partner_u32(w1, half);
+        const int rL = 8 * (ks & 1);
+        const unsigned wL0 = pk_bf16(ps[rL], ps[rL + 1]);
+        const unsigned wL1 = pk_bf16(ps[rL + 2], ps[rL + 3]);
+        const unsigned wH0 = pk_bf16(ps[rL + 4], ps[rL + 5]);
+        const unsigned wH1 = pk_bf16(ps[rL + 6], ps[rL + 7]);
+        auto rA = __builtin_amdgcn_permlane32_swap(wL0, wH0, false, false);
+        auto rB = __builtin_amdgcn_permlane32_swap(wL1, wH1, false, false);
         unsigned fr[4];
         if (half == 0) {
-          fr[0] = w0; fr[1] = w1; fr[2] = pw0; fr[3] = pw1;
+          fr[0] = wL0; fr[1] = wL1; fr[2] = rA[1]; fr[3] = rB[1];
         } else {
-          fr[0] = pw0; fr[1] = pw1; fr[2] = w0; fr[3] = w1;
+          fr[0] = rA[0]; fr[1] = rB[0]; fr[2] = wH0; fr[3] = wH1;
         }
         paf[ks] = *reinterpret_cast<const bf16x8*>(fr);
       }
@@ -2564,12 +2575,14 @@ std::vector<torch::Tensor> fa_fwd(torch::Tensor q, torch::Tensor k, torch::Tenso
     tm = tile_map->data_ptr<uint8_t>();
   }
 
-  // 8-wave 32x32 ladder for dense/axial shapes (no static-mask support);
-  // DALLE_AMD_FA8=0 falls back to the 4-wave 16x16 kernel
-  static const bool fa8_on = []() {
-    const char* e = getenv("DALLE_AMD_FA8");
-    return e == nullptr || e[0] != '0';
-  }();
+  // 8-wave 32x32 ladder (DALLE_AMD_FA8=1 opt-in). Measured on MI355X at the
+  // flagship shape it LOSES to the 4-wave 16x16 kernel (dense 819 vs 722 us,
+  // axial 835 vs 502): at D=64 the ladder has half the MFMA work per softmax
+  // op of the guide's D=128 recipe and PMC shows it VALU-bound (42 VALU
+  // instrs per MFMA), while its 256-row blocks waste 7/8 of the staged grid
+  // lines under axial patterns. Kept correct + tested for D=128-class heads.
+  const char* fa8_env = getenv("DALLE_AMD_FA8");
+  const bool fa8_on = fa8_env != nullptr && fa8_env[0] == '1';
   if (fa8_on && sm == nullptr && tm == nullptr && nq >= 64) {
     dim3 grid8(((nq + FA8_QBLK - 1) / FA8_QBLK) * b * h);
     hipLaunchKernelGGL(fa8_fwd_d64_kernel, grid8, dim3(512), 0, cur_stream(),

@@ -449,6 +449,46 @@ def test_fa_fully_masked_rows_no_nan(ext):
         assert (g[1].float() == 0).all(), 'masked batch leaked gradient'
 
 
+def test_fa8_ladder_vs_oracle(ext, monkeypatch):
+    """Opt-in 8-wave 32x32 ladder forward vs the fp32 oracle (dense causal,
+    non-causal, key-masked, axial) — kept correct although the 16x16 kernel
+    is faster at D=64 (see the dispatch comment in hip_ops.hip)."""
+    monkeypatch.setenv('DALLE_AMD_FA8', '1')
+    torch.manual_seed(21)
+    for nq, nk, causal in ((256, 256, True), (1280, 1280, True),
+                           (192, 320, False), (300, 300, True)):
+        q = torch.randn(2, 3, nq, 64, device='cuda').bfloat16()
+        k = torch.randn(2, 3, nk, 64, device='cuda').bfloat16()
+        v = torch.randn(2, 3, nk, 64, device='cuda').bfloat16()
+        out, lse = ext.fa_fwd(q, k, v, 0.125, causal, None, None, None, False)
+        ref = fp32_oracle(q.float(), k.float(), v.float(), 0.125, causal)
+        err = (out.float() - ref).abs().max().item()
+        assert err < 0.02, (nq, nk, causal, err)
+    # key mask
+    km = torch.ones(2, 320, dtype=torch.bool, device='cuda')
+    km[:, 5:40] = False
+    q = torch.randn(2, 2, 320, 64, device='cuda').bfloat16()
+    k = torch.randn(2, 2, 320, 64, device='cuda').bfloat16()
+    v = torch.randn(2, 2, 320, 64, device='cuda').bfloat16()
+    out, _ = ext.fa_fwd(q, k, v, 0.125, True, km, None, None, False)
+    ref = fp32_oracle(q.float(), k.float(), v.float(), 0.125, True, km)
+    assert (out.float() - ref).abs().max().item() < 0.02
+    # axial mode (S=16, t=65, n=321)
+    from dalle_pytorch_amd.models.attention import axial_mask
+    S, t = 16, 65
+    n = t + S * S
+    q = torch.randn(1, 2, n, 64, device='cuda').bfloat16()
+    k = torch.randn(1, 2, n, 64, device='cuda').bfloat16()
+    v = torch.randn(1, 2, n, 64, device='cuda').bfloat16()
+    for axis in (0, 1):
+        out, _ = ext.fa_fwd(q, k, v, 0.125, True, None, None, None, False,
+                            t, S, axis)
+        sm = axial_mask(n, t, S, axis).cuda()
+        ref = fp32_oracle(q.float(), k.float(), v.float(), 0.125, True,
+                          None, sm)
+        assert (out.float() - ref).abs().max().item() < 0.02, axis
+
+
 def test_permlane_semantics(ext):
     """Pin v_permlane32_swap_b32: with (a, b) the results must be
     r0 = [a_lo | b_lo-from-partner...] — concretely, the fa8 kernel's
