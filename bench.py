@@ -51,6 +51,8 @@ def parse_args():
     p.add_argument('--gen_batch', type=int, default=64)
     p.add_argument('--no_graph', action='store_true',
                    help='disable HIP-graph decode replay (for kernel profiling)')
+    p.add_argument('--fp8', action='store_true',
+                   help='e4m3 forward path for the projection GEMMs')
     p.add_argument('--eager', action='store_true',
                    help='force eager ops on GPU (baseline comparison)')
     return p.parse_args()
@@ -95,6 +97,10 @@ def main():
         os.environ['DALLE_AMD_ALLOW_EAGER'] = '1'
         import dalle_pytorch_amd.ops.dispatch as _d
         _d._HIP, _d._TRIED = None, True
+
+    if args.fp8:
+        from dalle_pytorch_amd.ops.fp8 import set_fp8_enabled
+        set_fp8_enabled(True)
 
     from dalle_pytorch_amd.parallel import init_distributed, barrier
     from dalle_pytorch_amd.parallel import DataParallelEngine
@@ -201,7 +207,7 @@ def main():
             'n_gpus': world, 'steps': args.steps, 'warmup': args.warmup,
             'ms_per_step': round(ms_per_step, 2),
             'higher_is_better': True, 'scaling': 'weak',
-            'vs_baseline': None, 'dtype': 'bf16' if use_cuda else 'fp32',
+            'vs_baseline': None, 'dtype': 'fp8-fwd/bf16' if args.fp8 else 'bf16' if use_cuda else 'fp32',
             'data': 'synthetic',
             'config': {
                 'model': f"dalle-dim{cfg['dim']}-depth{cfg['depth']}-heads{cfg['heads']}",

@@ -31,6 +31,7 @@ from torch import nn
 
 from dalle_pytorch_amd.models.positional import apply_rotary_to_qkv
 from dalle_pytorch_amd.ops import attention_core, axial_attention
+from dalle_pytorch_amd.ops.fp8 import fp8_linear
 from dalle_pytorch_amd.ops.attention import build_tile_map
 from dalle_pytorch_amd.ops.rope import rope_split, rope_split_supported, trig_tables
 
@@ -44,7 +45,7 @@ def _qkv_heads(x, to_qkv, h, dim_head, rotary_pos_emb, offset=0):
     """Project to q/k/v [b,h,n,d] with rotary applied to q,k,v — fused on
     GPU (rope_split), eager elsewhere. rotary_pos_emb is the angle table
     [1, N, rot], consumed from position ``offset``."""
-    qkv = to_qkv(x)
+    qkv = fp8_linear(to_qkv, x)
     n = x.shape[1]
     if rope_split_supported(qkv, dim_head):
         cos = sin = None

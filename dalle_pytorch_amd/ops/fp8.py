@@ -1,0 +1,107 @@
+"""fp8 (e4m3) forward path for the projection GEMMs.
+
+gfx950 runs e4m3 MFMA at 2x the bf16 rate; measured on MI355X the raw
+ff1-shaped GEMM goes 1297 -> 2134 TF/s (scripts/probe_fp8.py). Master
+weights stay bf16/fp32 — fp8 exists only on the wire into the GEMM:
+
+* activations: one torch amax reduce + the one-pass quant_fp8 HIP kernel
+  (the eager cast chain costs more than the fp8 GEMM saves);
+* weights: quantized once per optimizer step (cached on the parameter's
+  version counter, so the reversible recompute pass reuses it);
+* backward: standard bf16 dgrad/wgrad — fp8 is forward-only, which under
+  reversible execution still covers 2 of the 4 GEMM passes per step.
+
+Enable with DALLE_AMD_FP8=1 (or ``set_fp8_enabled(True)``; bench.py
+exposes ``--fp8``). Default OFF: the headline BASELINE numbers are bf16.
+"""
+
+import os
+
+import torch
+import torch.nn.functional as F
+
+from dalle_pytorch_amd.ops.dispatch import hip_module, using_eager_fallback
+
+_FORCED = None
+FP8_MAX = 448.0
+
+
+def set_fp8_enabled(flag):
+    global _FORCED
+    _FORCED = bool(flag) if flag is not None else None
+
+
+def fp8_enabled():
+    if _FORCED is not None:
+        return _FORCED
+    return os.environ.get('DALLE_AMD_FP8', '0') == '1'
+
+
+_w_cache = {}
+
+
+def _quant(t, ext):
+    amax = t.detach().abs().amax().float()
+    return ext.quant_fp8(t.detach().contiguous(), amax), amax / FP8_MAX
+
+
+def _quant_weight(w, ext):
+    """Per-step cache: the reversible recompute re-runs every forward with
+    unchanged weights, so quantize once per (param, version)."""
+    key = id(w)
+    hit = _w_cache.get(key)
+    if hit is not None and hit[0] == w._version:
+        return hit[1], hit[2]
+    wb = w.detach()
+    if wb.dtype != torch.bfloat16:
+        wb = wb.to(torch.bfloat16)
+    wq, ws = _quant(wb, ext)
+    _w_cache[key] = (w._version, wq, ws)
+    return wq, ws
+
+
+class _Fp8LinearFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, weight, bias):
+        ext = hip_module()
+        shape = x.shape
+        x2 = x.reshape(-1, shape[-1])
+        xq, xs = _quant(x2, ext)
+        wq, ws = _quant_weight(weight, ext)
+        out = torch._scaled_mm(xq, wq.t(), scale_a=xs, scale_b=ws,
+                               bias=bias, out_dtype=torch.bfloat16)
+        ctx.save_for_backward(x2, weight)
+        ctx.has_bias = bias is not None
+        return out.reshape(*shape[:-1], weight.shape[0])
+
+    @staticmethod
+    def backward(ctx, dout):
+        x2, weight = ctx.saved_tensors
+        do2 = dout.reshape(-1, dout.shape[-1]).contiguous()
+        dx = (do2 @ weight.to(do2.dtype)).reshape(
+            *dout.shape[:-1], weight.shape[1])
+        dw = do2.t() @ x2.to(do2.dtype)
+        db = do2.sum(0) if ctx.has_bias else None
+        return dx, dw.to(weight.dtype), db
+
+
+def _fp8_ok(x, weight):
+    if not fp8_enabled() or using_eager_fallback(x):
+        return False
+    if x.dtype != torch.bfloat16:
+        return False
+    rows = x.numel() // x.shape[-1]
+    k, n = weight.shape[1], weight.shape[0]
+    return rows % 16 == 0 and k % 16 == 0 and n % 16 == 0 and rows >= 256
+
+
+def fp8_linear(linear_module, x):
+    """F.linear through the fp8 forward path when profitable, else the
+    module itself. Drop-in for ``linear_module(x)``."""
+    w = linear_module.weight
+    if _fp8_ok(x, w):
+        bias = linear_module.bias
+        if bias is not None and bias.dtype != torch.bfloat16:
+            bias = bias.to(torch.bfloat16)
+        return _Fp8LinearFn.apply(x, w, bias)
+    return linear_module(x)

@@ -1316,6 +1316,39 @@ void fa_bwd_dkv_kernel(
 
 
 // ---------------------------------------------------------------------------
+// bf16 -> fp8(e4m3, OCP) quantization for the fp8 linear path: one pass,
+// scale read from a device scalar (amax/448 computed by a torch reduce).
+// The eager chain (float cast, div, clamp, to(fp8)) is ~5 full-tensor
+// passes and costs more than the fp8 GEMM saves (measured probe_fp8).
+// ---------------------------------------------------------------------------
+
+__global__ __launch_bounds__(256)
+void quant_fp8_kernel(const short* __restrict__ x,      // [n] bf16
+                      const float* __restrict__ amax,   // [1]
+                      unsigned char* __restrict__ out,  // [n] e4m3
+                      long n) {
+  const float inv = 448.f / fmaxf(*amax, 1e-12f);
+  const long i0 = ((long)blockIdx.x * 256 + threadIdx.x) * 8;
+  if (i0 + 7 < n) {
+    int4v v = *reinterpret_cast<const int4v*>(x + i0);
+    const short* vs = reinterpret_cast<const short*>(&v);
+    unsigned short o4[4];
+    #pragma unroll
+    for (int p = 0; p < 4; ++p) {
+      const float a = fminf(fmaxf(bf2f(vs[2 * p]) * inv, -448.f), 448.f);
+      const float b_ = fminf(fmaxf(bf2f(vs[2 * p + 1]) * inv, -448.f), 448.f);
+      o4[p] = (unsigned short)__builtin_amdgcn_cvt_pk_fp8_f32(a, b_, 0, false);
+    }
+    *reinterpret_cast<int2*>(out + i0) = *reinterpret_cast<const int2*>(o4);
+  } else {
+    for (long i = i0; i < n; ++i) {
+      const float a = fminf(fmaxf(bf2f(x[i]) * inv, -448.f), 448.f);
+      out[i] = (unsigned char)__builtin_amdgcn_cvt_pk_fp8_f32(a, 0.f, 0, false);
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
 // Skinny-M GEMM: out[M, N] = x[M, K] @ W[N, K]^T (+ bias), M <= 128.
 //
 // The decode step's projections are M = batch (64..128) against multi-MB
@@ -3036,6 +3069,19 @@ std::vector<torch::Tensor> dec_prelude(
   return {xn, z};
 }
 
+torch::Tensor quant_fp8(torch::Tensor x, torch::Tensor amax) {
+  CHK(x.is_cuda() && x.dtype() == torch::kBFloat16 && x.is_contiguous());
+  CHK(amax.dtype() == torch::kFloat32);
+  const long n = x.numel();
+  auto out = torch::empty_like(x, x.options().dtype(torch::kFloat8_e4m3fn));
+  hipLaunchKernelGGL(quant_fp8_kernel, dim3((n / 8 + 255) / 256 + 1),
+                     dim3(256), 0, cur_stream(),
+                     reinterpret_cast<const short*>(x.data_ptr()),
+                     amax.data_ptr<float>(),
+                     reinterpret_cast<unsigned char*>(out.data_ptr()), n);
+  return out;
+}
+
 torch::Tensor skinny_gemm(torch::Tensor x, torch::Tensor w,
                           std::optional<torch::Tensor> bias) {
   CHK(x.is_cuda() && x.dtype() == torch::kBFloat16 && x.is_contiguous());
@@ -3116,6 +3162,7 @@ std::vector<torch::Tensor> resls_bwd(torch::Tensor dout, torch::Tensor y,
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("resls_fwd", &resls_fwd, "fused residual + per-channel scale fwd");
   m.def("resls_bwd", &resls_bwd, "fused residual + per-channel scale bwd");
+  m.def("quant_fp8", &quant_fp8, "bf16 -> e4m3 one-pass quantize");
   m.def("skinny_gemm", &skinny_gemm,
         "skinny-M weights-streaming GEMM (decode projections)",
         py::arg("x"), py::arg("w"), py::arg("bias") = std::nullopt);

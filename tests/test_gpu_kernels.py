@@ -449,6 +449,63 @@ def test_fa_fully_masked_rows_no_nan(ext):
         assert (g[1].float() == 0).all(), 'masked batch leaked gradient'
 
 
+def test_fp8_linear_vs_bf16(ext, monkeypatch):
+    """fp8 forward path: numerics close to bf16, gradients match the bf16
+    master-weight gradients in direction, and a short training run's loss
+    curve tracks the bf16 run (VERDICT item 6 sanity bar)."""
+    from dalle_pytorch_amd.ops import fp8 as fp8_mod
+    torch.manual_seed(31)
+    lin = torch.nn.Linear(1024, 4096).cuda().bfloat16()
+    x = torch.randn(512, 1024, device='cuda', dtype=torch.bfloat16,
+                    requires_grad=True)
+    ref = lin(x)
+    gref = torch.autograd.grad(ref.float().square().sum(), (x, lin.weight))
+
+    monkeypatch.setenv('DALLE_AMD_FP8', '1')
+    out = fp8_mod.fp8_linear(lin, x)
+    g = torch.autograd.grad(out.float().square().sum(), (x, lin.weight))
+    rel = (out.float() - ref.float()).abs().mean().item() / \
+        ref.float().abs().mean().item()
+    assert rel < 0.08, rel   # e4m3 quantization noise bound
+    for a, b in zip(g, gref):
+        cos = torch.nn.functional.cosine_similarity(
+            a.float().flatten(), b.float().flatten(), dim=0).item()
+        assert cos > 0.98, cos
+
+    # loss-curve sanity: tiny flagship-shaped model, 8 Adam steps
+    from dalle_pytorch_amd import DALLE, DiscreteVAE
+
+    def run_losses():
+        torch.manual_seed(5)
+        vae = DiscreteVAE(image_size=64, num_layers=2, num_tokens=64,
+                          codebook_dim=64, hidden_dim=16).cuda()
+        d = DALLE(dim=1024, vae=vae, num_text_tokens=200, text_seq_len=16,
+                  depth=2, heads=16, dim_head=64,
+                  attn_types=('axial_row',), shift_tokens=True).cuda()
+        opt = torch.optim.Adam([p for p in d.parameters() if p.requires_grad],
+                               lr=3e-4)
+        torch.manual_seed(7)
+        text = torch.randint(1, 200, (16, 16), device='cuda')
+        imgs = torch.rand(16, 3, 64, 64, device='cuda')
+        losses = []
+        for _ in range(8):
+            with torch.autocast('cuda', dtype=torch.bfloat16):
+                loss = d(text, imgs, return_loss=True)
+            opt.zero_grad()
+            loss.backward()
+            opt.step()
+            losses.append(loss.item())
+        return losses
+
+    fp8_losses = run_losses()
+    monkeypatch.setenv('DALLE_AMD_FP8', '0')
+    bf16_losses = run_losses()
+    # same trajectory within quantization noise; both must actually descend
+    assert fp8_losses[-1] < fp8_losses[0]
+    for a, b in zip(fp8_losses, bf16_losses):
+        assert abs(a - b) / abs(b) < 0.05, (fp8_losses, bf16_losses)
+
+
 def test_fa8_ladder_vs_oracle(ext, monkeypatch):
     """Opt-in 8-wave 32x32 ladder forward vs the fp32 oracle (dense causal,
     non-causal, key-masked, axial) — kept correct although the 16x16 kernel
