@@ -149,6 +149,9 @@ def main():
         engine.clip_grad_norm_(0.5)
         opt.step()
         engine.zero_grad()
+        if args.fp8:
+            from dalle_pytorch_amd.ops.fp8 import fp8_mark_step
+            fp8_mark_step()
         return loss
 
     decoder = None

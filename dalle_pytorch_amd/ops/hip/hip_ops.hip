@@ -1323,6 +1323,32 @@ void fa_bwd_dkv_kernel(
 // ---------------------------------------------------------------------------
 
 __global__ __launch_bounds__(256)
+void amax_bf16_kernel(const short* __restrict__ x, float* __restrict__ out,
+                      long n) {
+  // |max| over bf16: one pass, wave reduce, one atomicMax per wave (positive
+  // floats compare as their uint bit patterns; out is pre-zeroed)
+  float m = 0.f;
+  for (long i = ((long)blockIdx.x * 256 + threadIdx.x) * 8; i + 7 < n;
+       i += (long)gridDim.x * 256 * 8) {
+    int4v v = *reinterpret_cast<const int4v*>(x + i);
+    const short* vs = reinterpret_cast<const short*>(&v);
+    #pragma unroll
+    for (int e = 0; e < 8; ++e) m = fmaxf(m, fabsf(bf2f(vs[e])));
+  }
+  if (blockIdx.x == 0 && threadIdx.x == 0) {
+    const long tail = (n / 8) * 8;
+    for (long i = tail; i < n; ++i) m = fmaxf(m, fabsf(bf2f(x[i])));
+  }
+  #pragma unroll
+  for (int s = 32; s > 0; s >>= 1) m = fmaxf(m, __shfl_xor(m, s));
+  if ((threadIdx.x & 63) == 0) {
+    union { float f; unsigned u; } c;
+    c.f = m;
+    atomicMax(reinterpret_cast<unsigned*>(out), c.u);
+  }
+}
+
+__global__ __launch_bounds__(256)
 void quant_fp8_kernel(const short* __restrict__ x,      // [n] bf16
                       const float* __restrict__ amax,   // [1]
                       unsigned char* __restrict__ out,  // [n] e4m3
@@ -3069,6 +3095,17 @@ std::vector<torch::Tensor> dec_prelude(
   return {xn, z};
 }
 
+torch::Tensor amax_bf16(torch::Tensor x) {
+  CHK(x.is_cuda() && x.dtype() == torch::kBFloat16 && x.is_contiguous());
+  const long n = x.numel();
+  auto out = torch::zeros({1}, x.options().dtype(torch::kFloat32));
+  const long nb = std::min<long>((n / 8 + 255) / 256 + 1, 4096);
+  hipLaunchKernelGGL(amax_bf16_kernel, dim3(nb), dim3(256), 0, cur_stream(),
+                     reinterpret_cast<const short*>(x.data_ptr()),
+                     out.data_ptr<float>(), n);
+  return out;
+}
+
 torch::Tensor quant_fp8(torch::Tensor x, torch::Tensor amax) {
   CHK(x.is_cuda() && x.dtype() == torch::kBFloat16 && x.is_contiguous());
   CHK(amax.dtype() == torch::kFloat32);
@@ -3163,6 +3200,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("resls_fwd", &resls_fwd, "fused residual + per-channel scale fwd");
   m.def("resls_bwd", &resls_bwd, "fused residual + per-channel scale bwd");
   m.def("quant_fp8", &quant_fp8, "bf16 -> e4m3 one-pass quantize");
+  m.def("amax_bf16", &amax_bf16, "abs-max of a bf16 tensor (one pass)");
   m.def("skinny_gemm", &skinny_gemm,
         "skinny-M weights-streaming GEMM (decode projections)",
         py::arg("x"), py::arg("w"), py::arg("bias") = std::nullopt);

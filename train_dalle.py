@@ -131,6 +131,7 @@ def get_tokenizer(args):
 def main(argv=None):
     args = parse_args(argv)
     from dalle_pytorch_amd.utils.tunable import maybe_enable_tunableop
+    from dalle_pytorch_amd.ops.fp8 import fp8_mark_step as _fp8_mark_step
     maybe_enable_tunableop()
     rank, world, local_rank = init_distributed()
     is_root = rank == 0
@@ -307,6 +308,7 @@ def main(argv=None):
                 if args.clip_grad_norm:
                     engine.clip_grad_norm_(args.clip_grad_norm)
                 opt.step()
+                _fp8_mark_step()
                 engine.zero_grad()
 
             avg_loss = average_scalar(loss)
@@ -345,6 +347,7 @@ def main(argv=None):
                 ploss.backward()
                 engine.finish_gradient_sync()
                 opt.step()
+                _fp8_mark_step()
                 engine.zero_grad()
                 if device.type == 'cuda':
                     torch.cuda.synchronize()
