@@ -1358,17 +1358,17 @@ void amax_bf16_kernel(const short* __restrict__ x, float* __restrict__ out,
   for (int s = 32; s > 0; s >>= 1) m = fmaxf(m, __shfl_xor(m, s));
   if ((threadIdx.x & 63) == 0) {
     union { float f; unsigned u; } c;
-    c.f = m;
-    atomicMax(reinterpret_cast<unsigned*>(out), c.u);
+    c.f = fmaxf(m, 1e-12f) * (1.f / 448.f);   // scale, not amax: saves the
+    atomicMax(reinterpret_cast<unsigned*>(out), c.u);  // per-call div kernel
   }
 }
 
 __global__ __launch_bounds__(256)
 void quant_fp8_kernel(const short* __restrict__ x,      // [n] bf16
-                      const float* __restrict__ amax,   // [1]
+                      const float* __restrict__ scale,  // [1] = amax/448
                       unsigned char* __restrict__ out,  // [n] e4m3
                       long n) {
-  const float inv = 448.f / fmaxf(*amax, 1e-12f);
+  const float inv = 1.f / fmaxf(*scale, 1e-30f);
   const long i0 = ((long)blockIdx.x * 256 + threadIdx.x) * 8;
   if (i0 + 7 < n) {
     int4v v = *reinterpret_cast<const int4v*>(x + i0);
