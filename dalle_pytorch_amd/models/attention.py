@@ -204,7 +204,7 @@ class Attention(nn.Module):
             causal=self.causal and offset == 0,
             key_mask=mask, static_mask=static, static_tiles=tiles,
             static_tiles_t=tiles_t, fold_heads=True)
-        return self.to_out(out)
+        return self.to_out[1](fp8_linear(self.to_out[0], out))
 
 
 class _StaticMaskSparseAttention(nn.Module):
@@ -253,7 +253,7 @@ class _StaticMaskSparseAttention(nn.Module):
                              key_mask=km, static_mask=static,
                              static_tiles=tiles, static_tiles_t=tiles_t,
                              fold_heads=True)
-        return self.to_out(out)
+        return self.to_out[1](fp8_linear(self.to_out[0], out))
 
 
 class SparseAxialCausalAttention(_StaticMaskSparseAttention):
@@ -304,7 +304,7 @@ class SparseAxialCausalAttention(_StaticMaskSparseAttention):
         km = self._key_mask(mask, b, n, t, x.device)
         out = axial_attention(q, k, v, self.scale, t, self.image_size,
                               self.axis, key_mask=km)
-        return self.to_out(out)
+        return self.to_out[1](fp8_linear(self.to_out[0], out))
 
 
 class SparseConvCausalAttention(_StaticMaskSparseAttention):
