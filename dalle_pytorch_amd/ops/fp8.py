@@ -105,11 +105,12 @@ def _fp8_ok(x, weight):
     k, n = weight.shape[1], weight.shape[0]
     if rows % 16 or k % 16 or n % 16 or rows < 256:
         return False
-    # profitability: quantizing the input costs ~rows*k traffic, the GEMM
-    # saves ~35%% of its time; measured per-shape on MI355X (probe_fp8b) the
-    # win holds down to N=K (ff2's K=4096 input quant is the worst case and
-    # still nets ~+35 us after the amax fix)
-    return n * 4 >= k
+    # profitability gate, measured per-shape on MI355X (probe_fp8b + end-to-
+    # end A/B): ff1 (N=8192,K=1024) +375 us/call and qkv (N=3072) +123 win;
+    # ff2 (K=4096) and the square out-proj are net NEUTRAL in isolation and
+    # LOSE ~1%% end-to-end (launch-gap overhead of the extra quant kernels),
+    # so only clearly-profitable shapes pass
+    return n * 2 >= k * 3
 
 
 def fp8_linear(linear_module, x):
