@@ -33,6 +33,11 @@ CONFIGS = {
                    reversible=True, batch_size=64, vae='dvae'),
     'd': dict(dim=1024, depth=64, heads=16, attn_types=('axial_row', 'axial_col'),
               reversible=True, batch_size=4, vae='vqgan16k'),
+    # the other two sparse attention families at the flagship shape
+    'c_conv': dict(dim=1024, depth=12, heads=16, attn_types=('conv_like',),
+                   reversible=True, batch_size=64, vae='dvae'),
+    'c_sparse': dict(dim=1024, depth=12, heads=16, attn_types=('sparse',),
+                     reversible=True, batch_size=64, vae='dvae'),
     # plumbing smoke (CPU-runnable, used by the distributed-launch test)
     'tiny': dict(dim=64, depth=1, heads=1, attn_types=('full',),
                  reversible=False, batch_size=2, vae='dvae',
