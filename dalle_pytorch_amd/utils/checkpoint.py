@@ -62,6 +62,13 @@ def build_dalle_from_checkpoint(ckpt, vae=None, strict=True, extra_hparams=None)
     hparams = dict(ckpt['hparams'])
     hparams.update(extra_hparams or {})
     dalle = DALLE(vae=vae, **hparams)
+    # adopt the checkpoint's storage dtype (a reference --fp16 run saves
+    # fp16 weights via dalle.half(), train_dalle.py:430-432) — plain
+    # load_state_dict would silently cast them up to the fresh model's fp32
+    float_dtypes = {t.dtype for t in ckpt['weights'].values()
+                    if torch.is_tensor(t) and t.is_floating_point()}
+    if len(float_dtypes) == 1:
+        dalle = dalle.to(next(iter(float_dtypes)))
     dalle.load_state_dict(ckpt['weights'], strict=strict)
     return dalle, vae
 

@@ -56,6 +56,12 @@ def parse_args(argv=None):
     p.add_argument('--fp16', action='store_true',
                    help='mixed precision (bf16 autocast on MI355X)')
     p.add_argument('--amp', action='store_true', help='alias of --fp16 here')
+    p.add_argument('--half_weights', choices=['fp16', 'bf16'], default=None,
+                   help='store the model weights in half precision, as the '
+                        'reference --fp16 does via dalle.half() '
+                        '(train_dalle.py:430-432); bf16 keeps the HIP kernel '
+                        'path, fp16 matches reference checkpoint dtype '
+                        'exactly (runs on torch fallbacks)')
     p.add_argument('--wandb_name', default='dalle_train_transformer')
     p.add_argument('--wandb_entity', default=None)
     p.add_argument('--name_suffix', default='')
@@ -200,6 +206,11 @@ def main(argv=None):
     if resume_ckpt is not None:
         dalle.load_state_dict(resume_ckpt['weights'])
     dalle = dalle.to(device)
+    if args.half_weights:
+        # reference --fp16 semantics: the weights themselves are halved
+        # (reference train_dalle.py:430-432 does dalle.half())
+        dalle = dalle.to(torch.float16 if args.half_weights == 'fp16'
+                         else torch.bfloat16)
 
     # --------------------------------------------------------- dataset
     text_seq_len = dalle_params['text_seq_len']  # checkpoint-authoritative on resume
@@ -268,7 +279,8 @@ def main(argv=None):
                 str(out_dir / 'trace.json')))
         profiler.__enter__()
 
-    autocast_enabled = (args.fp16 or args.amp) and device.type == 'cuda'
+    autocast_enabled = ((args.fp16 or args.amp) and device.type == 'cuda'
+                        and not args.half_weights)
 
     def save(epoch):
         if not is_root:
