@@ -563,26 +563,30 @@ class FastDecoder:
         try:
             ntt, nit = d.num_text_tokens, d.num_image_tokens
             logits = self.prefill(text)[:, ntt:ntt + nit]
-            step_fn = self._graph_step if self.use_graph else self.step
             # reference top-k semantics: k is a fraction of the FULL vocab
             k = max(int((1 - filter_thres) * d.total_tokens), 1)
             k = min(k, nit)
-            out_tokens = []
-            for i in range(d.image_seq_len):
-                li = logits.float()
-                if guided:
-                    cond, null = li[:nb], li[nb:]
-                    li = null + (cond - null) * cond_scale
-                vals, idx = li.topk(k, dim=-1)
-                filtered = torch.full_like(li, -torch.finfo(li.dtype).max)
-                filtered.scatter_(1, idx, vals)
-                token = gumbel_sample(filtered, temperature=temperature)
-                out_tokens.append(token)
-                if i + 1 < d.image_seq_len:
-                    feed = torch.cat((token, token), dim=0) if guided else token
-                    logits = step_fn(feed)
-            img_seq = torch.stack(out_tokens, dim=1)
-            images = d.vae.decode(img_seq)
+            self._out_buf = torch.empty(nb, d.image_seq_len, dtype=torch.long,
+                                        device=self.device)
+            self._gen_ptr = torch.zeros(1, dtype=torch.long, device=self.device)
+            sample_args = (nb, guided, cond_scale, k, temperature)
+
+            # token 0 from the prefill logits (outside any graph)
+            li = logits.float()
+            if guided:
+                cond, null = li[:nb], li[nb:]
+                li = null + (cond - null) * cond_scale
+            vals, idx = li.topk(k, dim=-1)
+            filtered = torch.full_like(li, -torch.finfo(li.dtype).max)
+            filtered.scatter_(1, idx, vals)
+            token = gumbel_sample(filtered, temperature=temperature)
+            self._out_buf.index_copy_(1, self._gen_ptr, token.unsqueeze(1))
+            self._gen_ptr += 1
+
+            step_fn = self._graph_token_step if self.use_graph else self._token_step
+            for _ in range(d.image_seq_len - 1):
+                token = step_fn(token, *sample_args)
+            images = d.vae.decode(self._out_buf)
         finally:
             self._img_head = False
             d.train(was_training)
@@ -592,10 +596,11 @@ class FastDecoder:
         return ([st.k.clone() if st.k is not None else None for st in self.states],
                 [st.v.clone() if st.v is not None else None for st in self.states],
                 [st.ring.clone() if st.ring is not None else None for st in self.states],
-                self.offset_t.clone())
+                self.offset_t.clone(),
+                self._gen_ptr.clone() if getattr(self, '_gen_ptr', None) is not None else None)
 
     def _restore(self, snap):
-        ks, vs, rings, off = snap
+        ks, vs, rings, off, ptr = snap
         for st, k, v, r in zip(self.states, ks, vs, rings):
             if k is not None:
                 st.k.copy_(k)
@@ -603,6 +608,47 @@ class FastDecoder:
             if r is not None:
                 st.ring.copy_(r)
         self.offset_t.copy_(off)
+        if ptr is not None:
+            self._gen_ptr.copy_(ptr)
+
+    # ------------------------------------------------- fully in-graph step
+
+    def _token_step(self, token, nb, guided, cond_scale, k, temperature):
+        """token ids [nb] -> next token ids [nb], sampling included — the
+        whole thing is graph-capturable (RNG advances via the graph-safe
+        philox state), so a generation step replays with ZERO eager kernels."""
+        feed = torch.cat((token, token), dim=0) if guided else token
+        logits = self.step(feed).float()
+        if guided:
+            cond, null = logits[:nb], logits[nb:]
+            logits = null + (cond - null) * cond_scale
+        vals, idx = logits.topk(k, dim=-1)
+        filtered = torch.full_like(logits, -torch.finfo(logits.dtype).max)
+        filtered.scatter_(1, idx, vals)
+        nxt = gumbel_sample(filtered, temperature=temperature)
+        self._out_buf.index_copy_(1, self._gen_ptr, nxt.unsqueeze(1))
+        self._gen_ptr += 1
+        return nxt
+
+    def _graph_token_step(self, token, *args):
+        if self._graph is None:
+            self._g_token = token.clone()
+            snap = self._snapshot()
+            s = torch.cuda.Stream()
+            s.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(s):
+                for _ in range(2):
+                    self._g_next = self._token_step(self._g_token, *args)
+            torch.cuda.current_stream().wait_stream(s)
+            self._restore(snap)
+            self._graph = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(self._graph):
+                self._g_next = self._token_step(self._g_token, *args)
+            self._graph.replay()
+            return self._g_next
+        self._g_token.copy_(token)
+        self._graph.replay()
+        return self._g_next
 
     def _graph_step(self, token):
         if self._graph is None:
