@@ -231,3 +231,6 @@ def main():
 
 if __name__ == '__main__':
     main()
+    import torch.distributed as dist
+    if dist.is_available() and dist.is_initialized():
+        dist.destroy_process_group()
