@@ -179,6 +179,10 @@ class DALLE(nn.Module):
 
     @torch.no_grad()
     def generate_texts(self, tokenizer, text=None, *, filter_thres=0.5, temperature=1.):
+        # accept either a tokenizer object or the reference's module-with-
+        # singleton convention (reference generate.py passes the module and
+        # dalle_pytorch.py:469 reads tokenizer.tokenizer)
+        tokenizer = getattr(tokenizer, 'tokenizer', tokenizer)
         was_training = self.training
         self.eval()
         device = next(self.parameters()).device
@@ -186,7 +190,7 @@ class DALLE(nn.Module):
             text_tokens = torch.tensor([[0]], device=device)
         else:
             text_tokens = torch.tensor(
-                tokenizer.tokenizer.encode(text), device=device).unsqueeze(0)
+                tokenizer.encode(text), device=device).unsqueeze(0)
 
         for _ in range(text_tokens.shape[1], self.text_seq_len):
             tokens = self.text_emb(text_tokens)
@@ -210,7 +214,7 @@ class DALLE(nn.Module):
         self.train(was_training)
         pad_tokens = set(range(self.num_text_tokens - self.text_seq_len,
                                self.num_text_tokens))
-        texts = [tokenizer.tokenizer.decode(t, pad_tokens=pad_tokens)
+        texts = [tokenizer.decode(t, pad_tokens=pad_tokens)
                  for t in text_tokens]
         return text_tokens, texts
 

@@ -60,6 +60,22 @@ def test_train_dalle_then_generate(tmp_path):
     outs = list((tmp_path / 'gen').glob('**/*'))
     assert any(p.suffix in ('.png', '.pt') for p in outs)
 
+    # classifier-free guidance on the fast path (doubled-batch decoder)
+    generate.main([
+        '--dalle_path', str(ckpt_path), '--text', 'a guided test',
+        '--num_images', '1', '--batch_size', '1', '--cond_scale', '2.0',
+        '--outputs_dir', str(tmp_path / 'gen_guided')])
+    assert any(p.suffix == '.png'
+               for p in (tmp_path / 'gen_guided').glob('**/*'))
+
+    # text completion first, then fast-path image decode
+    generate.main([
+        '--dalle_path', str(ckpt_path), '--text', 'a', '--gentxt',
+        '--num_images', '1', '--batch_size', '1',
+        '--outputs_dir', str(tmp_path / 'gen_txt')])
+    assert any(p.suffix == '.png'
+               for p in (tmp_path / 'gen_txt').glob('**/*'))
+
 
 def test_train_dalle_grad_accum(tmp_path):
     import train_vae
