@@ -166,6 +166,10 @@ void fa_fwd_d64_kernel(
   // Q fragments (B-operand of the swapped QK^T):
   // lane holds Q[lq][8*grp + e + 32*c], c = 0,1
   const int qphys = axial ? ax_phys(qrow, ax_t, ax_logS, ax_axis) : qrow;
+  // first key of this q row's own grid line (INT_MAX for text rows): the
+  // axial element test reduces to 4 branchless compares
+  const int ax_klo = (axial && qrow < nq && qrow >= ax_t)
+      ? ax_t + (((qrow - ax_t) >> ax_logS) << ax_logS) : 0x7fffffff;
   bf16x8 qfrag[2];
   {
     const bool qok = qrow < nq;
@@ -324,7 +328,8 @@ void fa_fwd_d64_kernel(
       #pragma unroll
       for (int i = 0; i < 16; ++i) {
         const int kg = kbase + (i >> 2) * 16 + grp * 4 + (i & 3);
-        bool ok = (kg < nk) & (qrow < nq) && ax_ok(qrow, kg, ax_t, ax_logS);
+        bool ok = (kg < nk) & (qrow < nq) & (kg <= qrow) &
+                  ((kg < ax_t) | (kg >= ax_klo));
         if (key_mask != nullptr && ok)
           ok &= key_mask[(long)batch * nk + ax_phys(kg, ax_t, ax_logS, ax_axis)];
         s16[i] = ok ? s16[i] * scale : NEG_INF;
@@ -814,6 +819,8 @@ void fa_bwd_dq_kernel(
 
   const bool axial = ax_axis >= 0;
   const int qphys = axial ? ax_phys(qrow, ax_t, ax_logS, ax_axis) : qrow;
+  const int ax_klo = (axial && qrow < nq && qrow >= ax_t)
+      ? ax_t + (((qrow - ax_t) >> ax_logS) << ax_logS) : 0x7fffffff;
   bf16x8 qfrag[2], dofrag[2];
   float lse_q = 0.f, D_q = 0.f;
   {
@@ -966,7 +973,8 @@ void fa_bwd_dq_kernel(
       #pragma unroll
       for (int i = 0; i < 16; ++i) {
         const int kg = kbase + (i >> 2) * 16 + grp * 4 + (i & 3);
-        bool ok = (kg < nk) & (qrow < nq) && ax_ok(qrow, kg, ax_t, ax_logS);
+        bool ok = (kg < nk) & (qrow < nq) & (kg <= qrow) &
+                  ((kg < ax_t) | (kg >= ax_klo));
         if (key_mask != nullptr && ok)
           ok &= key_mask[(long)batch * nk + ax_phys(kg, ax_t, ax_logS, ax_axis)];
         const float p = ok ? __expf(s16[i] * scale - lse_q) : 0.f;
@@ -1068,6 +1076,17 @@ void fa_bwd_dkv_kernel(
 
   const bool axial = ax_axis >= 0;
   const int kphys = axial ? ax_phys(krow, ax_t, ax_logS, ax_axis) : krow;
+  // per accumulator row r: one past the last q row that may attend key
+  // k0+wave*16+grp*4+r (text keys: everything causal; image keys: own line)
+  int ax_kend[4];
+  if (axial) {
+    #pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int key = k0 + wave * 16 + grp * 4 + r;
+      ax_kend[r] = key < ax_t ? nq
+          : min(nq, ax_t + ((((key - ax_t) >> ax_logS) + 1) << ax_logS));
+    }
+  }
   bf16x8 kfrag[2], vfrag[2];
   {
     const bool kok = krow < nk;
@@ -1238,7 +1257,7 @@ void fa_bwd_dkv_kernel(
         #pragma unroll
         for (int r = 0; r < 4; ++r) {
           const int key = k0 + wave * 16 + grp * 4 + r;
-          bool ok = (key < nk) & (qg < nq) && ax_ok(qg, key, ax_t, ax_logS);
+          bool ok = (key < nk) & (qg < nq) & (qg >= key) & (qg < ax_kend[r]);
           if (key_mask != nullptr && ok)
             ok &= key_mask[(long)batch * nk + ax_phys(key, ax_t, ax_logS, ax_axis)];
           float p = 0.f, ds = 0.f;
