@@ -383,8 +383,9 @@ class DALLE(nn.Module):
         ntt = self.num_text_tokens
         h = self.to_logits[0](out)
         head_w, head_b = self.to_logits[1].weight, self.to_logits[1].bias
-        logits_text = F.linear(h[:, :tlen], head_w[:ntt], head_b[:ntt])
-        logits_img = F.linear(h[:, tlen:], head_w[ntt:], head_b[ntt:])
+        from dalle_pytorch_amd.ops.fp8 import fp8_linear_raw
+        logits_text = fp8_linear_raw(h[:, :tlen], head_w[:ntt], head_b[:ntt])
+        logits_img = fp8_linear_raw(h[:, tlen:], head_w[ntt:], head_b[ntt:])
         loss_text = F.cross_entropy(
             logits_text.reshape(-1, ntt), labels[:, :tlen].reshape(-1))
         loss_img = F.cross_entropy(
