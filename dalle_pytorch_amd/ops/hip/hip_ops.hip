@@ -1950,45 +1950,67 @@ __global__ void geglu_bwd_kernel(const T* __restrict__ x,
 // pad/cat/fill kernels per call.
 // ---------------------------------------------------------------------------
 
-__global__ void token_shift_kernel(
-    const char* __restrict__ src, char* __restrict__ dst,
-    int n, int row_bytes, int text_len, int S, int backward) {
-  const int cpr = row_bytes / 16;
-  const long c = (long)blockIdx.x * blockDim.x + threadIdx.x;
-  if (c >= (long)n * cpr) return;
-  const int pos = c / cpr;
-  const int off = (int)(c - (long)pos * cpr) * 16;
-  const long row0 = (long)blockIdx.y * n * row_bytes;
-
-  const int half = row_bytes / 2;
-  const int quarter = row_bytes / 4;
-
-  int srcpos = pos;
-  bool zero = false;
+DEVFN void token_shift_src(int pos, int off, int n, int text_len, int S,
+                           int half, int quarter, int backward,
+                           int* srcpos, bool* zero) {
+  *srcpos = pos;
+  *zero = false;
   if (!backward) {
     if (pos < text_len) {
-      if (off < half) { srcpos = pos - 1; zero = pos == 0; }
+      if (off < half) { *srcpos = pos - 1; *zero = pos == 0; }
     } else {
       const int g = pos - text_len;
       const int gr = g / S, gc = g - gr * S;
-      if (off < quarter) { srcpos = pos - S; zero = gr == 0; }
-      else if (off < half) { srcpos = pos - 1; zero = gc == 0; }
+      if (off < quarter) { *srcpos = pos - S; *zero = gr == 0; }
+      else if (off < half) { *srcpos = pos - 1; *zero = gc == 0; }
     }
   } else {
     if (pos < text_len) {
-      if (off < half) { srcpos = pos + 1; zero = pos + 1 >= text_len; }
+      if (off < half) { *srcpos = pos + 1; *zero = pos + 1 >= text_len; }
     } else {
       const int g = pos - text_len;
       const int gr = g / S, gc = g - gr * S;
-      if (off < quarter) { srcpos = pos + S; zero = gr + 1 >= S || pos + S >= n; }
-      else if (off < half) { srcpos = pos + 1; zero = gc + 1 >= S || pos + 1 >= n; }
+      if (off < quarter) { *srcpos = pos + S; *zero = gr + 1 >= S || pos + S >= n; }
+      else if (off < half) { *srcpos = pos + 1; *zero = gc + 1 >= S || pos + 1 >= n; }
     }
   }
+}
 
-  int4v val{0, 0, 0, 0};
-  if (!zero)
-    val = *reinterpret_cast<const int4v*>(src + row0 + (long)srcpos * row_bytes + off);
-  *reinterpret_cast<int4v*>(dst + row0 + (long)pos * row_bytes + off) = val;
+__global__ void token_shift_kernel(
+    const char* __restrict__ src, char* __restrict__ dst,
+    int n, int row_bytes, int text_len, int S, int backward) {
+  // 4 chunks per thread with the gather loads batched ahead of the stores
+  const int cpr = row_bytes / 16;
+  const long total = (long)n * cpr;
+  const long stride = (long)gridDim.x * blockDim.x;
+  const long row0 = (long)blockIdx.y * n * row_bytes;
+  const int half = row_bytes / 2;
+  const int quarter = row_bytes / 4;
+
+  long cc[4];
+  int4v val[4];
+  #pragma unroll
+  for (int u = 0; u < 4; ++u) {
+    cc[u] = (long)blockIdx.x * blockDim.x + threadIdx.x + u * stride;
+    val[u] = int4v{0, 0, 0, 0};
+    if (cc[u] < total) {
+      const int pos = (int)(cc[u] / cpr);
+      const int off = (int)(cc[u] - (long)pos * cpr) * 16;
+      int srcpos; bool zero;
+      token_shift_src(pos, off, n, text_len, S, half, quarter, backward,
+                      &srcpos, &zero);
+      if (!zero)
+        val[u] = *reinterpret_cast<const int4v*>(
+            src + row0 + (long)srcpos * row_bytes + off);
+    }
+  }
+  #pragma unroll
+  for (int u = 0; u < 4; ++u) {
+    if (cc[u] >= total) continue;
+    const int pos = (int)(cc[u] / cpr);
+    const int off = (int)(cc[u] - (long)pos * cpr) * 16;
+    *reinterpret_cast<int4v*>(dst + row0 + (long)pos * row_bytes + off) = val[u];
+  }
 }
 
 // ---------------------------------------------------------------------------
@@ -3028,7 +3050,7 @@ torch::Tensor token_shift(torch::Tensor x, int64_t text_len, int64_t image_size,
   CHK(row_bytes % 64 == 0);   // 16B chunks must not cross quarter bounds
   auto out = torch::empty_like(x);
   const long chunks = (long)n * (row_bytes / 16);
-  dim3 grid((chunks + 255) / 256, b);
+  dim3 grid((chunks + 1023) / 1024, b);   // 4 chunks per thread
   hipLaunchKernelGGL(token_shift_kernel, grid, dim3(256), 0, cur_stream(),
                      reinterpret_cast<const char*>(x.data_ptr()),
                      reinterpret_cast<char*>(out.data_ptr()),
