@@ -2514,50 +2514,61 @@ void ln_fwd_kernel(const short* __restrict__ x, const float* __restrict__ w,
                    const float* __restrict__ bia, short* __restrict__ y,
                    float* __restrict__ mean_out, float* __restrict__ rstd_out,
                    long rows, float eps) {
+  // one wave per TWO rows, both rows' loads issued before either row's
+  // compute: the row-serial form left HBM latency exposed (~3.9 TB/s)
   constexpr int per = PER;
   constexpr int dim = PER * 64;
-  const long row = blockIdx.x * 4 + (threadIdx.x >> 6);
+  const long row0 = ((long)blockIdx.x * 4 + (threadIdx.x >> 6)) * 2;
   const int lane = threadIdx.x & 63;
-  if (row >= rows) return;
-  const short* xr = x + row * (long)dim;
+  if (row0 >= rows) return;
+  const bool two = row0 + 1 < rows;
 
-  float sum = 0.f, sq = 0.f;
-  float vals[PER];
+  float vals[2][PER];
   #pragma unroll
-  for (int i = 0; i < per; i += 8) {
-    int4v v = *reinterpret_cast<const int4v*>(xr + lane * per + i);
-    const short* vs = reinterpret_cast<const short*>(&v);
+  for (int rr = 0; rr < 2; ++rr) {
+    if (rr && !two) break;
+    const short* xr = x + (row0 + rr) * (long)dim;
     #pragma unroll
-    for (int e = 0; e < 8; ++e) {
-      const float f = bf2f(vs[e]);
-      vals[i + e] = f;
-      sum += f;
-      sq += f * f;
+    for (int i = 0; i < per; i += 8) {
+      int4v v = *reinterpret_cast<const int4v*>(xr + lane * per + i);
+      const short* vs = reinterpret_cast<const short*>(&v);
+      #pragma unroll
+      for (int e = 0; e < 8; ++e) vals[rr][i + e] = bf2f(vs[e]);
     }
   }
   #pragma unroll
-  for (int sft = 32; sft > 0; sft >>= 1) {
-    sum += __shfl_xor(sum, sft);
-    sq += __shfl_xor(sq, sft);
-  }
-  const float mu = sum / dim;
-  const float var = sq / dim - mu * mu;
-  const float rstd = __frsqrt_rn(var + eps);
-  if (lane == 0) {
-    mean_out[row] = mu;
-    rstd_out[row] = rstd;
-  }
-  short* yr = y + row * (long)dim;
-  #pragma unroll
-  for (int i = 0; i < per; i += 8) {
-    short out8[8];
+  for (int rr = 0; rr < 2; ++rr) {
+    if (rr && !two) break;
+    float sum = 0.f, sq = 0.f;
     #pragma unroll
-    for (int e = 0; e < 8; ++e) {
-      const int d = lane * per + i + e;
-      out8[e] = f2bf((vals[i + e] - mu) * rstd * w[d] + bia[d]);
+    for (int i = 0; i < per; ++i) {
+      sum += vals[rr][i];
+      sq += vals[rr][i] * vals[rr][i];
     }
-    *reinterpret_cast<int4v*>(yr + lane * per + i) =
-        *reinterpret_cast<const int4v*>(out8);
+    #pragma unroll
+    for (int sft = 32; sft > 0; sft >>= 1) {
+      sum += __shfl_xor(sum, sft);
+      sq += __shfl_xor(sq, sft);
+    }
+    const float mu = sum / dim;
+    const float var = sq / dim - mu * mu;
+    const float rstd = __frsqrt_rn(var + eps);
+    if (lane == 0) {
+      mean_out[row0 + rr] = mu;
+      rstd_out[row0 + rr] = rstd;
+    }
+    short* yr = y + (row0 + rr) * (long)dim;
+    #pragma unroll
+    for (int i = 0; i < per; i += 8) {
+      short out8[8];
+      #pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        const int d = lane * per + i + e;
+        out8[e] = f2bf((vals[rr][i + e] - mu) * rstd * w[d] + bia[d]);
+      }
+      *reinterpret_cast<int4v*>(yr + lane * per + i) =
+          *reinterpret_cast<const int4v*>(out8);
+    }
   }
 }
 
@@ -3083,7 +3094,7 @@ std::vector<torch::Tensor> ln_fwd(torch::Tensor x, torch::Tensor w,
   auto rstd = torch::empty_like(mean);
   auto wf = w.to(torch::kFloat32).contiguous();
   auto bf_ = b.to(torch::kFloat32).contiguous();
-  dim3 grid((rows + 3) / 4);
+  dim3 grid((rows + 7) / 8);   // 4 waves x 2 rows per block
   auto launch = [&](auto kern) {
     hipLaunchKernelGGL(kern, grid, dim3(256), 0, cur_stream(),
                        reinterpret_cast<const short*>(x.data_ptr()),
