@@ -622,10 +622,16 @@ class FastDecoder:
         if guided:
             cond, null = logits[:nb], logits[nb:]
             logits = null + (cond - null) * cond_scale
-        vals, idx = logits.topk(k, dim=-1)
-        filtered = torch.full_like(logits, -torch.finfo(logits.dtype).max)
-        filtered.scatter_(1, idx, vals)
-        nxt = gumbel_sample(filtered, temperature=temperature)
+        if self._fused_decode:
+            from dalle_pytorch_amd.ops.dispatch import hip_module
+            noise = torch.rand_like(logits)   # graph-capture-safe philox
+            nxt = hip_module().sample_topk_gumbel(
+                logits.contiguous(), noise, k, max(temperature, 1e-10))
+        else:
+            vals, idx = logits.topk(k, dim=-1)
+            filtered = torch.full_like(logits, -torch.finfo(logits.dtype).max)
+            filtered.scatter_(1, idx, vals)
+            nxt = gumbel_sample(filtered, temperature=temperature)
         self._out_buf.index_copy_(1, self._gen_ptr, nxt.unsqueeze(1))
         self._gen_ptr += 1
         return nxt

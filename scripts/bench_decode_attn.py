@@ -34,6 +34,9 @@ def timeit(fn, iters=300):
     torch.cuda.synchronize()
     return (time.perf_counter() - t0)/iters*1e6
 
-us = timeit(lambda: ext.fa_decode(qkv, kc, vc, cos, sin, off, None, 0.125, live, cntc))
-traffic = b*h*int(cnt[1100])*256  # K+V bytes
-print(f'fa_decode(one-pass): {us:.1f} us  eff {traffic/us/1e3:.2f} TB/s')
+off_t = torch.tensor([1100], device=dev)
+for clip in (32, 64, 128, 224, 269):
+    cnt_c = cnt.clamp(max=clip).contiguous()
+    us = timeit(lambda: ext.fa_decode(qkv, kc, vc, cos, sin, off_t, None, 0.125, live, cnt_c))
+    traffic = b*h*min(clip, int(cnt[1100]))*256
+    print(f'live={min(clip,int(cnt[1100])):4d}: {us:6.1f} us  eff {traffic/us/1e6:.2f} TB/s')
