@@ -2391,7 +2391,7 @@ void fa_decode_part_list_kernel(
 // the whole head (lane-parallel dots, shuffle softmax, unrolled PV with 4
 // loads in flight) — no key-split scratch, no combine kernel, KS=1. Four
 // heads per 256-thread block; grid (ceil(h/4), b).
-#define DEC_LMAX 1344
+#define DEC_LMAX 5376
 __global__ __launch_bounds__(256)
 void fa_decode_one_kernel(
     const short* __restrict__ qkv,    // [b, 3*h*64]
