@@ -17,8 +17,10 @@ engine runs the same math with fully static shapes:
   tensor.
 
 Because every step is shape-static it can be captured as a HIP graph
-(``torch.cuda.CUDAGraph`` on ROCm): one graph replay per generated token.
-Sampling (top-k + gumbel) stays outside the graph.
+(``torch.cuda.CUDAGraph`` on ROCm): one graph replay per generated token,
+INCLUDING sampling — the fused top-k+gumbel kernel and the out-buffer
+write are captured, so a token step is a single replay with zero eager
+kernels (the graph-safe philox state advances the noise each replay).
 
 Correctness is pinned by tests/test_decode_engine.py: engine generation ==
 the model's dict-cache generation (itself verified bitwise-equal to

@@ -1578,19 +1578,24 @@ void skinny_gemm_kernel(
   #pragma unroll
   for (int i = 0; i < 8; ++i) acc[i] = f32x4{0, 0, 0, 0};
 
-  // double-buffered A-frag stream over k; B-frags come from LDS
-  bf16x8 af = (wok && kn > 0)
-      ? *reinterpret_cast<const bf16x8*>(wrow + k0 + 8 * grp)
-      : bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
-  for (int k = 0; k < kn; k += 32) {
-    const bf16x8 cur = af;
-    if (wok && k + 32 < kn)
-      af = *reinterpret_cast<const bf16x8*>(wrow + k0 + k + 32 + 8 * grp);
+  // the ENTIRE k-slice's W fragments issued up front (kslice <= 512 ->
+  // <= 16 frags, 64 VGPRs): one-ahead prefetch left each HBM round-trip
+  // exposed (same disease as the decode PV loop, fixed the same way)
+  bf16x8 af[16];
+  const int ksteps = (kn + 31) / 32;
+  #pragma unroll
+  for (int t = 0; t < 16; ++t) {
+    if (t < ksteps && wok)
+      af[t] = *reinterpret_cast<const bf16x8*>(wrow + k0 + 32 * t + 8 * grp);
+    else
+      af[t] = bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
+  }
+  for (int t = 0; t < ksteps; ++t) {
     #pragma unroll 4
     for (int mt = 0; mt < mt_n; ++mt) {
       const bf16x8 xf = *reinterpret_cast<const bf16x8*>(
-          &Xs[(mt * 16 + lq) * xpitch + k + 8 * grp]);
-      acc[mt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(cur, xf, acc[mt], 0, 0, 0);
+          &Xs[(mt * 16 + lq) * xpitch + 32 * t + 8 * grp]);
+      acc[mt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af[t], xf, acc[mt], 0, 0, 0);
     }
   }
 
@@ -3370,6 +3375,7 @@ torch::Tensor skinny_gemm(torch::Tensor x, torch::Tensor w,
   const int m_pad = ((M + 15) / 16) * 16;
   int ksplit = std::min<int>({16, std::max(1, 512 / ntiles), (K + 63) / 64});
   int kslice = ((K + ksplit - 1) / ksplit + 31) & ~31;
+  if (kslice > 512) kslice = 512;   // A-frag register budget (16 frags)
   // x slice must fit LDS: m_pad * (kslice + 8) * 2 bytes
   while ((long)m_pad * (kslice + 8) * 2 > 131072 && kslice > 32)
     kslice = ((kslice / 2) + 31) & ~31;
