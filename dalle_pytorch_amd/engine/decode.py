@@ -624,7 +624,7 @@ class FastDecoder:
         if guided:
             cond, null = logits[:nb], logits[nb:]
             logits = null + (cond - null) * cond_scale
-        if self._fused_decode:
+        if self._fused_decode and logits.shape[-1] <= 8192:
             from dalle_pytorch_amd.ops.dispatch import hip_module
             noise = torch.rand_like(logits)   # graph-capture-safe philox
             nxt = hip_module().sample_topk_gumbel(
