@@ -54,6 +54,9 @@ def parse_args():
     p.add_argument('--batch_size', type=int, default=None, help='per-GPU batch')
     p.add_argument('--mode', type=str, default='train', choices=['train', 'generate'])
     p.add_argument('--gen_batch', type=int, default=64)
+    p.add_argument('--gen_cond_scale', type=float, default=1.0,
+                   help='classifier-free guidance scale for --mode generate '
+                        '(the decoder runs cond+null as one doubled batch)')
     p.add_argument('--no_graph', action='store_true',
                    help='disable HIP-graph decode replay (for kernel profiling)')
     p.add_argument('--fp8', action='store_true',
@@ -164,13 +167,16 @@ def main():
         dalle.eval()
         if not args.eager:
             from dalle_pytorch_amd.engine import FastDecoder
-            decoder = FastDecoder(dalle, batch_size=args.gen_batch,
-                                  use_graph=use_cuda and not args.no_graph)
+            guided = args.gen_cond_scale != 1.0
+            decoder = FastDecoder(
+                dalle, batch_size=args.gen_batch * (2 if guided else 1),
+                use_graph=use_cuda and not args.no_graph)
 
     def gen_step(_):
         text = pool[0][0][:1].repeat(args.gen_batch, 1)
         if decoder is not None:
-            decoder.generate(text, filter_thres=0.9)
+            decoder.generate(text, filter_thres=0.9,
+                             cond_scale=args.gen_cond_scale)
         else:
             with autocast:
                 dalle.generate_images(text, use_cache=True, filter_thres=0.9)
