@@ -35,7 +35,7 @@ from dalle_pytorch_amd.ops.fused import layer_norm as fused_layer_norm
 
 from dalle_pytorch_amd.models import attention as attn_mod
 from dalle_pytorch_amd.models.transformer import CachedAs, NonCached, PreShiftToken
-from dalle_pytorch_amd.models.dalle import top_k, gumbel_sample
+from dalle_pytorch_amd.models.dalle import gumbel_sample
 from dalle_pytorch_amd.ops import attention_core
 
 

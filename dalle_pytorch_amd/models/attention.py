@@ -26,7 +26,6 @@ import os
 from math import ceil
 
 import torch
-import torch.nn.functional as F
 from torch import nn
 
 from dalle_pytorch_amd.models.positional import apply_rotary_to_qkv
