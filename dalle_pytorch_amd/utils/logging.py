@@ -70,3 +70,18 @@ def _sanitize(obj):
     if isinstance(obj, (int, float, str, bool)) or obj is None:
         return obj
     return str(obj)
+
+    def log_artifact(self, path, name='trained-dalle', type_='model'):
+        """Model-artifact upload per save (reference train_dalle.py:584-587,
+        wandb.save + artifact logging); JSONL mode records the path only."""
+        if not self.enabled:
+            return
+        if self.wandb is not None:
+            try:
+                art = self.wandb.Artifact(name, type=type_)
+                art.add_file(str(path))
+                self.wandb.log_artifact(art)
+            except Exception:
+                pass
+        else:
+            self.log({'checkpoint_saved': str(path)})

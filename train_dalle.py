@@ -293,6 +293,7 @@ def main(argv=None):
             save_dalle_checkpoint(step_path, dalle, dalle_params, vae_params,
                                   epoch=epoch, vae_class_name=vae_class_name)
             rotate_checkpoints(out_dir, f'{run_name}-step*.pt', args.keep_n_checkpoints)
+        logger.log_artifact(ckpt_path)
 
     # fail-early checkpoint (reference train_dalle.py:591-594)
     global_step = 0
