@@ -61,16 +61,6 @@ class RunLogger:
         if self._jsonl is not None:
             self._jsonl.close()
 
-
-def _sanitize(obj):
-    if isinstance(obj, dict):
-        return {k: _sanitize(v) for k, v in obj.items()}
-    if isinstance(obj, (list, tuple)):
-        return [_sanitize(v) for v in obj]
-    if isinstance(obj, (int, float, str, bool)) or obj is None:
-        return obj
-    return str(obj)
-
     def log_artifact(self, path, name='trained-dalle', type_='model'):
         """Model-artifact upload per save (reference train_dalle.py:584-587,
         wandb.save + artifact logging); JSONL mode records the path only."""
@@ -85,3 +75,13 @@ def _sanitize(obj):
                 pass
         else:
             self.log({'checkpoint_saved': str(path)})
+
+
+def _sanitize(obj):
+    if isinstance(obj, dict):
+        return {k: _sanitize(v) for k, v in obj.items()}
+    if isinstance(obj, (list, tuple)):
+        return [_sanitize(v) for v in obj]
+    if isinstance(obj, (int, float, str, bool)) or obj is None:
+        return obj
+    return str(obj)
