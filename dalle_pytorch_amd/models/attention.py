@@ -339,6 +339,12 @@ class SparseAttention(_StaticMaskSparseAttention):
     (block=16, global blocks = the text prefix, num_random_blocks =
     seq/block/4, unidirectional); random choices are drawn once from a fixed
     seed so the layout is reproducible across ranks and resumes.
+
+    NOTE: the random-block choices use this module's own seeded generator,
+    NOT DeepSpeed's — a model trained with DeepSpeed's sparse attention has
+    a different (framework-internal) random layout, so such checkpoints
+    load but attend through a different random-block pattern. This is a
+    deliberate native redesign, not an oversight.
     """
 
     def __init__(self, dim, seq_len, causal=True, heads=8, dim_head=64,
