@@ -1387,15 +1387,20 @@ void sample_topk_gumbel_kernel(
   const int tid = threadIdx.x;
   const float* lr = logits + (long)row * V;
 
+  // FIXED 32-slot register cache (dynamic trip counts spill to scratch);
+  // out-of-range slots hold -inf and drop out of every reduction
   float v[32];
-  const int nv = (V - tid + 255) / 256;
-  #pragma unroll 8
-  for (int j = 0; j < nv; ++j) v[j] = lr[tid + j * 256];
+  #pragma unroll
+  for (int j = 0; j < 32; ++j) {
+    const int i = tid + j * 256;
+    v[j] = i < V ? lr[i] : -INFINITY;
+  }
 
   float mx = -INFINITY, mn = INFINITY;
-  for (int j = 0; j < nv; ++j) {
+  #pragma unroll
+  for (int j = 0; j < 32; ++j) {
     mx = fmaxf(mx, v[j]);
-    mn = fminf(mn, v[j]);
+    if (v[j] != -INFINITY) mn = fminf(mn, v[j]);
   }
   #pragma unroll
   for (int sft = 32; sft > 0; sft >>= 1) {
@@ -1412,7 +1417,8 @@ void sample_topk_gumbel_kernel(
   for (int it = 0; it < 24 && lo < hi; ++it) {
     const float mid = 0.5f * (lo + hi);
     int cnt = 0;
-    for (int j = 0; j < nv; ++j) cnt += v[j] > mid;
+    #pragma unroll
+    for (int j = 0; j < 32; ++j) cnt += v[j] > mid;
     #pragma unroll
     for (int sft = 32; sft > 0; sft >>= 1) cnt += __shfl_xor(cnt, sft);
     if ((tid & 63) == 0) redi[tid >> 6] = cnt;
@@ -1427,7 +1433,8 @@ void sample_topk_gumbel_kernel(
   const float* ur = noise + (long)row * V;
   float best = -INFINITY;
   int besti = 0;
-  for (int j = 0; j < nv; ++j) {
+  #pragma unroll 8
+  for (int j = 0; j < 32; ++j) {
     const float x = v[j];
     if (x < tau) continue;
     const int i = tid + j * 256;
