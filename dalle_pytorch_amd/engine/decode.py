@@ -214,20 +214,35 @@ class FastDecoder:
         if st.info['norm_out'] is not None:
             w['lno_w'] = castf(st.info['norm_out'].weight)
             w['lno_b'] = castf(st.info['norm_out'].bias)
+        castT = lambda t: None if t is None else \
+            t.detach().to(self.device, self.dtype).t().contiguous()
         if st.is_attn:
             w['qkv'] = cast(st.leaf.to_qkv.weight)
+            w['qkv_T'] = castT(st.leaf.to_qkv.weight)
             w['out_w'] = cast(st.leaf.to_out[0].weight)
+            w['out_T'] = castT(st.leaf.to_out[0].weight)
             w['out_b'] = cast(st.leaf.to_out[0].bias)
             w['out_b32'] = castf(st.leaf.to_out[0].bias)
         else:
             net = st.leaf.net
             w['ff1_w'] = cast(net[0].weight)
+            w['ff1_T'] = castT(net[0].weight)
             w['ff1_b'] = cast(net[0].bias)
             w['ff1_b32'] = castf(net[0].bias)
             w['ff2_w'] = cast(net[3].weight)
+            w['ff2_T'] = castT(net[3].weight)
             w['ff2_b'] = cast(net[3].bias)
             w['ff2_b32'] = castf(net[3].bias)
         return w
+
+    def _lin_t(self, x, wT, bias):
+        """Decode linear with a pre-transposed weight: hipBLASLt picks a
+        plain-NN kernel instead of the slow transposed-B one at M<=128
+        (decode weights are static, so the transposed copy is free)."""
+        rows = x.numel() // x.shape[-1]
+        x2 = x.reshape(rows, x.shape[-1])
+        out = torch.addmm(bias, x2, wT) if bias is not None else x2 @ wT
+        return out.view(*x.shape[:-1], wT.shape[1])
 
     def _lin(self, x, w, bias32, bias):
         """Decode-step linear. The skinny-M weights-streaming kernel is kept
@@ -265,7 +280,7 @@ class FastDecoder:
         offset 0; decode has n == 1)."""
         leaf = st.leaf
         h, d = leaf.heads, leaf.dim_head
-        qkv = (self._lin(x, st.w['qkv'], None, None) if n == 1
+        qkv = (self._lin_t(x, st.w['qkv_T'], None) if n == 1
                else F.linear(x, st.w['qkv']))
         if n == 1 and self._fused_decode:
             from dalle_pytorch_amd.ops.dispatch import hip_module
@@ -275,7 +290,7 @@ class FastDecoder:
                 self.sin if self.rotary else None,
                 offset_t, st.pattern, leaf.scale,
                 st.live, st.live_cnt).view(self.b, 1, h * d)
-            return self._lin(out, st.w['out_w'], st.w['out_b32'], st.w['out_b'])
+            return self._lin_t(out, st.w['out_T'], st.w['out_b'])
         q, k, v = (t.reshape(self.b, n, h, d).permute(0, 2, 1, 3)
                    for t in qkv.chunk(3, dim=-1))
         if self.rotary:
@@ -376,9 +391,9 @@ class FastDecoder:
             y = self._attn(st, y, offset_t, n)
         elif n == 1:
             from dalle_pytorch_amd.ops import geglu
-            y = self._lin(y, st.w['ff1_w'], st.w['ff1_b32'], st.w['ff1_b'])
+            y = self._lin_t(y, st.w['ff1_T'], st.w['ff1_b'])
             y = geglu(y)
-            y = self._lin(y, st.w['ff2_w'], st.w['ff2_b32'], st.w['ff2_b'])
+            y = self._lin_t(y, st.w['ff2_T'], st.w['ff2_b'])
         else:
             from dalle_pytorch_amd.ops import geglu
             y = F.linear(y, st.w['ff1_w'], st.w['ff1_b'])
