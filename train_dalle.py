@@ -246,8 +246,10 @@ def main(argv=None):
 
     # ------------------------------------------------------ optimizer
     engine = DataParallelEngine(dalle)
+    # fused=True runs the whole Adam update as one multi-tensor HIP kernel
+    # instead of foreach's several full passes over 320M params
     opt = Adam((p for p in dalle.parameters() if p.requires_grad),
-               lr=args.learning_rate)
+               lr=args.learning_rate, fused=device.type == 'cuda')
     if opt_state:
         opt.load_state_dict(opt_state)
     scheduler = None
