@@ -152,6 +152,10 @@ class FastDecoder:
             'b_img32': castf(None if head_lin.bias is None
                              else head_lin.bias[ntt:ntt + nit]),
         }
+        if self.head_w['w_img'].dtype == torch.bfloat16:
+            self.head_w['w_img_pk'] = self._sk2_pack(self.head_w['w_img'])
+        else:
+            self.head_w['w_img_pk'] = None
 
         from dalle_pytorch_amd.ops.dispatch import hip_available
         self.offset_t = torch.zeros(1, dtype=torch.long, device=self.device)
@@ -168,6 +172,9 @@ class FastDecoder:
         self._g_logits = None
         if self._fused_decode:
             self._build_live_tables()
+        import os as _os
+        self._sk2_on = (self._fused_decode
+                        and _os.environ.get('DALLE_AMD_SK2', '1') == '1')
         self._fused_prelude = self._fused_decode and all(
             st.info['norm_out'] is None
             and st.info['scale'] is not None
@@ -196,6 +203,17 @@ class FastDecoder:
                                   idx).to(torch.int32).contiguous()
             st.live_cnt = cnt.contiguous()
 
+    @staticmethod
+    def _sk2_pack(w):
+        """Pack [N, K] bf16 into MFMA A-fragment order [N/16][K/32][lane][8]
+        (lane = kgroup*16 + col) so the sk2 decode GEMM streams each weight
+        tile as one contiguous region. None if the shape doesn't qualify."""
+        N, K = w.shape
+        if N % 32 or K % 512 or not w.is_cuda or w.dtype != torch.bfloat16:
+            return None
+        return (w.reshape(N // 16, 16, K // 32, 4, 8)
+                 .permute(0, 2, 3, 1, 4).contiguous())
+
     def _materialize(self, st):
         """Pre-cast this branch's weights to the engine dtype once: under
         autocast the casts would otherwise be captured into the decode graph
@@ -219,21 +237,42 @@ class FastDecoder:
         if st.is_attn:
             w['qkv'] = cast(st.leaf.to_qkv.weight)
             w['qkv_T'] = castT(st.leaf.to_qkv.weight)
+            w['qkv_pk'] = self._sk2_pack(w['qkv'])
             w['out_w'] = cast(st.leaf.to_out[0].weight)
             w['out_T'] = castT(st.leaf.to_out[0].weight)
+            w['out_pk'] = self._sk2_pack(w['out_w'])
             w['out_b'] = cast(st.leaf.to_out[0].bias)
             w['out_b32'] = castf(st.leaf.to_out[0].bias)
         else:
             net = st.leaf.net
             w['ff1_w'] = cast(net[0].weight)
             w['ff1_T'] = castT(net[0].weight)
+            w['ff1_pk'] = self._sk2_pack(w['ff1_w'])
+            # the geglu-fused epilogue (mode 1) needs N = 2 * ff2-in width
+            if w['ff1_pk'] is not None and net[0].weight.shape[0] % 64:
+                w['ff1_pk'] = None
             w['ff1_b'] = cast(net[0].bias)
             w['ff1_b32'] = castf(net[0].bias)
             w['ff2_w'] = cast(net[3].weight)
             w['ff2_T'] = castT(net[3].weight)
+            w['ff2_pk'] = self._sk2_pack(w['ff2_w'])
             w['ff2_b'] = cast(net[3].bias)
             w['ff2_b32'] = castf(net[3].bias)
         return w
+
+    def _sk2_ok(self, rows, pk):
+        return (self._sk2_on and pk is not None
+                and rows in (16, 32, 64, 128))
+
+    def _sk2(self, x, pk, bias32, N, K, mode):
+        """Weights-streaming decode GEMM on a pre-packed tile layout with
+        the epilogue (bias / geglu / fp32 head) fused — one dispatch where
+        hipBLASLt + eager epilogues took two or three (see sk2_kernel)."""
+        from dalle_pytorch_amd.ops.dispatch import hip_module
+        rows = x.numel() // K
+        out = hip_module().sk2(x.reshape(rows, K), pk, bias32, N, K, mode)
+        no = N // 2 if mode == 1 else N
+        return out.view(*x.shape[:-1], no)
 
     def _lin_t(self, x, wT, bias):
         """Decode linear with a pre-transposed weight: hipBLASLt picks a
@@ -280,8 +319,12 @@ class FastDecoder:
         offset 0; decode has n == 1)."""
         leaf = st.leaf
         h, d = leaf.heads, leaf.dim_head
-        qkv = (self._lin_t(x, st.w['qkv_T'], None) if n == 1
-               else F.linear(x, st.w['qkv']))
+        if n == 1 and self._sk2_ok(x.numel() // x.shape[-1], st.w['qkv_pk']):
+            qkv = self._sk2(x, st.w['qkv_pk'], None,
+                            st.w['qkv'].shape[0], st.w['qkv'].shape[1], 0)
+        else:
+            qkv = (self._lin_t(x, st.w['qkv_T'], None) if n == 1
+                   else F.linear(x, st.w['qkv']))
         if n == 1 and self._fused_decode:
             from dalle_pytorch_amd.ops.dispatch import hip_module
             out = hip_module().fa_decode(
@@ -290,6 +333,10 @@ class FastDecoder:
                 self.sin if self.rotary else None,
                 offset_t, st.pattern, leaf.scale,
                 st.live, st.live_cnt).view(self.b, 1, h * d)
+            if self._sk2_ok(self.b, st.w['out_pk']):
+                return self._sk2(out, st.w['out_pk'], st.w['out_b32'],
+                                 st.w['out_w'].shape[0],
+                                 st.w['out_w'].shape[1], 0)
             return self._lin_t(out, st.w['out_T'], st.w['out_b'])
         q, k, v = (t.reshape(self.b, n, h, d).permute(0, 2, 1, 3)
                    for t in qkv.chunk(3, dim=-1))
@@ -389,6 +436,13 @@ class FastDecoder:
                 self._shift_decode(st, y, offset_t)
         if st.is_attn:
             y = self._attn(st, y, offset_t, n)
+        elif n == 1 and self._sk2_ok(y.numel() // y.shape[-1],
+                                     st.w['ff1_pk']) \
+                and st.w['ff2_pk'] is not None:
+            y = self._sk2(y, st.w['ff1_pk'], st.w['ff1_b32'],
+                          st.w['ff1_w'].shape[0], st.w['ff1_w'].shape[1], 1)
+            y = self._sk2(y, st.w['ff2_pk'], st.w['ff2_b32'],
+                          st.w['ff2_w'].shape[0], st.w['ff2_w'].shape[1], 0)
         elif n == 1:
             from dalle_pytorch_amd.ops import geglu
             y = self._lin_t(y, st.w['ff1_T'], st.w['ff1_b'])
@@ -501,6 +555,11 @@ class FastDecoder:
             x = x / x.amax(dim=-1, keepdim=True)
         x = F.layer_norm(x, (x.shape[-1],), self.head_w['ln_w'],
                          self.head_w['ln_b'], d.to_logits[0].eps)
+        if self._sk2_ok(x.numel() // x.shape[-1], self.head_w['w_img_pk']):
+            return self._sk2(x, self.head_w['w_img_pk'],
+                             self.head_w['b_img32'],
+                             self.head_w['w_img'].shape[0],
+                             self.head_w['w_img'].shape[1], 2)
         return self._lin(x, self.head_w['w_img'], self.head_w['b_img32'],
                          self.head_w['b_img'])
 

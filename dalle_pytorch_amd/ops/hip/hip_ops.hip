@@ -1654,6 +1654,144 @@ void skinny_cast_kernel(
 }
 
 // ---------------------------------------------------------------------------
+// sk2: second-generation decode GEMM, out[M, N] = x[M, K] @ W[N, K]^T (+bias),
+// M in {16, 32, 64, 128}. One kernel, fused epilogue, no trailing cast pass.
+//
+// The token-decode projections are tiny-M GEMMs against multi-MB weights —
+// pure weight-streaming problems that hipBLASLt executes at 12-14 us inside
+// the decode graph (measured profiles/rocprof_generate_r2b.txt: 51% of the
+// whole token step) against a ~1-3 us weight-read roofline. Design:
+//   * W is pre-packed ONCE on the host (decode weights are static) into MFMA
+//     A-fragment order [N/16][K/32][lane][8], so each wave streams ONE fully
+//     contiguous region of HBM with dwordx4 loads — no strided access, no LDS
+//     staging, no shared-memory bank concerns.
+//   * One workgroup per 16-column tile of W; its 4 waves split K four ways
+//     (grid = N/16 >= 64 blocks even for the square out-proj, vs the 48-tile
+//     starved hipBLASLt MT64x64 launch). Partials meet in LDS; the epilogue
+//     (bias add + bf16 cast, the GEGLU gate product, or an fp32 store for the
+//     sampler head) runs in the same kernel.
+//   * x ([M,K] <= 256 KB) stays L2-resident and is read directly as MFMA
+//     B-fragments; the 4x re-read across waves is free next to W traffic.
+//   * Loads are software-pipelined DEPTH chunks ahead with compile-time ring
+//     indices (a dynamic ring index spills the staging array to scratch —
+//     the sample_topk lesson).
+// MODE 0: bias epilogue. MODE 1: GEGLU pair — this block also streams the
+// gate tile at column nt*16 + N/2 and writes value*gelu(gate) (out width
+// N/2), replacing the separate geglu kernel dispatch. MODE 2: fp32 out+bias
+// (the image-vocab head feeding the fp32 sampler).
+// ---------------------------------------------------------------------------
+
+DEVFN float gelu_f(float x);   // defined with the geglu kernels below
+
+template <int MT, int MODE>
+__global__ __launch_bounds__(256, 2)
+void sk2_kernel(const short* __restrict__ x,     // [MT*16, K] bf16
+                const short* __restrict__ wp,    // packed [N/16][K/32][64][8]
+                const float* __restrict__ bias,  // [N] fp32 or null
+                void* __restrict__ out,          // [MT*16, NO]
+                int N, int K) {
+  constexpr int NW = (MODE == 1) ? 2 : 1;        // weight streams per block
+  constexpr int DEPTH = (MT >= 8 || (MT >= 4 && NW == 2)) ? 2 : 4;
+  const int nt = blockIdx.x;
+  const int wave = threadIdx.x >> 6;             // = this wave's k-split
+  const int lane = threadIdx.x & 63;
+  const int lq = lane & 15;
+  const int kg = lane >> 4;
+  const int KC = K >> 5;                         // 32-wide k-chunks total
+  const int KCW = KC >> 2;                       // chunks per wave
+  const long tile_elems = (long)KC * 512;        // shorts per packed n-tile
+
+  const short* wb0 = wp + (long)nt * tile_elems
+                        + ((long)wave * KCW) * 512 + lane * 8;
+  const short* wb1 = (MODE == 1)
+      ? wp + ((long)nt + (N >> 5)) * tile_elems
+           + ((long)wave * KCW) * 512 + lane * 8
+      : nullptr;
+  const short* xb = x + (long)lq * K + kg * 8 + (long)wave * KCW * 32;
+
+  f32x4 acc[NW][MT];
+  #pragma unroll
+  for (int s = 0; s < NW; ++s)
+    #pragma unroll
+    for (int mt = 0; mt < MT; ++mt) acc[s][mt] = f32x4{0, 0, 0, 0};
+
+  bf16x8 wrg[DEPTH][NW];
+  bf16x8 xrg[DEPTH][MT];
+
+  // prologue: fill the ring (KCW >= DEPTH, host-enforced via K % 512 == 0)
+  #pragma unroll
+  for (int p = 0; p < DEPTH; ++p) {
+    wrg[p][0] = *reinterpret_cast<const bf16x8*>(wb0 + (long)p * 512);
+    if (MODE == 1)
+      wrg[p][NW - 1] = *reinterpret_cast<const bf16x8*>(wb1 + (long)p * 512);
+    #pragma unroll
+    for (int mt = 0; mt < MT; ++mt)
+      xrg[p][mt] = *reinterpret_cast<const bf16x8*>(xb + (long)mt * 16 * K
+                                                       + (long)p * 32);
+  }
+
+  for (int base = 0; base < KCW; base += DEPTH) {
+    #pragma unroll
+    for (int p = 0; p < DEPTH; ++p) {
+      __builtin_amdgcn_s_setprio(1);
+      #pragma unroll
+      for (int mt = 0; mt < MT; ++mt) {
+        acc[0][mt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            wrg[p][0], xrg[p][mt], acc[0][mt], 0, 0, 0);
+        if (MODE == 1)
+          acc[NW - 1][mt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              wrg[p][NW - 1], xrg[p][mt], acc[NW - 1][mt], 0, 0, 0);
+      }
+      __builtin_amdgcn_s_setprio(0);
+      const int nx = base + DEPTH + p;
+      if (nx < KCW) {
+        wrg[p][0] = *reinterpret_cast<const bf16x8*>(wb0 + (long)nx * 512);
+        if (MODE == 1)
+          wrg[p][NW - 1] =
+              *reinterpret_cast<const bf16x8*>(wb1 + (long)nx * 512);
+        #pragma unroll
+        for (int mt = 0; mt < MT; ++mt)
+          xrg[p][mt] = *reinterpret_cast<const bf16x8*>(
+              xb + (long)mt * 16 * K + (long)nx * 32);
+      }
+    }
+  }
+
+  // k-split reduce through LDS. Layout [wave][s][mt][n][m]; MFMA C frag:
+  // m (x row) = lane&15, n (w row) = grp*4 + r.
+  __shared__ float red[4][NW][MT][16][16];
+  #pragma unroll
+  for (int s = 0; s < NW; ++s)
+    #pragma unroll
+    for (int mt = 0; mt < MT; ++mt)
+      #pragma unroll
+      for (int r = 0; r < 4; ++r)
+        red[wave][s][mt][kg * 4 + r][lq] = acc[s][mt][r];
+  __syncthreads();
+
+  const int NO = (MODE == 1) ? (N >> 1) : N;
+  for (int i = threadIdx.x; i < MT * 256; i += 256) {
+    const int mt = i >> 8, m = (i >> 4) & 15, n = i & 15;
+    float v = red[0][0][mt][n][m] + red[1][0][mt][n][m]
+            + red[2][0][mt][n][m] + red[3][0][mt][n][m];
+    const long o = (long)(mt * 16 + m) * NO + nt * 16 + n;
+    if (MODE == 1) {
+      float g = red[0][NW - 1][mt][n][m] + red[1][NW - 1][mt][n][m]
+              + red[2][NW - 1][mt][n][m] + red[3][NW - 1][mt][n][m];
+      if (bias != nullptr) {
+        v += bias[nt * 16 + n];
+        g += bias[(N >> 1) + nt * 16 + n];
+      }
+      reinterpret_cast<short*>(out)[o] = f2bf(v * gelu_f(g));
+    } else {
+      if (bias != nullptr) v += bias[nt * 16 + n];
+      if (MODE == 2) reinterpret_cast<float*>(out)[o] = v;
+      else           reinterpret_cast<short*>(out)[o] = f2bf(v);
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
 // Fused residual + LayerScale: out = x + gamma * y (gamma per-channel).
 // The eager chain (scale cast, y*gamma temp, x+temp) is 5 full-tensor passes
 // + a 1-element cast kernel per call; this is 3 passes, one launch. Backward:
@@ -3411,6 +3549,51 @@ torch::Tensor skinny_gemm(torch::Tensor x, torch::Tensor w,
   return out;
 }
 
+torch::Tensor sk2(torch::Tensor x, torch::Tensor wp,
+                  std::optional<torch::Tensor> bias,
+                  long N, long K, long mode) {
+  CHK(x.is_cuda() && x.dtype() == torch::kBFloat16 && x.is_contiguous());
+  CHK(wp.dtype() == torch::kBFloat16 && wp.is_contiguous());
+  const long rows = x.numel() / K;
+  CHK(x.size(-1) == K);
+  CHK(rows == 16 || rows == 32 || rows == 64 || rows == 128);
+  CHK(K % 512 == 0 && N % 32 == 0);
+  CHK(wp.numel() == N * K);
+  CHK(mode >= 0 && mode <= 2);
+  const float* bp = nullptr;
+  if (bias.has_value()) {
+    CHK(bias->dtype() == torch::kFloat32 && bias->is_contiguous());
+    CHK(bias->numel() == N);
+    bp = bias->data_ptr<float>();
+  }
+  const long NO = (mode == 1) ? N / 2 : N;
+  auto out = torch::empty({rows, NO},
+      x.options().dtype(mode == 2 ? torch::kFloat32 : torch::kBFloat16));
+  const dim3 grid(mode == 1 ? N / 32 : N / 16);
+  const short* xp = reinterpret_cast<const short*>(x.data_ptr());
+  const short* wpp = reinterpret_cast<const short*>(wp.data_ptr());
+  void* op = out.data_ptr();
+  const int Ni = (int)N, Ki = (int)K;
+  #define SK2_LAUNCH(MT, MODE)                                              \
+    hipLaunchKernelGGL((sk2_kernel<MT, MODE>), grid, dim3(256), 0,          \
+                       cur_stream(), xp, wpp, bp, op, Ni, Ki)
+  #define SK2_MT(MT)                                                        \
+    switch (mode) {                                                         \
+      case 0: SK2_LAUNCH(MT, 0); break;                                     \
+      case 1: SK2_LAUNCH(MT, 1); break;                                     \
+      default: SK2_LAUNCH(MT, 2); break;                                    \
+    }
+  switch (rows) {
+    case 16: SK2_MT(1); break;
+    case 32: SK2_MT(2); break;
+    case 64: SK2_MT(4); break;
+    default: SK2_MT(8); break;
+  }
+  #undef SK2_MT
+  #undef SK2_LAUNCH
+  return out;
+}
+
 torch::Tensor resls_fwd(torch::Tensor x, torch::Tensor y, torch::Tensor gamma) {
   CHK(x.is_cuda() && x.dtype() == torch::kBFloat16);
   CHK(x.is_contiguous() && y.is_contiguous() && gamma.is_contiguous());
@@ -3457,6 +3640,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("sample_topk_gumbel", &sample_topk_gumbel,
         "fused top-k threshold + gumbel argmax sampling");
   m.def("amax_bf16", &amax_bf16, "abs-max of a bf16 tensor (one pass)");
+  m.def("sk2", &sk2,
+        "decode skinny GEMM on packed weights (fused bias/geglu/fp32 head)");
   m.def("skinny_gemm", &skinny_gemm,
         "skinny-M weights-streaming GEMM (decode projections)",
         py::arg("x"), py::arg("w"), py::arg("bias") = std::nullopt);

@@ -757,3 +757,41 @@ def test_layer_norm_vs_oracle(ext):
                             (b.grad, br.grad, 'db')):
         rel = (got.float() - want).abs().max() / want.abs().max().clamp(min=1e-6)
         assert rel < 2e-2, f'{name} {rel.item()}'
+
+
+@pytest.mark.gpu
+def test_sk2_decode_gemm_matches_linear():
+    """sk2 (packed weights-streaming decode GEMM) vs fp32 F.linear across
+    row counts and all three epilogue modes (bias / geglu / fp32 head)."""
+    import dalle_pytorch_amd._hip as ext
+    from dalle_pytorch_amd.engine.decode import FastDecoder
+    torch.manual_seed(0)
+    dev = 'cuda'
+    for rows, K, N in [(64, 1024, 3072), (64, 1024, 1024), (128, 1024, 1024),
+                       (32, 4096, 1024), (16, 1024, 512), (64, 512, 2048)]:
+        x = (torch.randn(rows, K, device=dev) * 0.3).bfloat16()
+        w = (torch.randn(N, K, device=dev) * 0.05).bfloat16()
+        b = torch.randn(N, device=dev)
+        pk = FastDecoder._sk2_pack(w)
+        assert pk is not None
+        out = ext.sk2(x, pk, b, N, K, 0)
+        ref = F.linear(x.float(), w.float(), b)
+        torch.testing.assert_close(out.float(), ref, rtol=3e-2, atol=3e-2)
+        out_nb = ext.sk2(x, pk, None, N, K, 0)
+        torch.testing.assert_close(out_nb.float(), ref - b, rtol=3e-2,
+                                   atol=3e-2)
+    # mode 1: fused geglu epilogue
+    rows, K, N = 64, 1024, 8192
+    x = (torch.randn(rows, K, device='cuda') * 0.3).bfloat16()
+    w = (torch.randn(N, K, device='cuda') * 0.05).bfloat16()
+    b = torch.randn(N, device='cuda')
+    pk = FastDecoder._sk2_pack(w)
+    y = F.linear(x.float(), w.float(), b)
+    v, g = y.chunk(2, dim=-1)
+    ref = v * F.gelu(g)
+    out = ext.sk2(x, pk, b, N, K, 1)
+    torch.testing.assert_close(out.float(), ref, rtol=3e-2, atol=3e-2)
+    # mode 2: fp32 output (image-vocab head)
+    out32 = ext.sk2(x, pk, b, N, K, 2)
+    assert out32.dtype == torch.float32
+    torch.testing.assert_close(out32, y, rtol=3e-2, atol=3e-2)
