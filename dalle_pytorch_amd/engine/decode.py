@@ -260,9 +260,15 @@ class FastDecoder:
             w['ff2_b32'] = castf(net[3].bias)
         return w
 
-    def _sk2_ok(self, rows, pk):
-        return (self._sk2_on and pk is not None
-                and rows in (16, 32, 64, 128))
+    def _sk2_ok(self, rows, pk, mode=0):
+        # measured on-box (scripts/bench_sk2.py): at 128 rows the MT=8
+        # variants tie or lose to hipBLASLt except the geglu-fused ff1;
+        # at <= 64 rows sk2 wins every decode shape
+        if not (self._sk2_on and pk is not None):
+            return False
+        if rows in (16, 32, 64):
+            return True
+        return rows == 128 and mode == 1
 
     def _sk2(self, x, pk, bias32, N, K, mode):
         """Weights-streaming decode GEMM on a pre-packed tile layout with
@@ -319,7 +325,7 @@ class FastDecoder:
         offset 0; decode has n == 1)."""
         leaf = st.leaf
         h, d = leaf.heads, leaf.dim_head
-        if n == 1 and self._sk2_ok(x.numel() // x.shape[-1], st.w['qkv_pk']):
+        if n == 1 and self._sk2_ok(x.numel() // x.shape[-1], st.w['qkv_pk'], 0):
             qkv = self._sk2(x, st.w['qkv_pk'], None,
                             st.w['qkv'].shape[0], st.w['qkv'].shape[1], 0)
         else:
@@ -333,7 +339,7 @@ class FastDecoder:
                 self.sin if self.rotary else None,
                 offset_t, st.pattern, leaf.scale,
                 st.live, st.live_cnt).view(self.b, 1, h * d)
-            if self._sk2_ok(self.b, st.w['out_pk']):
+            if self._sk2_ok(self.b, st.w['out_pk'], 0):
                 return self._sk2(out, st.w['out_pk'], st.w['out_b32'],
                                  st.w['out_w'].shape[0],
                                  st.w['out_w'].shape[1], 0)
@@ -437,8 +443,9 @@ class FastDecoder:
         if st.is_attn:
             y = self._attn(st, y, offset_t, n)
         elif n == 1 and self._sk2_ok(y.numel() // y.shape[-1],
-                                     st.w['ff1_pk']) \
-                and st.w['ff2_pk'] is not None:
+                                     st.w['ff1_pk'], 1) \
+                and self._sk2_ok(y.numel() // y.shape[-1],
+                                 st.w['ff2_pk'], 0):
             y = self._sk2(y, st.w['ff1_pk'], st.w['ff1_b32'],
                           st.w['ff1_w'].shape[0], st.w['ff1_w'].shape[1], 1)
             y = self._sk2(y, st.w['ff2_pk'], st.w['ff2_b32'],
@@ -555,7 +562,7 @@ class FastDecoder:
             x = x / x.amax(dim=-1, keepdim=True)
         x = F.layer_norm(x, (x.shape[-1],), self.head_w['ln_w'],
                          self.head_w['ln_b'], d.to_logits[0].eps)
-        if self._sk2_ok(x.numel() // x.shape[-1], self.head_w['w_img_pk']):
+        if self._sk2_ok(x.numel() // x.shape[-1], self.head_w['w_img_pk'], 2):
             return self._sk2(x, self.head_w['w_img_pk'],
                              self.head_w['b_img32'],
                              self.head_w['w_img'].shape[0],

@@ -1691,7 +1691,12 @@ void sk2_kernel(const short* __restrict__ x,     // [MT*16, K] bf16
                 void* __restrict__ out,          // [MT*16, NO]
                 int N, int K) {
   constexpr int NW = (MODE == 1) ? 2 : 1;        // weight streams per block
-  constexpr int DEPTH = (MT >= 8 || (MT >= 4 && NW == 2)) ? 2 : 4;
+  // ring depth bounds the per-wave outstanding loads (the compiler drains
+  // vmcnt(0) once per ring cycle, so bytes-in-flight = DEPTH * frags * 16 B);
+  // deeper is faster until VGPR staging (DEPTH * (NW + MT) * 4 regs) costs
+  // occupancy. KCW % DEPTH == 0 must hold (K % 512 == 0 gives KCW in 8/16/32).
+  constexpr int DEPTH = (MT >= 8 || (MT >= 4 && NW == 2)) ? 2
+                        : (MT >= 4 ? 8 : 8);
   const int nt = blockIdx.x;
   const int wave = threadIdx.x >> 6;             // = this wave's k-split
   const int lane = threadIdx.x & 63;

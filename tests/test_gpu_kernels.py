@@ -763,6 +763,7 @@ def test_layer_norm_vs_oracle(ext):
 def test_sk2_decode_gemm_matches_linear():
     """sk2 (packed weights-streaming decode GEMM) vs fp32 F.linear across
     row counts and all three epilogue modes (bias / geglu / fp32 head)."""
+    import torch.nn.functional as F
     import dalle_pytorch_amd._hip as ext
     from dalle_pytorch_amd.engine.decode import FastDecoder
     torch.manual_seed(0)
