@@ -175,6 +175,9 @@ class FastDecoder:
         import os as _os
         self._sk2_on = (self._fused_decode
                         and _os.environ.get('DALLE_AMD_SK2', '1') == '1')
+        # A/B lever: shapes with K above this go through hipBLASLt instead
+        # (the K=4096 ff2 is the one sk2 shape still far from its roofline)
+        self._sk2_maxk = int(_os.environ.get('DALLE_AMD_SK2_MAXK', '8192'))
         self._fused_prelude = self._fused_decode and all(
             st.info['norm_out'] is None
             and st.info['scale'] is not None
@@ -260,11 +263,11 @@ class FastDecoder:
             w['ff2_b32'] = castf(net[3].bias)
         return w
 
-    def _sk2_ok(self, rows, pk, mode=0):
+    def _sk2_ok(self, rows, pk, mode=0, k_dim=0):
         # measured on-box (scripts/bench_sk2.py): at 128 rows the MT=8
         # variants tie or lose to hipBLASLt except the geglu-fused ff1;
         # at <= 64 rows sk2 wins every decode shape
-        if not (self._sk2_on and pk is not None):
+        if not (self._sk2_on and pk is not None) or k_dim > self._sk2_maxk:
             return False
         if rows in (16, 32, 64):
             return True
@@ -444,8 +447,8 @@ class FastDecoder:
             y = self._attn(st, y, offset_t, n)
         elif n == 1 and self._sk2_ok(y.numel() // y.shape[-1],
                                      st.w['ff1_pk'], 1) \
-                and self._sk2_ok(y.numel() // y.shape[-1],
-                                 st.w['ff2_pk'], 0):
+                and self._sk2_ok(y.numel() // y.shape[-1], st.w['ff2_pk'],
+                                 0, st.w['ff2_w'].shape[1]):
             y = self._sk2(y, st.w['ff1_pk'], st.w['ff1_b32'],
                           st.w['ff1_w'].shape[0], st.w['ff1_w'].shape[1], 1)
             y = self._sk2(y, st.w['ff2_pk'], st.w['ff2_b32'],
