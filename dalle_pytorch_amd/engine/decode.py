@@ -175,9 +175,10 @@ class FastDecoder:
         import os as _os
         self._sk2_on = (self._fused_decode
                         and _os.environ.get('DALLE_AMD_SK2', '1') == '1')
-        # A/B lever: shapes with K above this go through hipBLASLt instead
-        # (the K=4096 ff2 is the one sk2 shape still far from its roofline)
-        self._sk2_maxk = int(_os.environ.get('DALLE_AMD_SK2_MAXK', '8192'))
+        # the K=4096 ff2 is the one decode shape where hipBLASLt still wins
+        # (14 us in-situ F.linear vs 17.9 us sk2 — sk2's ring drains on HBM
+        # latency at only 64 blocks); default keeps ff2 on hipBLASLt
+        self._sk2_maxk = int(_os.environ.get('DALLE_AMD_SK2_MAXK', '2048'))
         self._fused_prelude = self._fused_decode and all(
             st.info['norm_out'] is None
             and st.info['scale'] is not None
@@ -445,19 +446,22 @@ class FastDecoder:
                 self._shift_decode(st, y, offset_t)
         if st.is_attn:
             y = self._attn(st, y, offset_t, n)
-        elif n == 1 and self._sk2_ok(y.numel() // y.shape[-1],
-                                     st.w['ff1_pk'], 1) \
-                and self._sk2_ok(y.numel() // y.shape[-1], st.w['ff2_pk'],
-                                 0, st.w['ff2_w'].shape[1]):
-            y = self._sk2(y, st.w['ff1_pk'], st.w['ff1_b32'],
-                          st.w['ff1_w'].shape[0], st.w['ff1_w'].shape[1], 1)
-            y = self._sk2(y, st.w['ff2_pk'], st.w['ff2_b32'],
-                          st.w['ff2_w'].shape[0], st.w['ff2_w'].shape[1], 0)
         elif n == 1:
-            from dalle_pytorch_amd.ops import geglu
-            y = self._lin_t(y, st.w['ff1_T'], st.w['ff1_b'])
-            y = geglu(y)
-            y = self._lin_t(y, st.w['ff2_T'], st.w['ff2_b'])
+            rows = y.numel() // y.shape[-1]
+            if self._sk2_ok(rows, st.w['ff1_pk'], 1):
+                y = self._sk2(y, st.w['ff1_pk'], st.w['ff1_b32'],
+                              st.w['ff1_w'].shape[0],
+                              st.w['ff1_w'].shape[1], 1)
+            else:
+                from dalle_pytorch_amd.ops import geglu
+                y = geglu(self._lin_t(y, st.w['ff1_T'], st.w['ff1_b']))
+            if self._sk2_ok(rows, st.w['ff2_pk'], 0,
+                            st.w['ff2_w'].shape[1]):
+                y = self._sk2(y, st.w['ff2_pk'], st.w['ff2_b32'],
+                              st.w['ff2_w'].shape[0],
+                              st.w['ff2_w'].shape[1], 0)
+            else:
+                y = self._lin_t(y, st.w['ff2_T'], st.w['ff2_b'])
         else:
             from dalle_pytorch_amd.ops import geglu
             y = F.linear(y, st.w['ff1_w'], st.w['ff1_b'])
@@ -478,9 +482,17 @@ class FastDecoder:
         """Branch compute after the (fused) LN+shift prelude: attention or FF."""
         if st.is_attn:
             return self._attn(st, z, offset_t, 1)
-        from dalle_pytorch_amd.ops import geglu
-        y = self._lin(z, st.w['ff1_w'], st.w['ff1_b32'], st.w['ff1_b'])
-        y = geglu(y)
+        rows = z.numel() // z.shape[-1]
+        if self._sk2_ok(rows, st.w['ff1_pk'], 1):
+            y = self._sk2(z, st.w['ff1_pk'], st.w['ff1_b32'],
+                          st.w['ff1_w'].shape[0], st.w['ff1_w'].shape[1], 1)
+        else:
+            from dalle_pytorch_amd.ops import geglu
+            y = geglu(self._lin(z, st.w['ff1_w'], st.w['ff1_b32'],
+                                st.w['ff1_b']))
+        if self._sk2_ok(rows, st.w['ff2_pk'], 0, st.w['ff2_w'].shape[1]):
+            return self._sk2(y, st.w['ff2_pk'], st.w['ff2_b32'],
+                             st.w['ff2_w'].shape[0], st.w['ff2_w'].shape[1], 0)
         return self._lin(y, st.w['ff2_w'], st.w['ff2_b32'], st.w['ff2_b'])
 
     def _prelude(self, ext, stream, pend, st, off):
