@@ -213,7 +213,7 @@ class FastDecoder:
         (lane = kgroup*16 + col) so the sk2 decode GEMM streams each weight
         tile as one contiguous region. None if the shape doesn't qualify."""
         N, K = w.shape
-        if N % 32 or K % 512 or not w.is_cuda or w.dtype != torch.bfloat16:
+        if N % 32 or K % 1024 or not w.is_cuda or w.dtype != torch.bfloat16:
             return None
         return (w.reshape(N // 16, 16, K // 32, 4, 8)
                  .permute(0, 2, 3, 1, 4).contiguous())

@@ -3562,7 +3562,10 @@ torch::Tensor sk2(torch::Tensor x, torch::Tensor wp,
   const long rows = x.numel() / K;
   CHK(x.size(-1) == K);
   CHK(rows == 16 || rows == 32 || rows == 64 || rows == 128);
-  CHK(K % 512 == 0 && N % 32 == 0);
+  // K % 1024 == 0 guarantees KCW (= K/128 chunks per wave) is a multiple of
+  // every template's ring DEPTH (<= 8); smaller K would read past the last
+  // chunk in the prologue
+  CHK(K % 1024 == 0 && N % 32 == 0);
   CHK(wp.numel() == N * K);
   CHK(mode >= 0 && mode <= 2);
   const float* bp = nullptr;
