@@ -1683,8 +1683,8 @@ void skinny_cast_kernel(
 
 DEVFN float gelu_f(float x);   // defined with the geglu kernels below
 
-template <int MT, int MODE>
-__global__ __launch_bounds__(256, 2)
+template <int MT, int MODE, int KS = 4>
+__global__ __launch_bounds__(KS * 64, 2)
 void sk2_kernel(const short* __restrict__ x,     // [MT*16, K] bf16
                 const short* __restrict__ wp,    // packed [N/16][K/32][64][8]
                 const float* __restrict__ bias,  // [N] fp32 or null
@@ -1703,7 +1703,7 @@ void sk2_kernel(const short* __restrict__ x,     // [MT*16, K] bf16
   const int lq = lane & 15;
   const int kg = lane >> 4;
   const int KC = K >> 5;                         // 32-wide k-chunks total
-  const int KCW = KC >> 2;                       // chunks per wave
+  const int KCW = KC / KS;                       // chunks per wave
   const long tile_elems = (long)KC * 512;        // shorts per packed n-tile
 
   const short* wb0 = wp + (long)nt * tile_elems
@@ -1764,7 +1764,7 @@ void sk2_kernel(const short* __restrict__ x,     // [MT*16, K] bf16
 
   // k-split reduce through LDS. Layout [wave][s][mt][n][m]; MFMA C frag:
   // m (x row) = lane&15, n (w row) = grp*4 + r.
-  __shared__ float red[4][NW][MT][16][16];
+  __shared__ float red[KS][NW][MT][16][16];
   #pragma unroll
   for (int s = 0; s < NW; ++s)
     #pragma unroll
@@ -1775,14 +1775,15 @@ void sk2_kernel(const short* __restrict__ x,     // [MT*16, K] bf16
   __syncthreads();
 
   const int NO = (MODE == 1) ? (N >> 1) : N;
-  for (int i = threadIdx.x; i < MT * 256; i += 256) {
+  for (int i = threadIdx.x; i < MT * 256; i += KS * 64) {
     const int mt = i >> 8, m = (i >> 4) & 15, n = i & 15;
-    float v = red[0][0][mt][n][m] + red[1][0][mt][n][m]
-            + red[2][0][mt][n][m] + red[3][0][mt][n][m];
+    float v = 0.f, g = 0.f;
+    #pragma unroll
+    for (int s4 = 0; s4 < KS; ++s4) v += red[s4][0][mt][n][m];
     const long o = (long)(mt * 16 + m) * NO + nt * 16 + n;
     if (MODE == 1) {
-      float g = red[0][NW - 1][mt][n][m] + red[1][NW - 1][mt][n][m]
-              + red[2][NW - 1][mt][n][m] + red[3][NW - 1][mt][n][m];
+      #pragma unroll
+      for (int s4 = 0; s4 < KS; ++s4) g += red[s4][NW - 1][mt][n][m];
       if (bias != nullptr) {
         v += bias[nt * 16 + n];
         g += bias[(N >> 1) + nt * 16 + n];
@@ -3582,14 +3583,18 @@ torch::Tensor sk2(torch::Tensor x, torch::Tensor wp,
   const short* wpp = reinterpret_cast<const short*>(wp.data_ptr());
   void* op = out.data_ptr();
   const int Ni = (int)N, Ki = (int)K;
-  #define SK2_LAUNCH(MT, MODE)                                              \
-    hipLaunchKernelGGL((sk2_kernel<MT, MODE>), grid, dim3(256), 0,          \
+  // long-K bias shapes (ff2): 8-way k-split per block doubles the in-flight
+  // weight stream (the 4-wave variant is HBM-latency-bound at 64 blocks)
+  const bool ks8 = mode == 0 && K >= 4096 && K % 2048 == 0 && rows <= 64;
+  #define SK2_LAUNCH(MT, MODE, KS)                                          \
+    hipLaunchKernelGGL((sk2_kernel<MT, MODE, KS>), grid, dim3(KS * 64), 0,  \
                        cur_stream(), xp, wpp, bp, op, Ni, Ki)
   #define SK2_MT(MT)                                                        \
     switch (mode) {                                                         \
-      case 0: SK2_LAUNCH(MT, 0); break;                                     \
-      case 1: SK2_LAUNCH(MT, 1); break;                                     \
-      default: SK2_LAUNCH(MT, 2); break;                                    \
+      case 0: if (ks8) SK2_LAUNCH(MT, 0, 8); else SK2_LAUNCH(MT, 0, 4);     \
+              break;                                                        \
+      case 1: SK2_LAUNCH(MT, 1, 4); break;                                  \
+      default: SK2_LAUNCH(MT, 2, 4); break;                                 \
     }
   switch (rows) {
     case 16: SK2_MT(1); break;
