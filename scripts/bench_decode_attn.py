@@ -27,12 +27,19 @@ cntc = cnt.contiguous()
 print('lmax', lmax, 'live@1100', int(cnt[1100]))
 
 def timeit(fn, iters=300):
-    for _ in range(30): fn()
-    torch.cuda.synchronize()
+    # graph-replay timing: eager launches cost ~19 us of host overhead and
+    # mask the kernel (the skinny.txt lesson); capture 20 calls, replay
+    fn(); torch.cuda.synchronize()
+    g = torch.cuda.CUDAGraph()
+    with torch.cuda.graph(g):
+        for _ in range(20):
+            fn()
+    g.replay(); torch.cuda.synchronize()
     t0 = time.perf_counter()
-    for _ in range(iters): fn()
+    for _ in range(iters // 10):
+        g.replay()
     torch.cuda.synchronize()
-    return (time.perf_counter() - t0)/iters*1e6
+    return (time.perf_counter() - t0) / (iters // 10) / 20 * 1e6
 
 off_t = torch.tensor([1100], device=dev)
 for clip in (32, 64, 128, 224, 269):
