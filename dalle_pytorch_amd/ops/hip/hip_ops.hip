@@ -2659,11 +2659,10 @@ void fa_decode_one_kernel(
     m_loc = fmaxf(m_loc, p);
   };
   int j = lane;
-  for (; j + 192 < jn; j += 256) {   // 4 rows (512 B) in flight per lane
+  for (; j + 128 < jn; j += 192) {
     dot_one(j);
     dot_one(j + 64);
     dot_one(j + 128);
-    dot_one(j + 192);
   }
   for (; j < jn; j += 64) dot_one(j);
   #pragma unroll
@@ -2700,10 +2699,6 @@ void fa_decode_one_kernel(
     for (int e = 0; e < 4; ++e) a[e] += pj * bf2f(v4[e]);
   };
   int i = kg;
-  for (; i + 124 < jn; i += 128) {   // 32 V loads (256 B) in flight per lane
-    #pragma unroll
-    for (int u = 0; u < 32; ++u) pv_one(i + 4 * u, a4[u & 3]);
-  }
   for (; i + 60 < jn; i += 64) {
     #pragma unroll
     for (int u = 0; u < 16; ++u) pv_one(i + 4 * u, a4[u & 3]);
