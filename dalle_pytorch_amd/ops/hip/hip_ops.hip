@@ -1695,8 +1695,8 @@ void sk2_kernel(const short* __restrict__ x,     // [MT*16, K] bf16
   // vmcnt(0) once per ring cycle, so bytes-in-flight = DEPTH * frags * 16 B);
   // deeper is faster until VGPR staging (DEPTH * (NW + MT) * 4 regs) costs
   // occupancy. KCW % DEPTH == 0 must hold (K % 512 == 0 gives KCW in 8/16/32).
-  constexpr int DEPTH = (MT >= 8 || (MT >= 4 && NW == 2)) ? 2
-                        : (MT >= 4 ? 8 : 8);
+  constexpr int DEPTH = (NW == 2 && MT >= 4) ? 2
+                        : (MT >= 8 ? 4 : 8);
   const int nt = blockIdx.x;
   const int wave = threadIdx.x >> 6;             // = this wave's k-split
   const int lane = threadIdx.x & 63;
