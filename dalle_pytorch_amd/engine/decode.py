@@ -735,14 +735,18 @@ class FastDecoder:
         if self._fused_decode and logits.shape[-1] <= 8192:
             from dalle_pytorch_amd.ops.dispatch import hip_module
             noise = torch.rand_like(logits)   # graph-capture-safe philox
+            # the sampler writes the next-token feed buffer IN PLACE and
+            # appends to the sequence buffer itself: no index_copy_ kernel
+            # and no eager feed copy between graph replays
             nxt = hip_module().sample_topk_gumbel(
-                logits.contiguous(), noise, k, max(temperature, 1e-10))
+                logits.contiguous(), noise, k, max(temperature, 1e-10),
+                out_tok=token, seq=self._out_buf, seq_ptr=self._gen_ptr)
         else:
             vals, idx = logits.topk(k, dim=-1)
             filtered = torch.full_like(logits, -torch.finfo(logits.dtype).max)
             filtered.scatter_(1, idx, vals)
             nxt = gumbel_sample(filtered, temperature=temperature)
-        self._out_buf.index_copy_(1, self._gen_ptr, nxt.unsqueeze(1))
+            self._out_buf.index_copy_(1, self._gen_ptr, nxt.unsqueeze(1))
         self._gen_ptr += 1
         return nxt
 
@@ -762,8 +766,9 @@ class FastDecoder:
                 self._g_next = self._token_step(self._g_token, *args)
             self._graph.replay()
             return self._g_next
-        self._g_token.copy_(token)
-        self._graph.replay()
+        if token is not self._g_token:   # the sampler writes the feed
+            self._g_token.copy_(token)   # buffer in place, so steady-state
+        self._graph.replay()             # replays skip this copy entirely
         return self._g_next
 
     def _graph_step(self, token):

@@ -1376,8 +1376,10 @@ __global__ __launch_bounds__(256)
 void sample_topk_gumbel_kernel(
     const float* __restrict__ logits,   // [rows, V], V <= 8192
     const float* __restrict__ noise,    // [rows, V] uniform(0,1)
-    long* __restrict__ out,             // [rows]
-    int V, int k, float inv_temp) {
+    long* __restrict__ out,             // [rows] (doubles as the next feed)
+    long* __restrict__ seq,             // [rows, S] generated-sequence buffer
+    const long* __restrict__ seq_ptr,   // [1] write position, or null
+    int S, int V, int k, float inv_temp) {
   // row cached in REGISTERS (32 values/thread @ V=8192): the LDS-staged
   // form spent ~42 us in bisection LDS sweeps; this runs the 24 bisection
   // rounds over registers with one tiny cross-wave reduce each
@@ -1465,6 +1467,9 @@ void sample_topk_gumbel_kernel(
       }
     }
     out[row] = besti;
+    // fold the sequence-buffer write into the same dispatch (replaces the
+    // eager index_copy_ kernel + its index plumbing in the decode graph)
+    if (seq != nullptr) seq[(long)row * S + *seq_ptr] = besti;
   }
 }
 
@@ -3487,18 +3492,39 @@ torch::Tensor amax_bf16(torch::Tensor x) {
 }
 
 torch::Tensor sample_topk_gumbel(torch::Tensor logits, torch::Tensor noise,
-                                 int64_t k, double temperature) {
+                                 int64_t k, double temperature,
+                                 std::optional<torch::Tensor> out_tok,
+                                 std::optional<torch::Tensor> seq,
+                                 std::optional<torch::Tensor> seq_ptr) {
   CHK(logits.is_cuda() && logits.dtype() == torch::kFloat32 &&
       logits.is_contiguous() && logits.dim() == 2);
   CHK(noise.sizes() == logits.sizes() && noise.dtype() == torch::kFloat32);
   const int rows = logits.size(0), V = logits.size(1);
   CHK(V <= 8192);
-  auto out = torch::empty({rows}, logits.options().dtype(torch::kLong));
+  torch::Tensor out;
+  if (out_tok.has_value()) {   // write the next-token feed buffer in place
+    CHK(out_tok->dtype() == torch::kLong && out_tok->is_contiguous() &&
+        out_tok->numel() == rows);
+    out = *out_tok;
+  } else {
+    out = torch::empty({rows}, logits.options().dtype(torch::kLong));
+  }
+  long* seqp = nullptr;
+  const long* ptrp = nullptr;
+  int S = 0;
+  if (seq.has_value()) {
+    CHK(seq_ptr.has_value() && seq_ptr->dtype() == torch::kLong);
+    CHK(seq->dtype() == torch::kLong && seq->is_contiguous() &&
+        seq->size(0) == rows && seq->dim() == 2);
+    seqp = seq->data_ptr<long>();
+    ptrp = seq_ptr->data_ptr<long>();
+    S = (int)seq->size(1);
+  }
   hipLaunchKernelGGL(sample_topk_gumbel_kernel, dim3(rows), dim3(256),
                      0, cur_stream(),
                      logits.data_ptr<float>(),
                      noise.contiguous().data_ptr<float>(),
-                     out.data_ptr<long>(), V, (int)k,
+                     out.data_ptr<long>(), seqp, ptrp, S, V, (int)k,
                      (float)(1.0 / temperature));
   return out;
 }
@@ -3651,6 +3677,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("resls_bwd", &resls_bwd, "fused residual + per-channel scale bwd");
   m.def("quant_fp8", &quant_fp8, "bf16 -> e4m3 one-pass quantize");
   m.def("sample_topk_gumbel", &sample_topk_gumbel,
+        py::arg("logits"), py::arg("noise"), py::arg("k"),
+        py::arg("temperature"), py::arg("out_tok") = std::nullopt,
+        py::arg("seq") = std::nullopt, py::arg("seq_ptr") = std::nullopt,
         "fused top-k threshold + gumbel argmax sampling");
   m.def("amax_bf16", &amax_bf16, "abs-max of a bf16 tensor (one pass)");
   m.def("sk2", &sk2,
