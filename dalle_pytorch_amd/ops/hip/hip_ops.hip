@@ -181,6 +181,12 @@ void fa_fwd_d64_kernel(
     }
   }
 
+  // exp2-domain softmax: exp(x) lowers to v_exp_f32(x*log2e) on gfx9+, so
+  // folding log2e into the score scale (a multiply that happens anyway)
+  // saves one VALU mul per score element in kernels measured 92-96%
+  // VALU-bound (profiles/pmc_final_r2.txt); lse converts back to natural
+  // log once per row at the write (the external contract is unchanged)
+  const float scl2 = scale * 1.44269504088896f;
   float m_run = NEG_INF;   // per-q-row online-softmax state (q = lq)
   float l_run = 0.f;
   f32x4 acc[4] = {f32x4{0, 0, 0, 0}, f32x4{0, 0, 0, 0},
@@ -333,7 +339,7 @@ void fa_fwd_d64_kernel(
                  (tmap_row != nullptr && tile_full(kt)))));
     if (interior) {
       #pragma unroll
-      for (int i = 0; i < 16; ++i) s16[i] *= scale;
+      for (int i = 0; i < 16; ++i) s16[i] *= scl2;
     } else if (axial) {
       #pragma unroll
       for (int i = 0; i < 16; ++i) {
@@ -342,7 +348,7 @@ void fa_fwd_d64_kernel(
                   ((kg < ax_t) | (kg >= ax_klo));
         if (key_mask != nullptr && ok)
           ok &= key_mask[(long)batch * nk + ax_phys(kg, ax_t, ax_logS, ax_axis)];
-        s16[i] = ok ? s16[i] * scale : NEG_INF;
+        s16[i] = ok ? s16[i] * scl2 : NEG_INF;
       }
     } else {
       #pragma unroll
@@ -353,7 +359,7 @@ void fa_fwd_d64_kernel(
         if (key_mask != nullptr && ok) ok &= key_mask[(long)batch * nk + kg];
         if (static_mask != nullptr && ok)
           ok &= Mtile[wave * 16 + lq][kg - kbase] != 0;
-        s16[i] = ok ? s16[i] * scale : NEG_INF;
+        s16[i] = ok ? s16[i] * scl2 : NEG_INF;
       }
     }
 
@@ -366,14 +372,14 @@ void fa_fwd_d64_kernel(
 
     // defer-max (guide T13): if no row's max grew by more than 8, keep the
     // old running max — P stays bounded by e^8 and the O-rescale is skipped
-    const bool defer = __all(mt_part <= m_run + 8.f);
+    const bool defer = __all(mt_part <= m_run + 11.5416f);  // 8 nats
     const float m_new = defer ? m_run : fmaxf(m_run, mt_part);
 
     float lsum = 0.f;
     float p16[16];
     #pragma unroll
     for (int i = 0; i < 16; ++i) {
-      p16[i] = (s16[i] == NEG_INF) ? 0.f : __expf(s16[i] - m_new);
+      p16[i] = (s16[i] == NEG_INF) ? 0.f : exp2f(s16[i] - m_new);
       lsum += p16[i];
     }
     lsum += __shfl_xor(lsum, 16);
@@ -381,7 +387,7 @@ void fa_fwd_d64_kernel(
 
     float alpha = 1.f;
     if (!defer) {
-      alpha = (m_run == NEG_INF) ? 0.f : __expf(m_run - m_new);
+      alpha = (m_run == NEG_INF) ? 0.f : exp2f(m_run - m_new);
       if (m_new != NEG_INF) m_run = m_new;
     }
     l_run = l_run * alpha + lsum;
@@ -445,7 +451,8 @@ void fa_fwd_d64_kernel(
     }
   }
   if (grp == 0 && qrow < nq) {
-    lse[(long)bh * nq + qphys] = (l_run > 0.f) ? m_run + __logf(l_run) : NEG_INF;
+    lse[(long)bh * nq + qphys] = (l_run > 0.f)
+        ? (m_run + __log2f(l_run)) * 0.69314718055995f : NEG_INF;
   }
 }
 
@@ -544,6 +551,7 @@ void fa8_fwd_d64_kernel(
     }
   }
 
+  const float scl2 = scale * 1.44269504088896f;
   float m_run = NEG_INF, l_run = 0.f;
   f32x16 acc_o[2] = {};
   const int wave_qmax = q0 + wave * 32 + 31 + diag;   // causal bound
@@ -632,7 +640,7 @@ void fa8_fwd_d64_kernel(
         #pragma unroll
         for (int s = 0; s < 2; ++s)
           #pragma unroll
-          for (int r = 0; r < 16; ++r) s32[16 * s + r] = acc_s[s][r] * scale;
+          for (int r = 0; r < 16; ++r) s32[16 * s + r] = acc_s[s][r] * scl2;
       } else {
         #pragma unroll
         for (int s = 0; s < 2; ++s) {
@@ -649,7 +657,7 @@ void fa8_fwd_d64_kernel(
               const int kph = axial ? ax_phys(key, ax_t, ax_logS, ax_axis) : key;
               ok &= key_mask[(long)batch * nk + kph];
             }
-            s32[16 * s + r] = ok ? acc_s[s][r] * scale : NEG_INF;
+            s32[16 * s + r] = ok ? acc_s[s][r] * scl2 : NEG_INF;
           }
         }
       }
@@ -660,21 +668,21 @@ void fa8_fwd_d64_kernel(
       for (int i = 0; i < 32; ++i) mt = fmaxf(mt, s32[i]);
       mt = fmaxf(mt, partner_f32(mt, half));
 
-      const bool defer = __all(mt <= m_run + 8.f);
+      const bool defer = __all(mt <= m_run + 11.5416f);
       const float m_new = defer ? m_run : fmaxf(m_run, mt);
 
       float p32[32];
       float lsum = 0.f;
       #pragma unroll
       for (int i = 0; i < 32; ++i) {
-        p32[i] = (s32[i] == NEG_INF) ? 0.f : __expf(s32[i] - m_new);
+        p32[i] = (s32[i] == NEG_INF) ? 0.f : exp2f(s32[i] - m_new);
         lsum += p32[i];
       }
       lsum += partner_f32(lsum, half);
 
       float alpha = 1.f;
       if (!defer) {
-        alpha = (m_run == NEG_INF) ? 0.f : __expf(m_run - m_new);
+        alpha = (m_run == NEG_INF) ? 0.f : exp2f(m_run - m_new);
         if (m_new != NEG_INF) m_run = m_new;
         // alpha is per q (= lane&31); the PV accumulator rows are crow(r)
         // -> broadcast through this wave's 32-slot LDS row
@@ -756,7 +764,8 @@ void fa8_fwd_d64_kernel(
     }
   }
   if (lane < 32 && qrow < nq) {
-    lse[(long)bh * nq + qphys] = (l_run > 0.f) ? m_run + __logf(l_run) : NEG_INF;
+    lse[(long)bh * nq + qphys] = (l_run > 0.f)
+        ? (m_run + __log2f(l_run)) * 0.69314718055995f : NEG_INF;
   }
 }
 
@@ -832,6 +841,7 @@ void fa_bwd_dq_kernel(
   const int ax_klo = (axial && qrow < nq && qrow >= ax_t)
       ? ax_t + (((qrow - ax_t) >> ax_logS) << ax_logS) : 0x7fffffff;
   bf16x8 qfrag[2], dofrag[2];
+  const float scl2 = scale * 1.44269504088896f;
   float lse_q = 0.f, D_q = 0.f;
   {
     const bool qok = qrow < nq;
@@ -843,7 +853,8 @@ void fa_bwd_dq_kernel(
                       : bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
     }
     if (qok) {
-      lse_q = lse[(long)bh * nq + qphys];
+      // exp2-domain: p = exp2(s*scl2 - lse*log2e); ds keeps natural scale
+      lse_q = lse[(long)bh * nq + qphys] * 1.44269504088896f;
       D_q = Dv[(long)bh * nq + qphys];
     }
   }
@@ -986,7 +997,7 @@ void fa_bwd_dq_kernel(
     if (interior) {
       #pragma unroll
       for (int i = 0; i < 16; ++i) {
-        const float p = __expf(s16[i] * scale - lse_q);
+        const float p = exp2f(s16[i] * scl2 - lse_q);
         ds16[i] = p * (dp16[i] - D_q) * scale;
       }
     } else if (axial) {
@@ -997,7 +1008,7 @@ void fa_bwd_dq_kernel(
                   ((kg < ax_t) | (kg >= ax_klo));
         if (key_mask != nullptr && ok)
           ok &= key_mask[(long)batch * nk + ax_phys(kg, ax_t, ax_logS, ax_axis)];
-        const float p = ok ? __expf(s16[i] * scale - lse_q) : 0.f;
+        const float p = ok ? exp2f(s16[i] * scl2 - lse_q) : 0.f;
         ds16[i] = p * (dp16[i] - D_q) * scale;
       }
     } else {
@@ -1009,7 +1020,7 @@ void fa_bwd_dq_kernel(
         if (key_mask != nullptr && ok) ok &= key_mask[(long)batch * nk + kg];
         if (static_mask != nullptr && ok)
           ok &= Mtile[wave * 16 + lq][kg - kbase] != 0;
-        const float p = ok ? __expf(s16[i] * scale - lse_q) : 0.f;
+        const float p = ok ? exp2f(s16[i] * scl2 - lse_q) : 0.f;
         ds16[i] = p * (dp16[i] - D_q) * scale;
       }
     }
@@ -1233,6 +1244,7 @@ void fa_bwd_dkv_kernel(
     // B = Q/dO columns from LDS. Output [M=16 keys, N=16 q]: C row
     // grp*4+r = key within the wave's 16, C col = lane&15 = q in subtile.
     __builtin_amdgcn_s_setprio(1);
+    const float scl2 = scale * 1.44269504088896f;
     f32x4 st4[4], dpt4[4];
     #pragma unroll
     for (int mt = 0; mt < 4; ++mt) {
@@ -1263,11 +1275,11 @@ void fa_bwd_dkv_kernel(
       for (int mt = 0; mt < 4; ++mt) {
         const int qg = qbase + mt * 16 + lq;
         const int qph = axial ? ax_phys(qg, ax_t, ax_logS, ax_axis) : qg;
-        const float l = lse[(long)bh * nq + qph];
+        const float l = lse[(long)bh * nq + qph] * 1.44269504088896f;
         const float Dq = Dv[(long)bh * nq + qph];
         #pragma unroll
         for (int r = 0; r < 4; ++r) {
-          const float p = __expf(st4[mt][r] * scale - l);
+          const float p = exp2f(st4[mt][r] * scl2 - l);
           const float ds = p * (dpt4[mt][r] - Dq) * scale;
           const int cc = (mt * 16 + lq) ^ (grp << 3);   // swz by row>>2
           Pt[wave][grp * 4 + r][cc] = f2bf(p);
@@ -1279,7 +1291,8 @@ void fa_bwd_dkv_kernel(
       for (int mt = 0; mt < 4; ++mt) {
         const int qg = qbase + mt * 16 + lq;
         const int qph = (qg < nq) ? ax_phys(qg, ax_t, ax_logS, ax_axis) : 0;
-        const float l = (qg < nq) ? lse[(long)bh * nq + qph] : 0.f;
+        const float l = (qg < nq)
+            ? lse[(long)bh * nq + qph] * 1.44269504088896f : 0.f;
         const float Dq = (qg < nq) ? Dv[(long)bh * nq + qph] : 0.f;
         #pragma unroll
         for (int r = 0; r < 4; ++r) {
@@ -1289,7 +1302,7 @@ void fa_bwd_dkv_kernel(
             ok &= key_mask[(long)batch * nk + ax_phys(key, ax_t, ax_logS, ax_axis)];
           float p = 0.f, ds = 0.f;
           if (ok) {
-            p = __expf(st4[mt][r] * scale - l);
+            p = exp2f(st4[mt][r] * scl2 - l);
             ds = p * (dpt4[mt][r] - Dq) * scale;
           }
           const int cc = (mt * 16 + lq) ^ (grp << 3);
@@ -1311,9 +1324,9 @@ void fa_bwd_dkv_kernel(
             ok &= Mtile[mt * 16 + lq][key - k0] != 0;
           float p = 0.f, ds = 0.f;
           if (ok) {
-            const float l = lse[(long)bh * nq + qg];
+            const float l = lse[(long)bh * nq + qg] * 1.44269504088896f;
             const float Dq = Dv[(long)bh * nq + qg];
-            p = __expf(st4[mt][r] * scale - l);
+            p = exp2f(st4[mt][r] * scl2 - l);
             ds = p * (dpt4[mt][r] - Dq) * scale;
           }
           const int cc = (mt * 16 + lq) ^ (grp << 3);   // swz by row>>2
@@ -2370,7 +2383,7 @@ void fa_decode_part_kernel(
       kc[cbase] = f2bf(kv);
       vc[cbase] = f2bf(vv);
     }
-    qs[d] = qv * scale;
+    qs[d] = qv * (scale * 1.44269504088896f);
     ksn[d] = kv;
     vsn[d] = vv;
   }
@@ -2417,7 +2430,7 @@ void fa_decode_part_kernel(
   float lsum = 0.f;
   for (int i = tid; i < chunk; i += 256) {
     const float p = (Pl[i] == NEG_INF || m == NEG_INF)
-        ? 0.f : __expf(Pl[i] - m);
+        ? 0.f : exp2f(Pl[i] - m);
     Pl[i] = p;
     lsum += p;
   }
@@ -2504,7 +2517,7 @@ void fa_decode_part_list_kernel(
       kc[cbase] = f2bf(kv);
       vc[cbase] = f2bf(vv);
     }
-    qs[wave][d] = qv * scale;
+    qs[wave][d] = qv * (scale * 1.44269504088896f);
     ksn[wave][d] = kv;
     vsn[wave][d] = vv;
   }
@@ -2543,7 +2556,7 @@ void fa_decode_part_list_kernel(
   float m = dot;
   #pragma unroll
   for (int s = 32; s > 0; s >>= 1) m = fmaxf(m, __shfl_xor(m, s));
-  const float p = (dot == NEG_INF || m == NEG_INF) ? 0.f : __expf(dot - m);
+  const float p = (dot == NEG_INF || m == NEG_INF) ? 0.f : exp2f(dot - m);
   float lsum = p;
   #pragma unroll
   for (int s = 32; s > 0; s >>= 1) lsum += __shfl_xor(lsum, s);
@@ -2628,7 +2641,7 @@ void fa_decode_one_kernel(
     const long cbase = (((long)bi * h + head) * N + off) * 64 + d;
     kc[cbase] = f2bf(kv);
     vc[cbase] = f2bf(vv);
-    qs[wave][d] = qv * scale;
+    qs[wave][d] = qv * (scale * 1.44269504088896f);
     ksn[wave][d] = kv;
     vsn[wave][d] = vv;
   }
@@ -2675,7 +2688,7 @@ void fa_decode_one_kernel(
 
   float l_loc = 0.f;
   for (int j = lane; j < jn; j += 64) {
-    const float p = __expf(Pl[wave][j] - m_loc);
+    const float p = exp2f(Pl[wave][j] - m_loc);
     Pl[wave][j] = p;
     l_loc += p;
   }
@@ -2739,7 +2752,7 @@ void fa_decode_combine_kernel(
     for (int z = 0; z < KS; ++z) {
       const float mz = s0[z * 66];
       if (mz == NEG_INF) continue;
-      const float r = __expf(mz - m);
+      const float r = exp2f(mz - m);
       den += s0[z * 66 + 1] * r;
       acc += s0[z * 66 + 2 + lane] * r;
     }
