@@ -66,6 +66,13 @@ constexpr int KPAD = 72;        // K tile row stride (8-elem pad -> 2-way banks)
 constexpr int VPAD = 40;        // V^T tile row stride
 constexpr int PPAD = 40;        // P tile row stride
 
+
+// raw v_exp_f32 (base-2): libm exp2f lowers to a guarded cmp/cndmask/
+// v_exp/v_ldexp sequence (ISA-checked) that costs MORE than __expf's
+// mul+exp; the amdgcn builtin is the single instruction. Softmax args are
+// <= 0 and > -20000, far from the guard range.
+DEVFN float fexp2(float x) { return __builtin_amdgcn_exp2f(x); }
+
 DEVFN bf16x8 frag_from_lds(const short* base) {
   return *reinterpret_cast<const bf16x8*>(base);  // ds_read_b128
 }
@@ -379,7 +386,7 @@ void fa_fwd_d64_kernel(
     float p16[16];
     #pragma unroll
     for (int i = 0; i < 16; ++i) {
-      p16[i] = (s16[i] == NEG_INF) ? 0.f : exp2f(s16[i] - m_new);
+      p16[i] = (s16[i] == NEG_INF) ? 0.f : fexp2(s16[i] - m_new);
       lsum += p16[i];
     }
     lsum += __shfl_xor(lsum, 16);
@@ -387,7 +394,7 @@ void fa_fwd_d64_kernel(
 
     float alpha = 1.f;
     if (!defer) {
-      alpha = (m_run == NEG_INF) ? 0.f : exp2f(m_run - m_new);
+      alpha = (m_run == NEG_INF) ? 0.f : fexp2(m_run - m_new);
       if (m_new != NEG_INF) m_run = m_new;
     }
     l_run = l_run * alpha + lsum;
@@ -675,14 +682,14 @@ void fa8_fwd_d64_kernel(
       float lsum = 0.f;
       #pragma unroll
       for (int i = 0; i < 32; ++i) {
-        p32[i] = (s32[i] == NEG_INF) ? 0.f : exp2f(s32[i] - m_new);
+        p32[i] = (s32[i] == NEG_INF) ? 0.f : fexp2(s32[i] - m_new);
         lsum += p32[i];
       }
       lsum += partner_f32(lsum, half);
 
       float alpha = 1.f;
       if (!defer) {
-        alpha = (m_run == NEG_INF) ? 0.f : exp2f(m_run - m_new);
+        alpha = (m_run == NEG_INF) ? 0.f : fexp2(m_run - m_new);
         if (m_new != NEG_INF) m_run = m_new;
         // alpha is per q (= lane&31); the PV accumulator rows are crow(r)
         // -> broadcast through this wave's 32-slot LDS row
@@ -997,7 +1004,7 @@ void fa_bwd_dq_kernel(
     if (interior) {
       #pragma unroll
       for (int i = 0; i < 16; ++i) {
-        const float p = exp2f(s16[i] * scl2 - lse_q);
+        const float p = fexp2(s16[i] * scl2 - lse_q);
         ds16[i] = p * (dp16[i] - D_q) * scale;
       }
     } else if (axial) {
@@ -1008,7 +1015,7 @@ void fa_bwd_dq_kernel(
                   ((kg < ax_t) | (kg >= ax_klo));
         if (key_mask != nullptr && ok)
           ok &= key_mask[(long)batch * nk + ax_phys(kg, ax_t, ax_logS, ax_axis)];
-        const float p = ok ? exp2f(s16[i] * scl2 - lse_q) : 0.f;
+        const float p = ok ? fexp2(s16[i] * scl2 - lse_q) : 0.f;
         ds16[i] = p * (dp16[i] - D_q) * scale;
       }
     } else {
@@ -1020,7 +1027,7 @@ void fa_bwd_dq_kernel(
         if (key_mask != nullptr && ok) ok &= key_mask[(long)batch * nk + kg];
         if (static_mask != nullptr && ok)
           ok &= Mtile[wave * 16 + lq][kg - kbase] != 0;
-        const float p = ok ? exp2f(s16[i] * scl2 - lse_q) : 0.f;
+        const float p = ok ? fexp2(s16[i] * scl2 - lse_q) : 0.f;
         ds16[i] = p * (dp16[i] - D_q) * scale;
       }
     }
@@ -1279,7 +1286,7 @@ void fa_bwd_dkv_kernel(
         const float Dq = Dv[(long)bh * nq + qph];
         #pragma unroll
         for (int r = 0; r < 4; ++r) {
-          const float p = exp2f(st4[mt][r] * scl2 - l);
+          const float p = fexp2(st4[mt][r] * scl2 - l);
           const float ds = p * (dpt4[mt][r] - Dq) * scale;
           const int cc = (mt * 16 + lq) ^ (grp << 3);   // swz by row>>2
           Pt[wave][grp * 4 + r][cc] = f2bf(p);
@@ -1302,7 +1309,7 @@ void fa_bwd_dkv_kernel(
             ok &= key_mask[(long)batch * nk + ax_phys(key, ax_t, ax_logS, ax_axis)];
           float p = 0.f, ds = 0.f;
           if (ok) {
-            p = exp2f(st4[mt][r] * scl2 - l);
+            p = fexp2(st4[mt][r] * scl2 - l);
             ds = p * (dpt4[mt][r] - Dq) * scale;
           }
           const int cc = (mt * 16 + lq) ^ (grp << 3);
@@ -1326,7 +1333,7 @@ void fa_bwd_dkv_kernel(
           if (ok) {
             const float l = lse[(long)bh * nq + qg] * 1.44269504088896f;
             const float Dq = Dv[(long)bh * nq + qg];
-            p = exp2f(st4[mt][r] * scl2 - l);
+            p = fexp2(st4[mt][r] * scl2 - l);
             ds = p * (dpt4[mt][r] - Dq) * scale;
           }
           const int cc = (mt * 16 + lq) ^ (grp << 3);   // swz by row>>2
@@ -2430,7 +2437,7 @@ void fa_decode_part_kernel(
   float lsum = 0.f;
   for (int i = tid; i < chunk; i += 256) {
     const float p = (Pl[i] == NEG_INF || m == NEG_INF)
-        ? 0.f : exp2f(Pl[i] - m);
+        ? 0.f : fexp2(Pl[i] - m);
     Pl[i] = p;
     lsum += p;
   }
@@ -2556,7 +2563,7 @@ void fa_decode_part_list_kernel(
   float m = dot;
   #pragma unroll
   for (int s = 32; s > 0; s >>= 1) m = fmaxf(m, __shfl_xor(m, s));
-  const float p = (dot == NEG_INF || m == NEG_INF) ? 0.f : exp2f(dot - m);
+  const float p = (dot == NEG_INF || m == NEG_INF) ? 0.f : fexp2(dot - m);
   float lsum = p;
   #pragma unroll
   for (int s = 32; s > 0; s >>= 1) lsum += __shfl_xor(lsum, s);
@@ -2688,7 +2695,7 @@ void fa_decode_one_kernel(
 
   float l_loc = 0.f;
   for (int j = lane; j < jn; j += 64) {
-    const float p = exp2f(Pl[wave][j] - m_loc);
+    const float p = fexp2(Pl[wave][j] - m_loc);
     Pl[wave][j] = p;
     l_loc += p;
   }
@@ -2752,7 +2759,7 @@ void fa_decode_combine_kernel(
     for (int z = 0; z < KS; ++z) {
       const float mz = s0[z * 66];
       if (mz == NEG_INF) continue;
-      const float r = exp2f(mz - m);
+      const float r = fexp2(mz - m);
       den += s0[z * 66 + 1] * r;
       acc += s0[z * 66 + 2 + lane] * r;
     }
