@@ -126,3 +126,28 @@ def test_fused_decode_kernels_match_torch_path():
     rel = (a[valid].float() - b[valid].float()).abs().max() / \
         b[valid].float().abs().max()
     assert rel < 5e-2, rel.item()
+
+
+def test_sk2_pack_layout():
+    """The sk2 packed-weight layout must match the kernel's index math:
+    element (lane, e) of chunk (nt, kc) is w[nt*16 + (lane&15),
+    kc*32 + (lane>>4)*8 + e] (hip_ops.hip sk2_kernel). Pure-CPU check of
+    the reshape/permute formula used by FastDecoder._sk2_pack."""
+    N, K = 64, 1024
+    w = torch.arange(N * K, dtype=torch.float32).reshape(N, K)
+    pk = (w.reshape(N // 16, 16, K // 32, 4, 8)
+           .permute(0, 2, 3, 1, 4).contiguous())    # [nt][kc][kg][col][e]
+    flat = pk.reshape(N // 16, K // 32, 64, 8)      # lane = kg*16 + col
+    for nt, kc, lane, e in [(0, 0, 0, 0), (1, 3, 17, 5), (3, 31, 63, 7),
+                            (2, 10, 48, 2)]:
+        col, kg = lane & 15, lane >> 4
+        assert flat[nt, kc, lane, e] == w[nt * 16 + col, kc * 32 + kg * 8 + e]
+
+
+def test_sk2_pack_gates():
+    """_sk2_pack returns None for shapes the kernel contract rejects
+    (K % 1024, N % 32) and for non-bf16/non-cuda tensors (CPU here)."""
+    from dalle_pytorch_amd.engine.decode import FastDecoder
+    assert FastDecoder._sk2_pack(torch.zeros(64, 512)) is None     # K % 1024
+    assert FastDecoder._sk2_pack(torch.zeros(48, 1024)) is None    # N % 32
+    assert FastDecoder._sk2_pack(torch.zeros(64, 1024)) is None    # cpu/fp32
