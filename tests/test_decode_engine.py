@@ -151,3 +151,33 @@ def test_sk2_pack_gates():
     assert FastDecoder._sk2_pack(torch.zeros(64, 512)) is None     # K % 1024
     assert FastDecoder._sk2_pack(torch.zeros(48, 1024)) is None    # N % 32
     assert FastDecoder._sk2_pack(torch.zeros(64, 1024)) is None    # cpu/fp32
+
+
+@pytest.mark.gpu
+def test_sk2_decode_path_matches_unpacked():
+    """In-situ sk2 coverage: at batch 16 and dim 1024 every sk2 call site
+    engages (qkv/out mode 0, geglu-fused ff1 mode 1, fp32 head mode 2).
+    Parity against the same engine with sk2 disabled."""
+    torch.manual_seed(9)
+    d = tiny_dalle(attn_types=('axial_row', 'axial_col'), depth=2, dim=1024,
+                   heads=16, dim_head=64).cuda().eval()
+    text = torch.randint(1, 50, (16, 8), device='cuda')
+    token = torch.randint(0, 64, (16,), device='cuda')
+
+    dec_s = FastDecoder(d, batch_size=16, dtype=torch.bfloat16)
+    assert dec_s._fused_decode and dec_s._sk2_on
+    assert dec_s.states[0].w.get('qkv_pk') is not None
+    assert dec_s.head_w.get('w_img_pk') is not None
+    dec_u = FastDecoder(d, batch_size=16, dtype=torch.bfloat16)
+    dec_u._sk2_on = False
+
+    with torch.no_grad():
+        a = dec_s.prefill(text)
+        b = dec_u.prefill(text)
+        for _ in range(4):
+            a = dec_s.step(token)
+            b = dec_u.step(token)
+    valid = b > -1e30
+    rel = (a[valid].float() - b[valid].float()).abs().max() / \
+        b[valid].float().abs().max()
+    assert rel < 5e-2, rel.item()
