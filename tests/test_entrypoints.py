@@ -94,3 +94,46 @@ def test_train_dalle_grad_accum(tmp_path):
     assert (tmp_path / 'log.jsonl').exists()
     records = [json.loads(l) for l in (tmp_path / 'log.jsonl').read_text().splitlines()]
     assert any('loss' in r for r in records)
+
+
+def test_serve_endpoint(tmp_path):
+    """HTTP serving front end (examples/serve.py): health + generate round
+    trip on a tiny checkpoint through Starlette's in-process TestClient."""
+    import base64
+    import io
+    import train_vae
+    import train_dalle
+    train_vae.main([
+        '--image_size', '32', '--num_tokens', '32', '--num_layers', '2',
+        '--emb_dim', '16', '--hidden_dim', '8', '--batch_size', '2',
+        '--epochs', '1', '--stop_after_steps', '1',
+        '--output_dir', str(tmp_path)])
+    train_dalle.main([
+        '--vae_path', str(tmp_path / 'vae-final.pt'), '--synthetic',
+        '--dim', '32', '--depth', '1', '--heads', '2', '--dim_head', '16',
+        '--text_seq_len', '8', '--batch_size', '2', '--epochs', '1',
+        '--stop_after_steps', '1', '--output_dir', str(tmp_path)])
+
+    sys.path.insert(0, str(REPO / 'examples'))
+    import serve
+    from dalle_pytorch_amd.utils.checkpoint import (
+        build_dalle_from_checkpoint, load_dalle_checkpoint)
+    from dalle_pytorch_amd.utils.tokenizer import tokenizer as tok
+    ckpt = load_dalle_checkpoint(tmp_path / 'dalle.pt')
+    dalle, _ = build_dalle_from_checkpoint(ckpt)
+    app = serve.build_app(dalle.eval(), torch.device('cpu'), tok)
+
+    from starlette.testclient import TestClient
+    client = TestClient(app)
+    r = client.get('/health')
+    assert r.status_code == 200 and r.json()['status'] == 'ok'
+    r = client.post('/generate', json={
+        'text': 'a tiny test', 'num_images': 1, 'seed': 0})
+    assert r.status_code == 200, r.text
+    body = r.json()
+    assert body['format'] == 'png' and len(body['images']) == 1
+    from PIL import Image
+    img = Image.open(io.BytesIO(base64.b64decode(body['images'][0])))
+    assert img.size == (32, 32)
+    r = client.post('/generate', json={'text': 'x', 'num_images': 999})
+    assert r.status_code == 400
