@@ -324,15 +324,42 @@ class FastDecoder:
             return torch.cat((out, tail), dim=-1) if tail.shape[-1] else out
         return one(q), one(k), one(v)
 
+    def _proj(self, x, st_w, key, bias32_key, bias_key, mode=0):
+        """Single-token projection: sk2 on the packed weight when profitable,
+        else the measured-best hipBLASLt form for the shape."""
+        w = st_w[key]
+        rows = x.numel() // x.shape[-1]
+        pk = st_w.get(key[:-2] + '_pk') if key.endswith('_w') else \
+            st_w.get(key + '_pk')
+        if self._sk2_ok(rows, pk, mode, w.shape[1]):
+            return self._sk2(x, pk, st_w.get(bias32_key), w.shape[0],
+                             w.shape[1], mode)
+        return None
+
+    def _ff_decode(self, st, y):
+        """Single-token feed-forward: geglu-fused sk2 ff1 when available,
+        hipBLASLt ff2 (measured faster than sk2 at K=4096)."""
+        rows = y.numel() // y.shape[-1]
+        if self._sk2_ok(rows, st.w['ff1_pk'], 1):
+            y = self._sk2(y, st.w['ff1_pk'], st.w['ff1_b32'],
+                          st.w['ff1_w'].shape[0], st.w['ff1_w'].shape[1], 1)
+        else:
+            from dalle_pytorch_amd.ops import geglu
+            y = geglu(self._lin(y, st.w['ff1_w'], st.w['ff1_b32'],
+                                st.w['ff1_b']))
+        if self._sk2_ok(rows, st.w['ff2_pk'], 0, st.w['ff2_w'].shape[1]):
+            return self._sk2(y, st.w['ff2_pk'], st.w['ff2_b32'],
+                             st.w['ff2_w'].shape[0], st.w['ff2_w'].shape[1],
+                             0)
+        return self._lin(y, st.w['ff2_w'], st.w['ff2_b32'], st.w['ff2_b'])
+
     def _attn(self, st, x, offset_t, n):
         """x [b, n, dim] at positions offset..offset+n-1 (prefill has
         offset 0; decode has n == 1)."""
         leaf = st.leaf
         h, d = leaf.heads, leaf.dim_head
-        if n == 1 and self._sk2_ok(x.numel() // x.shape[-1], st.w['qkv_pk'], 0):
-            qkv = self._sk2(x, st.w['qkv_pk'], None,
-                            st.w['qkv'].shape[0], st.w['qkv'].shape[1], 0)
-        else:
+        qkv = self._proj(x, st.w, 'qkv', None, None) if n == 1 else None
+        if qkv is None:
             qkv = (self._lin_t(x, st.w['qkv_T'], None) if n == 1
                    else F.linear(x, st.w['qkv']))
         if n == 1 and self._fused_decode:
@@ -343,11 +370,9 @@ class FastDecoder:
                 self.sin if self.rotary else None,
                 offset_t, st.pattern, leaf.scale,
                 st.live, st.live_cnt).view(self.b, 1, h * d)
-            if self._sk2_ok(self.b, st.w['out_pk'], 0):
-                return self._sk2(out, st.w['out_pk'], st.w['out_b32'],
-                                 st.w['out_w'].shape[0],
-                                 st.w['out_w'].shape[1], 0)
-            return self._lin_t(out, st.w['out_T'], st.w['out_b'])
+            y = self._proj(out, st.w, 'out_w', 'out_b32', 'out_b')
+            return y if y is not None else \
+                self._lin_t(out, st.w['out_T'], st.w['out_b'])
         q, k, v = (t.reshape(self.b, n, h, d).permute(0, 2, 1, 3)
                    for t in qkv.chunk(3, dim=-1))
         if self.rotary:
@@ -447,21 +472,7 @@ class FastDecoder:
         if st.is_attn:
             y = self._attn(st, y, offset_t, n)
         elif n == 1:
-            rows = y.numel() // y.shape[-1]
-            if self._sk2_ok(rows, st.w['ff1_pk'], 1):
-                y = self._sk2(y, st.w['ff1_pk'], st.w['ff1_b32'],
-                              st.w['ff1_w'].shape[0],
-                              st.w['ff1_w'].shape[1], 1)
-            else:
-                from dalle_pytorch_amd.ops import geglu
-                y = geglu(self._lin_t(y, st.w['ff1_T'], st.w['ff1_b']))
-            if self._sk2_ok(rows, st.w['ff2_pk'], 0,
-                            st.w['ff2_w'].shape[1]):
-                y = self._sk2(y, st.w['ff2_pk'], st.w['ff2_b32'],
-                              st.w['ff2_w'].shape[0],
-                              st.w['ff2_w'].shape[1], 0)
-            else:
-                y = self._lin_t(y, st.w['ff2_T'], st.w['ff2_b'])
+            y = self._ff_decode(st, y)
         else:
             from dalle_pytorch_amd.ops import geglu
             y = F.linear(y, st.w['ff1_w'], st.w['ff1_b'])
@@ -482,18 +493,7 @@ class FastDecoder:
         """Branch compute after the (fused) LN+shift prelude: attention or FF."""
         if st.is_attn:
             return self._attn(st, z, offset_t, 1)
-        rows = z.numel() // z.shape[-1]
-        if self._sk2_ok(rows, st.w['ff1_pk'], 1):
-            y = self._sk2(z, st.w['ff1_pk'], st.w['ff1_b32'],
-                          st.w['ff1_w'].shape[0], st.w['ff1_w'].shape[1], 1)
-        else:
-            from dalle_pytorch_amd.ops import geglu
-            y = geglu(self._lin(z, st.w['ff1_w'], st.w['ff1_b32'],
-                                st.w['ff1_b']))
-        if self._sk2_ok(rows, st.w['ff2_pk'], 0, st.w['ff2_w'].shape[1]):
-            return self._sk2(y, st.w['ff2_pk'], st.w['ff2_b32'],
-                             st.w['ff2_w'].shape[0], st.w['ff2_w'].shape[1], 0)
-        return self._lin(y, st.w['ff2_w'], st.w['ff2_b32'], st.w['ff2_b'])
+        return self._ff_decode(st, z)
 
     def _prelude(self, ext, stream, pend, st, off):
         """One fused kernel: apply the previous branch's residual to the
@@ -577,13 +577,10 @@ class FastDecoder:
             x = x / x.amax(dim=-1, keepdim=True)
         x = F.layer_norm(x, (x.shape[-1],), self.head_w['ln_w'],
                          self.head_w['ln_b'], d.to_logits[0].eps)
-        if self._sk2_ok(x.numel() // x.shape[-1], self.head_w['w_img_pk'], 2):
-            return self._sk2(x, self.head_w['w_img_pk'],
-                             self.head_w['b_img32'],
-                             self.head_w['w_img'].shape[0],
-                             self.head_w['w_img'].shape[1], 2)
-        return self._lin(x, self.head_w['w_img'], self.head_w['b_img32'],
-                         self.head_w['b_img'])
+        y = self._proj(x, self.head_w, 'w_img', 'b_img32', 'b_img', mode=2)
+        return y if y is not None else \
+            self._lin(x, self.head_w['w_img'], self.head_w['b_img32'],
+                      self.head_w['b_img'])
 
     # ----------------------------------------------------------- prefill
 
