@@ -164,6 +164,9 @@ def main():
 
     decoder = None
     if args.mode == 'generate':
+        if args.fp8:
+            # opt-in fp8-weight decode (sk2 e4m3 packs); dtype reported below
+            os.environ['DALLE_AMD_FP8_DECODE'] = '1'
         dalle.eval()
         if not args.eager:
             from dalle_pytorch_amd.engine import FastDecoder
@@ -212,16 +215,20 @@ def main():
             global_batch = bsz * world
             value = global_batch * image_seq_len * args.steps / elapsed
             metric, unit = 'img-tokens/sec', 'img_tokens_per_sec'
+            dtype = 'fp8-fwd/bf16' if args.fp8 else \
+                'bf16' if use_cuda else 'fp32'
         else:
             value = args.gen_batch * world * args.steps / elapsed
             metric, unit = 'gen imgs/sec', 'images_per_sec'
+            dtype = 'bf16/fp8-weights' if args.fp8 else \
+                'bf16' if use_cuda else 'fp32'
             global_batch = args.gen_batch * world
         print(json.dumps({
             'metric': metric, 'value': round(value, 2), 'unit': unit,
             'n_gpus': world, 'steps': args.steps, 'warmup': args.warmup,
             'ms_per_step': round(ms_per_step, 2),
             'higher_is_better': True, 'scaling': 'weak',
-            'vs_baseline': None, 'dtype': 'fp8-fwd/bf16' if args.fp8 else 'bf16' if use_cuda else 'fp32',
+            'vs_baseline': None, 'dtype': dtype,
             'data': 'synthetic',
             'config': {
                 'model': f"dalle-dim{cfg['dim']}-depth{cfg['depth']}-heads{cfg['heads']}",

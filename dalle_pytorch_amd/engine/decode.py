@@ -218,6 +218,23 @@ class FastDecoder:
         return (w.reshape(N // 16, 16, K // 32, 4, 8)
                  .permute(0, 2, 3, 1, 4).contiguous())
 
+    @staticmethod
+    def _sk2_pack_fp8(w):
+        """e4m3 variant of _sk2_pack: per-tensor scale (amax/448), packed
+        as bytes in the same fragment order. Halves the weight stream of
+        the weight-bound decode GEMMs; opt-in via DALLE_AMD_FP8_DECODE=1
+        (a quality trade — logits move ~0.1-1%% per element)."""
+        N, K = w.shape
+        if N % 32 or K % 1024 or not w.is_cuda or w.dtype != torch.bfloat16:
+            return None
+        amax = w.detach().abs().amax().float().clamp(min=1e-12)
+        scale = amax / 448.0
+        q = (w.detach().float() / scale).clamp(-448., 448.) \
+            .to(torch.float8_e4m3fn)
+        pk = (q.view(torch.uint8).reshape(N // 16, 16, K // 32, 4, 8)
+               .permute(0, 2, 3, 1, 4).contiguous())
+        return pk, float(scale)
+
     def _materialize(self, st):
         """Pre-cast this branch's weights to the engine dtype once: under
         autocast the casts would otherwise be captured into the decode graph
@@ -245,6 +262,9 @@ class FastDecoder:
             w['out_w'] = cast(st.leaf.to_out[0].weight)
             w['out_T'] = castT(st.leaf.to_out[0].weight)
             w['out_pk'] = self._sk2_pack(w['out_w'])
+            if self._want_fp8_decode():
+                w['qkv_pk8'] = self._sk2_pack_fp8(w['qkv'])
+                w['out_w_pk8'] = self._sk2_pack_fp8(w['out_w'])
             w['out_b'] = cast(st.leaf.to_out[0].bias)
             w['out_b32'] = castf(st.leaf.to_out[0].bias)
         else:
@@ -260,9 +280,17 @@ class FastDecoder:
             w['ff2_w'] = cast(net[3].weight)
             w['ff2_T'] = castT(net[3].weight)
             w['ff2_pk'] = self._sk2_pack(w['ff2_w'])
+            if self._want_fp8_decode():
+                w['ff1_w_pk8'] = self._sk2_pack_fp8(w['ff1_w'])
+                w['ff2_w_pk8'] = self._sk2_pack_fp8(w['ff2_w'])
             w['ff2_b'] = cast(net[3].bias)
             w['ff2_b32'] = castf(net[3].bias)
         return w
+
+    @staticmethod
+    def _want_fp8_decode():
+        import os
+        return os.environ.get('DALLE_AMD_FP8_DECODE', '0') == '1'
 
     def _sk2_ok(self, rows, pk, mode=0, k_dim=0):
         # measured on-box (scripts/bench_sk2.py): at 128 rows the MT=8
@@ -274,13 +302,15 @@ class FastDecoder:
             return True
         return rows == 128 and mode == 1
 
-    def _sk2(self, x, pk, bias32, N, K, mode):
+    def _sk2(self, x, pk, bias32, N, K, mode, wscale=1.0):
         """Weights-streaming decode GEMM on a pre-packed tile layout with
         the epilogue (bias / geglu / fp32 head) fused — one dispatch where
-        hipBLASLt + eager epilogues took two or three (see sk2_kernel)."""
+        hipBLASLt + eager epilogues took two or three (see sk2_kernel).
+        A uint8/e4m3 pack streams fp8 weights (wscale = its amax/448)."""
         from dalle_pytorch_amd.ops.dispatch import hip_module
         rows = x.numel() // K
-        out = hip_module().sk2(x.reshape(rows, K), pk, bias32, N, K, mode)
+        out = hip_module().sk2(x.reshape(rows, K), pk, bias32, N, K, mode,
+                               wscale)
         no = N // 2 if mode == 1 else N
         return out.view(*x.shape[:-1], no)
 
@@ -329,6 +359,10 @@ class FastDecoder:
         else the measured-best hipBLASLt form for the shape."""
         w = st_w[key]
         rows = x.numel() // x.shape[-1]
+        pk8 = st_w.get(key + '_pk8')
+        if pk8 is not None and self._sk2_ok(rows, pk8[0], mode):
+            return self._sk2(x, pk8[0], st_w.get(bias32_key), w.shape[0],
+                             w.shape[1], mode, pk8[1])
         pk = st_w.get(key[:-2] + '_pk') if key.endswith('_w') else \
             st_w.get(key + '_pk')
         if self._sk2_ok(rows, pk, mode, w.shape[1]):
@@ -338,8 +372,18 @@ class FastDecoder:
 
     def _ff_decode(self, st, y):
         """Single-token feed-forward: geglu-fused sk2 ff1 when available,
-        hipBLASLt ff2 (measured faster than sk2 at K=4096)."""
+        hipBLASLt ff2 (measured faster than sk2 at K=4096). fp8 packs
+        (opt-in) take both — at half the weight bytes sk2 wins ff2 too."""
         rows = y.numel() // y.shape[-1]
+        pk8_1, pk8_2 = st.w.get('ff1_w_pk8'), st.w.get('ff2_w_pk8')
+        if pk8_1 is not None and pk8_2 is not None \
+                and self._sk2_ok(rows, pk8_1[0], 1):
+            y = self._sk2(y, pk8_1[0], st.w['ff1_b32'],
+                          st.w['ff1_w'].shape[0], st.w['ff1_w'].shape[1],
+                          1, pk8_1[1])
+            return self._sk2(y, pk8_2[0], st.w['ff2_b32'],
+                             st.w['ff2_w'].shape[0], st.w['ff2_w'].shape[1],
+                             0, pk8_2[1])
         if self._sk2_ok(rows, st.w['ff1_pk'], 1):
             y = self._sk2(y, st.w['ff1_pk'], st.w['ff1_b32'],
                           st.w['ff1_w'].shape[0], st.w['ff1_w'].shape[1], 1)

@@ -1708,13 +1708,20 @@ void skinny_cast_kernel(
 
 DEVFN float gelu_f(float x);   // defined with the geglu kernels below
 
-template <int MT, int MODE, int KS = 4>
+// F8 = true streams e4m3 weights (HALF the bytes of the weight-bound
+// stream; the non-scaled 16x16x32_fp8_fp8 MFMA runs at the bf16 rate, which
+// is irrelevant here) and converts the L2-resident bf16 activations to fp8
+// in-register (the kernel is latency-bound with idle VALU). One per-tensor
+// weight scale is folded into the epilogue; activations ride unscaled
+// (post-LN magnitudes sit far inside e4m3's +-448 range; clamped anyway).
+// Opt-in quality trade (DALLE_AMD_FP8_DECODE=1) — the bf16 path is default.
+template <int MT, int MODE, int KS = 4, bool F8 = false>
 __global__ __launch_bounds__(KS * 64, 2)
 void sk2_kernel(const short* __restrict__ x,     // [MT*16, K] bf16
                 const short* __restrict__ wp,    // packed [N/16][K/32][64][8]
                 const float* __restrict__ bias,  // [N] fp32 or null
                 void* __restrict__ out,          // [MT*16, NO]
-                int N, int K) {
+                int N, int K, float wscale = 1.f) {
   constexpr int NW = (MODE == 1) ? 2 : 1;        // weight streams per block
   // ring depth bounds the per-wave outstanding loads (the compiler drains
   // vmcnt(0) once per ring cycle, so bytes-in-flight = DEPTH * frags * 16 B);
@@ -1731,11 +1738,14 @@ void sk2_kernel(const short* __restrict__ x,     // [MT*16, K] bf16
   const int KCW = KC / KS;                       // chunks per wave
   const long tile_elems = (long)KC * 512;        // shorts per packed n-tile
 
-  const short* wb0 = wp + (long)nt * tile_elems
-                        + ((long)wave * KCW) * 512 + lane * 8;
-  const short* wb1 = (MODE == 1)
-      ? wp + ((long)nt + (N >> 5)) * tile_elems
-           + ((long)wave * KCW) * 512 + lane * 8
+  // fp8 packs are bytes at the same [nt][kc][lane][8] indexing
+  const char* wraw = reinterpret_cast<const char*>(wp);
+  const long esz = F8 ? 1 : 2;
+  const char* wb0 = wraw + ((long)nt * tile_elems
+                        + ((long)wave * KCW) * 512 + lane * 8) * esz;
+  const char* wb1 = (MODE == 1)
+      ? wraw + (((long)nt + (N >> 5)) * tile_elems
+           + ((long)wave * KCW) * 512 + lane * 8) * esz
       : nullptr;
   const short* xb = x + (long)lq * K + kg * 8 + (long)wave * KCW * 32;
 
@@ -1748,12 +1758,47 @@ void sk2_kernel(const short* __restrict__ x,     // [MT*16, K] bf16
   bf16x8 wrg[DEPTH][NW];
   bf16x8 xrg[DEPTH][MT];
 
-  // prologue: fill the ring (KCW >= DEPTH, host-enforced via K % 512 == 0)
+  const long wstep = F8 ? 512 : 1024;   // bytes per (lane-sliced) chunk
+  auto load_w = [&](const char* base, long idx) -> bf16x8 {
+    if (F8) {   // 8 e4m3 bytes in the low half of the frag registers
+      const int2 b8 = *reinterpret_cast<const int2*>(base + idx * wstep);
+      bf16x8 f{};
+      *reinterpret_cast<int2*>(&f) = b8;
+      return f;
+    }
+    return *reinterpret_cast<const bf16x8*>(base + idx * wstep);
+  };
+  // bf16 -> e4m3 in-register: 8 clamps + 4 pack-converts per fragment
+  auto to_fp8 = [&](bf16x8 v) -> long {
+    const short* s = reinterpret_cast<const short*>(&v);
+    unsigned lo = 0, hi = 0;
+    #pragma unroll
+    for (int p2 = 0; p2 < 2; ++p2) {
+      const float a = fminf(fmaxf(bf2f(s[4 * p2]), -448.f), 448.f);
+      const float b = fminf(fmaxf(bf2f(s[4 * p2 + 1]), -448.f), 448.f);
+      const float c = fminf(fmaxf(bf2f(s[4 * p2 + 2]), -448.f), 448.f);
+      const float d = fminf(fmaxf(bf2f(s[4 * p2 + 3]), -448.f), 448.f);
+      unsigned& w32 = p2 ? hi : lo;
+      w32 = __builtin_amdgcn_cvt_pk_fp8_f32(a, b, w32, false);
+      w32 = __builtin_amdgcn_cvt_pk_fp8_f32(c, d, w32, true);
+    }
+    return ((long)(unsigned long)hi << 32) | lo;
+  };
+  auto mfma_any = [&](bf16x8 wf, bf16x8 xf, f32x4 c) -> f32x4 {
+    if (F8) {
+      const long wa = *reinterpret_cast<const long*>(&wf);
+      return __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(wa, to_fp8(xf),
+                                                        c, 0, 0, 0);
+    }
+    return __builtin_amdgcn_mfma_f32_16x16x32_bf16(wf, xf, c, 0, 0, 0);
+  };
+
+  // prologue: fill the ring (KCW >= DEPTH, host-enforced via K % 1024 == 0)
   #pragma unroll
   for (int p = 0; p < DEPTH; ++p) {
-    wrg[p][0] = *reinterpret_cast<const bf16x8*>(wb0 + (long)p * 512);
+    wrg[p][0] = load_w(wb0, p);
     if (MODE == 1)
-      wrg[p][NW - 1] = *reinterpret_cast<const bf16x8*>(wb1 + (long)p * 512);
+      wrg[p][NW - 1] = load_w(wb1, p);
     #pragma unroll
     for (int mt = 0; mt < MT; ++mt)
       xrg[p][mt] = *reinterpret_cast<const bf16x8*>(xb + (long)mt * 16 * K
@@ -1766,19 +1811,17 @@ void sk2_kernel(const short* __restrict__ x,     // [MT*16, K] bf16
       __builtin_amdgcn_s_setprio(1);
       #pragma unroll
       for (int mt = 0; mt < MT; ++mt) {
-        acc[0][mt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-            wrg[p][0], xrg[p][mt], acc[0][mt], 0, 0, 0);
+        acc[0][mt] = mfma_any(wrg[p][0], xrg[p][mt], acc[0][mt]);
         if (MODE == 1)
-          acc[NW - 1][mt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              wrg[p][NW - 1], xrg[p][mt], acc[NW - 1][mt], 0, 0, 0);
+          acc[NW - 1][mt] = mfma_any(wrg[p][NW - 1], xrg[p][mt],
+                                     acc[NW - 1][mt]);
       }
       __builtin_amdgcn_s_setprio(0);
       const int nx = base + DEPTH + p;
       if (nx < KCW) {
-        wrg[p][0] = *reinterpret_cast<const bf16x8*>(wb0 + (long)nx * 512);
+        wrg[p][0] = load_w(wb0, nx);
         if (MODE == 1)
-          wrg[p][NW - 1] =
-              *reinterpret_cast<const bf16x8*>(wb1 + (long)nx * 512);
+          wrg[p][NW - 1] = load_w(wb1, nx);
         #pragma unroll
         for (int mt = 0; mt < MT; ++mt)
           xrg[p][mt] = *reinterpret_cast<const bf16x8*>(
@@ -1805,10 +1848,12 @@ void sk2_kernel(const short* __restrict__ x,     // [MT*16, K] bf16
     float v = 0.f, g = 0.f;
     #pragma unroll
     for (int s4 = 0; s4 < KS; ++s4) v += red[s4][0][mt][n][m];
+    if (F8) v *= wscale;
     const long o = (long)(mt * 16 + m) * NO + nt * 16 + n;
     if (MODE == 1) {
       #pragma unroll
       for (int s4 = 0; s4 < KS; ++s4) g += red[s4][NW - 1][mt][n][m];
+      if (F8) g *= wscale;
       if (bias != nullptr) {
         v += bias[nt * 16 + n];
         g += bias[(N >> 1) + nt * 16 + n];
@@ -3603,9 +3648,11 @@ torch::Tensor skinny_gemm(torch::Tensor x, torch::Tensor w,
 
 torch::Tensor sk2(torch::Tensor x, torch::Tensor wp,
                   std::optional<torch::Tensor> bias,
-                  long N, long K, long mode) {
+                  long N, long K, long mode, double wscale) {
   CHK(x.is_cuda() && x.dtype() == torch::kBFloat16 && x.is_contiguous());
-  CHK(wp.dtype() == torch::kBFloat16 && wp.is_contiguous());
+  const bool f8 = wp.dtype() == torch::kFloat8_e4m3fn ||
+                  wp.dtype() == torch::kUInt8;
+  CHK((wp.dtype() == torch::kBFloat16 || f8) && wp.is_contiguous());
   const long rows = x.numel() / K;
   CHK(x.size(-1) == K);
   CHK(rows == 16 || rows == 32 || rows == 64 || rows == 128);
@@ -3632,9 +3679,16 @@ torch::Tensor sk2(torch::Tensor x, torch::Tensor wp,
   // long-K bias shapes (ff2): 8-way k-split per block doubles the in-flight
   // weight stream (the 4-wave variant is HBM-latency-bound at 64 blocks)
   const bool ks8 = mode == 0 && K >= 4096 && K % 2048 == 0 && rows <= 64;
+  const float ws = (float)wscale;
   #define SK2_LAUNCH(MT, MODE, KS)                                          \
-    hipLaunchKernelGGL((sk2_kernel<MT, MODE, KS>), grid, dim3(KS * 64), 0,  \
-                       cur_stream(), xp, wpp, bp, op, Ni, Ki)
+    if (f8)                                                                 \
+      hipLaunchKernelGGL((sk2_kernel<MT, MODE, KS, true>), grid,            \
+                         dim3(KS * 64), 0, cur_stream(), xp, wpp, bp, op,   \
+                         Ni, Ki, ws);                                       \
+    else                                                                    \
+      hipLaunchKernelGGL((sk2_kernel<MT, MODE, KS, false>), grid,           \
+                         dim3(KS * 64), 0, cur_stream(), xp, wpp, bp, op,   \
+                         Ni, Ki, ws)
   #define SK2_MT(MT)                                                        \
     switch (mode) {                                                         \
       case 0: if (ks8) SK2_LAUNCH(MT, 0, 8); else SK2_LAUNCH(MT, 0, 4);     \
@@ -3703,7 +3757,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "fused top-k threshold + gumbel argmax sampling");
   m.def("amax_bf16", &amax_bf16, "abs-max of a bf16 tensor (one pass)");
   m.def("sk2", &sk2,
-        "decode skinny GEMM on packed weights (fused bias/geglu/fp32 head)");
+        py::arg("x"), py::arg("wp"), py::arg("bias"), py::arg("N"),
+        py::arg("K"), py::arg("mode"), py::arg("wscale") = 1.0,
+        "decode skinny GEMM on packed bf16/e4m3 weights "
+        "(fused bias/geglu/fp32 head)");
   m.def("skinny_gemm", &skinny_gemm,
         "skinny-M weights-streaming GEMM (decode projections)",
         py::arg("x"), py::arg("w"), py::arg("bias") = std::nullopt);

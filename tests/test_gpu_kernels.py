@@ -796,3 +796,66 @@ def test_sk2_decode_gemm_matches_linear():
     out32 = ext.sk2(x, pk, b, N, K, 2)
     assert out32.dtype == torch.float32
     torch.testing.assert_close(out32, y, rtol=3e-2, atol=3e-2)
+
+
+@pytest.mark.gpu
+def test_sk2_fp8_weights_close_to_bf16():
+    """Opt-in fp8-weight decode GEMM (sk2 e4m3 pack): per-tensor-scaled
+    quantization error only — compare against the fp32 reference computed
+    from the DEQUANTIZED weights (exact-path check) and loosely against the
+    unquantized weights (end-to-end error budget)."""
+    import dalle_pytorch_amd._hip as ext
+    from dalle_pytorch_amd.engine.decode import FastDecoder
+    torch.manual_seed(1)
+    for rows, K, N, mode in [(64, 1024, 3072, 0), (64, 1024, 8192, 1),
+                             (64, 4096, 1024, 0), (32, 1024, 1024, 0)]:
+        x = (torch.randn(rows, K, device='cuda') * 0.3).bfloat16()
+        w = (torch.randn(N, K, device='cuda') * 0.05).bfloat16()
+        b = torch.randn(N, device='cuda')
+        pk8, ws = FastDecoder._sk2_pack_fp8(w)
+        out = ext.sk2(x, pk8, b, N, K, mode, ws).float()
+        # exact reference on dequantized weights (x also rides through e4m3)
+        wdq = (w.float() / ws).clamp(-448, 448).to(torch.float8_e4m3fn) \
+            .float() * ws
+        xdq = x.float().clamp(-448, 448).to(torch.float8_e4m3fn).float()
+        y = F.linear(xdq, wdq, b)
+        if mode == 1:
+            v, g = y.chunk(2, dim=-1)
+            y = v * F.gelu(g)
+        torch.testing.assert_close(out, y, rtol=3e-2, atol=3e-2)
+        # loose end-to-end budget vs the unquantized math
+        yref = F.linear(x.float(), w.float(), b)
+        if mode == 1:
+            v, g = yref.chunk(2, dim=-1)
+            yref = v * F.gelu(g)
+        rel = (out - yref).norm() / yref.norm()
+        assert rel < 0.05, rel.item()
+
+
+@pytest.mark.gpu
+def test_fp8_decode_engine_close_to_bf16(monkeypatch):
+    """DALLE_AMD_FP8_DECODE=1: the engine runs fp8-weight sk2 GEMMs; step
+    logits stay within the fp8 error budget of the bf16 engine."""
+    monkeypatch.setenv('DALLE_AMD_FP8_DECODE', '1')
+    from dalle_pytorch_amd.engine.decode import FastDecoder
+    from tests.test_decode_engine import tiny_dalle
+    torch.manual_seed(11)
+    d = tiny_dalle(attn_types=('axial_row', 'axial_col'), depth=2, dim=1024,
+                   heads=16, dim_head=64).cuda().eval()
+    text = torch.randint(1, 50, (16, 8), device='cuda')
+    token = torch.randint(0, 64, (16,), device='cuda')
+    dec8 = FastDecoder(d, batch_size=16, dtype=torch.bfloat16)
+    assert dec8.states[0].w.get('qkv_pk8') is not None
+    monkeypatch.setenv('DALLE_AMD_FP8_DECODE', '0')
+    dec = FastDecoder(d, batch_size=16, dtype=torch.bfloat16)
+    assert dec.states[0].w.get('qkv_pk8') is None
+    with torch.no_grad():
+        a = dec8.prefill(text)
+        b = dec.prefill(text)
+        for _ in range(3):
+            a = dec8.step(token)
+            b = dec.step(token)
+    valid = b > -1e30
+    rel = (a[valid].float() - b[valid].float()).norm() / \
+        b[valid].float().norm()
+    assert rel < 0.1, rel.item()
