@@ -55,3 +55,19 @@ def main():
 
 if __name__ == '__main__':
     main()
+
+
+def fp8_compare():
+    torch.manual_seed(0)
+    from dalle_pytorch_amd.engine.decode import FastDecoder as FD
+    for rows, K, N, mode in [(64, 1024, 3072, 0), (64, 1024, 8192, 1),
+                             (64, 4096, 1024, 0)]:
+        x = (torch.randn(rows, K, device='cuda') * 0.3).bfloat16()
+        w = (torch.randn(N, K, device='cuda') * 0.05).bfloat16()
+        b32 = torch.randn(N, device='cuda')
+        pk = FD._sk2_pack(w)
+        pk8, ws = FD._sk2_pack_fp8(w)
+        us_bf = gtime(lambda: ext.sk2(x, pk, b32, N, K, mode))
+        us_f8 = gtime(lambda: ext.sk2(x, pk8, b32, N, K, mode, ws))
+        print(f'M{rows} K{K} N{N} mode{mode}: bf16 {us_bf:6.2f}us  '
+              f'fp8 {us_f8:6.2f}us')
