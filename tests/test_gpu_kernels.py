@@ -804,6 +804,7 @@ def test_sk2_fp8_weights_close_to_bf16():
     quantization error only — compare against the fp32 reference computed
     from the DEQUANTIZED weights (exact-path check) and loosely against the
     unquantized weights (end-to-end error budget)."""
+    import torch.nn.functional as F
     import dalle_pytorch_amd._hip as ext
     from dalle_pytorch_amd.engine.decode import FastDecoder
     torch.manual_seed(1)
