@@ -164,12 +164,10 @@ def main():
 
     decoder = None
     if args.mode == 'generate':
-        # NOTE: fp8-weight decode (DALLE_AMD_FP8_DECODE=1) measured SLOWER
-        # than bf16 sk2 at these shapes (8.6 vs 6.9 us qkv — the kernels are
-        # latency-bound, not weight-byte-bound, and the in-register bf16->
-        # e4m3 conversion costs VALU); --fp8 does NOT enable it
-        if args.fp8 and os.environ.get('DALLE_AMD_FP8_DECODE') is None:
-            pass
+        # fp8-weight decode (DALLE_AMD_FP8_DECODE=1) measured SLOWER than
+        # bf16 sk2 at these shapes (8.6 vs 6.9 us qkv: the kernels are
+        # latency-bound, not weight-byte-bound, and the in-register
+        # bf16->e4m3 conversion costs VALU), so --fp8 does NOT enable it
         dalle.eval()
         if not args.eager:
             from dalle_pytorch_amd.engine import FastDecoder
