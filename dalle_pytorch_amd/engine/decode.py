@@ -270,7 +270,6 @@ class FastDecoder:
         else:
             net = st.leaf.net
             w['ff1_w'] = cast(net[0].weight)
-            w['ff1_T'] = castT(net[0].weight)
             w['ff1_pk'] = self._sk2_pack(w['ff1_w'])
             # the geglu-fused epilogue (mode 1) needs N = 2 * ff2-in width
             if w['ff1_pk'] is not None and net[0].weight.shape[0] % 64:
@@ -278,7 +277,6 @@ class FastDecoder:
             w['ff1_b'] = cast(net[0].bias)
             w['ff1_b32'] = castf(net[0].bias)
             w['ff2_w'] = cast(net[3].weight)
-            w['ff2_T'] = castT(net[3].weight)
             w['ff2_pk'] = self._sk2_pack(w['ff2_w'])
             if self._want_fp8_decode():
                 w['ff1_w_pk8'] = self._sk2_pack_fp8(w['ff1_w'])
